@@ -1,0 +1,313 @@
+"""Framework IR: the executable graph lowered from a TF GraphDef.
+
+The reference keeps the TF graph opaque and runs it through
+`tf.Session.run`; here the GraphDef is lowered into a small explicit IR
+(topo-sorted node list + weight dict) that the executor dispatches over
+MI355X HIP kernels (hot ops) or PyTorch-ROCm tensor ops (glue ops).
+
+Lowering passes (inference-oriented):
+  - strip Identity / NoOp / control inputs
+  - fold FusedBatchNorm(V2/V3) into the producing Conv2D's weight+bias
+  - fuse Conv2D + BiasAdd + Relu (+ residual AddV2) into one ConvBlock
+  - fuse MatMul + BiasAdd (+ activation) into Dense
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+
+from ..ckpt import graphdef as gd
+
+
+@dataclass
+class Node:
+    name: str
+    op: str
+    inputs: List[str]
+    attrs: dict = field(default_factory=dict)
+
+
+@dataclass
+class Graph:
+    nodes: List[Node]
+    consts: Dict[str, np.ndarray]
+    inputs: List[str]
+    outputs: List[str]
+
+    def node_map(self) -> Dict[str, Node]:
+        return {n.name: n for n in self.nodes}
+
+
+def _canon(name: str) -> str:
+    """Canonical tensor name: strip the ':0' output index and '^' ctrl marker."""
+    if name.startswith("^"):
+        return ""
+    if ":" in name:
+        base, idx = name.rsplit(":", 1)
+        if idx == "0":
+            return base
+        return name
+    return name
+
+
+def _attr_plain(av: gd.AttrValue):
+    for v in (av.s, av.i, av.f, av.b, av.type, av.shape, av.tensor,
+              av.list_i, av.list_f, av.list_s, av.list_type, av.list_shape):
+        if v is not None:
+            return v.decode() if isinstance(v, bytes) else v
+    return None
+
+
+def from_graphdef(g: gd.GraphDef,
+                  outputs: Optional[List[str]] = None) -> Graph:
+    """Lower a parsed GraphDef to the IR (no fusion yet — see `optimize`)."""
+    nodes: List[Node] = []
+    consts: Dict[str, np.ndarray] = {}
+    inputs: List[str] = []
+    for nd in g.node:
+        attrs = {k: _attr_plain(v) for k, v in nd.attr.items()}
+        if nd.op == "Const":
+            consts[nd.name] = nd.attr["value"].tensor
+            continue
+        if nd.op == "Placeholder":
+            inputs.append(nd.name)
+            continue
+        ins = [c for c in (_canon(i) for i in nd.input) if c]
+        nodes.append(Node(nd.name, nd.op, ins, attrs))
+
+    if outputs is None:
+        produced = {n.name for n in nodes}
+        consumed = set()
+        for n in nodes:
+            consumed.update(n.inputs)
+        outputs = [n for n in produced if n not in consumed]
+    return Graph(nodes, consts, inputs, outputs)
+
+
+# ------------------------------------------------------------- passes
+
+
+def strip_identity(graph: Graph) -> Graph:
+    """Remove Identity/NoOp nodes, rewiring consumers."""
+    alias: Dict[str, str] = {}
+    keep: List[Node] = []
+    for n in graph.nodes:
+        if n.op in ("Identity", "NoOp", "StopGradient", "CheckNumerics") and n.inputs:
+            alias[n.name] = n.inputs[0]
+        else:
+            keep.append(n)
+
+    def resolve(name: str) -> str:
+        seen = set()
+        while name in alias and name not in seen:
+            seen.add(name)
+            name = alias[name]
+        return name
+
+    for n in keep:
+        n.inputs = [resolve(i) for i in n.inputs]
+    outputs = [resolve(o) for o in graph.outputs]
+    return Graph(keep, graph.consts, graph.inputs, outputs)
+
+
+def fold_batchnorm(graph: Graph) -> Graph:
+    """Fold inference FusedBatchNorm(V2/V3) into the upstream Conv2D.
+
+    y = gamma * (x - mean) / sqrt(var + eps) + beta
+      = x * (gamma * rsqrt(var+eps)) + (beta - mean * gamma * rsqrt(var+eps))
+
+    When x is a Conv2D output: scale folds into the HWIO filter's O axis,
+    shift becomes a bias. When x is not a conv (or the conv has >1
+    consumer), the BN is rewritten to a Mul+Add pair via a 'BnScaleBias'
+    node instead.
+    """
+    consumers: Dict[str, int] = {}
+    for n in graph.nodes:
+        for i in n.inputs:
+            consumers[i] = consumers.get(i, 0) + 1
+    nmap = graph.node_map()
+    new_nodes: List[Node] = []
+    replaced: Dict[str, str] = {}
+
+    for n in graph.nodes:
+        if n.op in ("FusedBatchNorm", "FusedBatchNormV2", "FusedBatchNormV3"):
+            x, gamma_n, beta_n, mean_n, var_n = n.inputs[:5]
+            if not all(k in graph.consts for k in (gamma_n, beta_n, mean_n, var_n)):
+                new_nodes.append(n)
+                continue
+            eps = float(n.attrs.get("epsilon", 1e-3) or 1e-3)
+            gamma = graph.consts[gamma_n].astype(np.float64)
+            beta = graph.consts[beta_n].astype(np.float64)
+            mean = graph.consts[mean_n].astype(np.float64)
+            var = graph.consts[var_n].astype(np.float64)
+            scale = gamma / np.sqrt(var + eps)
+            shift = beta - mean * scale
+            prod = nmap.get(x)
+            if (prod is not None and prod.op == "Conv2D"
+                    and consumers.get(x, 0) == 1
+                    and prod.inputs[1] in graph.consts):
+                w = graph.consts[prod.inputs[1]]  # HWIO
+                graph.consts[prod.inputs[1]] = (
+                    w.astype(np.float64) * scale[None, None, None, :]
+                ).astype(w.dtype)
+                bias_name = n.name + "/folded_bias"
+                graph.consts[bias_name] = shift.astype(np.float32)
+                bn_bias = Node(n.name, "BiasAdd", [x, bias_name],
+                               {"data_format": "NHWC"})
+                new_nodes.append(bn_bias)
+            else:
+                sc_name = n.name + "/bn_scale"
+                sh_name = n.name + "/bn_shift"
+                graph.consts[sc_name] = scale.astype(np.float32)
+                graph.consts[sh_name] = shift.astype(np.float32)
+                new_nodes.append(Node(n.name, "BnScaleBias", [x, sc_name, sh_name], {}))
+            replaced[n.name] = n.name
+        else:
+            new_nodes.append(n)
+    return Graph(new_nodes, graph.consts, graph.inputs, graph.outputs)
+
+
+def fuse_conv_blocks(graph: Graph) -> Graph:
+    """Fuse Conv2D [+BiasAdd] [+AddV2 residual] [+Relu] -> ConvBlock,
+    and MatMul [+BiasAdd] [+Relu|Softmax kept separate] -> Dense.
+
+    ConvBlock attrs: strides, padding, plus fusion fields:
+      bias: const name or None; residual: tensor name or None; relu: bool.
+    """
+    consumers: Dict[str, List[Node]] = {}
+    for n in graph.nodes:
+        for i in n.inputs:
+            consumers.setdefault(i, []).append(n)
+    out_set = set(graph.outputs)
+    drop = set()
+    created = set()  # id() of fused nodes (they reuse the tail's name)
+    new_nodes: List[Node] = []
+    pos = {n.name: i for i, n in enumerate(graph.nodes)}
+
+    def sole_consumer(name: str) -> Optional[Node]:
+        cs = consumers.get(name, [])
+        if len(cs) == 1 and name not in out_set:
+            return cs[0]
+        return None
+
+    for n in graph.nodes:
+        if n.name in drop:
+            continue
+        if n.op in ("Conv2D", "DepthwiseConv2dNative"):
+            bias = None
+            residual = None
+            relu = False
+            tail = n
+            c = sole_consumer(tail.name)
+            if c is not None and c.op == "BiasAdd" and c.inputs[1] in graph.consts:
+                bias = c.inputs[1]
+                drop.add(c.name)
+                tail = c
+                c = sole_consumer(tail.name)
+            if c is not None and c.op in ("Add", "AddV2"):
+                other = [i for i in c.inputs if i != tail.name]
+                # Only the MAIN branch absorbs the residual Add: the other
+                # operand must already be produced (earlier in node order),
+                # else both branches would claim the Add (duplicate nodes).
+                if (len(other) == 1 and other[0] not in graph.consts
+                        and pos.get(other[0], 1 << 30) < pos[n.name]):
+                    residual = other[0]
+                    drop.add(c.name)
+                    tail = c
+                    c = sole_consumer(tail.name)
+            if c is not None and c.op == "Relu":
+                relu = True
+                drop.add(c.name)
+                tail = c
+            op = "ConvBlock" if n.op == "Conv2D" else "DepthwiseConvBlock"
+            attrs = dict(n.attrs)
+            attrs.update({"bias": bias, "residual": residual, "relu": relu})
+            ins = [n.inputs[0], n.inputs[1]]
+            if bias:
+                ins.append(bias)
+            if residual:
+                ins.append(residual)
+            nb = Node(tail.name, op, ins, attrs)
+            created.add(id(nb))
+            new_nodes.append(nb)
+        elif n.op == "MatMul":
+            bias = None
+            relu = False
+            tail = n
+            c = sole_consumer(tail.name)
+            if c is not None and c.op == "BiasAdd" and c.inputs[1] in graph.consts:
+                bias = c.inputs[1]
+                drop.add(c.name)
+                tail = c
+                c = sole_consumer(tail.name)
+            if c is not None and c.op == "Relu":
+                relu = True
+                drop.add(c.name)
+                tail = c
+            attrs = dict(n.attrs)
+            attrs.update({"bias": bias, "relu": relu})
+            ins = [n.inputs[0], n.inputs[1]]
+            if bias:
+                ins.append(bias)
+            dn = Node(tail.name, "Dense", ins, attrs)
+            created.add(id(dn))
+            new_nodes.append(dn)
+        else:
+            new_nodes.append(n)
+
+    new_nodes = [n for n in new_nodes
+                 if id(n) in created or n.name not in drop]
+    return Graph(new_nodes, graph.consts, graph.inputs, graph.outputs)
+
+
+def toposort(graph: Graph) -> Graph:
+    nmap = graph.node_map()
+    order: List[Node] = []
+    state: Dict[str, int] = {}
+
+    def visit(name: str):
+        n = nmap.get(name)
+        if n is None:
+            return
+        st = state.get(name, 0)
+        if st == 1:
+            raise ValueError(f"cycle at {name}")
+        if st == 2:
+            return
+        state[name] = 1
+        for i in n.inputs:
+            visit(i)
+        state[name] = 2
+        order.append(n)
+
+    import sys
+    old = sys.getrecursionlimit()
+    sys.setrecursionlimit(max(old, 10 * len(graph.nodes) + 1000))
+    try:
+        for o in graph.outputs:
+            visit(o)
+    finally:
+        sys.setrecursionlimit(old)
+    return Graph(order, graph.consts, graph.inputs, graph.outputs)
+
+
+def optimize(graph: Graph) -> Graph:
+    g = strip_identity(graph)
+    g = fold_batchnorm(g)
+    g = fuse_conv_blocks(g)
+    g = toposort(g)
+    return g
+
+
+def load_model(path: str, outputs: Optional[List[str]] = None) -> Graph:
+    """Load a frozen .pb file or a SavedModel directory into optimized IR."""
+    import os
+    if os.path.isdir(path):
+        from ..ckpt.savedmodel import load_savedmodel
+        graph, sig_out = load_savedmodel(path)
+        return optimize(graph if outputs is None else
+                        Graph(graph.nodes, graph.consts, graph.inputs, outputs))
+    g = gd.load_frozen_graph(path)
+    return optimize(from_graphdef(g, outputs))

@@ -1,0 +1,77 @@
+"""Inference engines behind the HTTP layer.
+
+`LocalEngine` — one in-process executor (CPU fp32, or single-GPU HIP
+path). `ReplicatedEngine` (parallel/replicas.py) fans batches out over
+DP replicas via RCCL; both present `infer(batch) -> probs` so the
+server/batcher are backend-agnostic (and tests can use a FakeEngine).
+"""
+from __future__ import annotations
+
+from typing import List, Optional
+
+import numpy as np
+import torch
+
+from ..conf.config import Config
+from .executor import Executor
+from .ir import Graph, load_model
+
+
+def build_graph(cfg: Config) -> Graph:
+    if cfg.model_path:
+        return load_model(cfg.model_path, outputs=[cfg.output_name])
+    # built-in random-init model families (no network: BASELINE protocol)
+    from .. import models as M
+    from ..ckpt import graphdef as gd
+    from . import ir
+    builders = {
+        "lenet": lambda: M.lenet.build_lenet(),
+        "resnet50": lambda: M.resnet50.build_resnet50(),
+    }
+    try:
+        from ..models import inception_v3, vit
+        builders["inception_v3"] = lambda: inception_v3.build_inception_v3()
+        builders["vit_b16"] = lambda: vit.build_vit_b16()
+    except ImportError:
+        pass
+    if cfg.model_name not in builders:
+        raise ValueError(f"unknown built-in model {cfg.model_name}")
+    g = builders[cfg.model_name]()
+    return ir.optimize(ir.from_graphdef(g, outputs=[cfg.output_name]))
+
+
+class LocalEngine:
+    def __init__(self, cfg: Config, graph: Optional[Graph] = None):
+        self.cfg = cfg
+        self.graph = graph if graph is not None else build_graph(cfg)
+        dev = cfg.device
+        if dev == "auto":
+            dev = "cuda" if torch.cuda.is_available() else "cpu"
+        self.device = dev
+        if dev.startswith("cuda"):
+            from .gpu_executor import GpuExecutor
+            self.ex = GpuExecutor(self.graph, device=dev, cfg=cfg)
+        else:
+            self.ex = Executor(self.graph, device=dev)
+
+    @torch.inference_mode()
+    def infer(self, batch: np.ndarray) -> np.ndarray:
+        x = torch.from_numpy(np.ascontiguousarray(batch, dtype=np.float32))
+        out = self.ex.run({self.cfg.input_name: x})[0]
+        return out.float().cpu().numpy()
+
+
+class FakeEngine:
+    """Deterministic engine for HTTP-layer tests: prob peak at a hash of
+    the image bytes so tests can assert routing without a model."""
+
+    def __init__(self, num_classes: int = 10):
+        self.num_classes = num_classes
+
+    def infer(self, batch: np.ndarray) -> np.ndarray:
+        n = batch.shape[0]
+        out = np.full((n, self.num_classes), 1e-6, np.float32)
+        for i in range(n):
+            k = int(abs(float(batch[i].sum())) * 1000) % self.num_classes
+            out[i, k] = 1.0
+        return out / out.sum(axis=1, keepdims=True)

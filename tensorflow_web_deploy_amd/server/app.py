@@ -1,0 +1,114 @@
+"""Flask HTTP serving surface (reference-parity endpoints).
+
+Reference class surface: `GET /` renders an upload form; `POST /predict`
+takes a multipart image upload, runs preprocess -> model forward ->
+top-k, returns predictions. We keep that surface and add operational
+endpoints (/healthz, /metrics) and a JSON response schema:
+
+  POST /predict  (multipart 'file', or raw image bytes)
+    -> {"predictions": [{"label": str, "class_id": int, "prob": float}...],
+        "model": str, "latency_ms": float}
+"""
+from __future__ import annotations
+
+import time
+from typing import Optional
+
+import numpy as np
+from flask import Flask, Response, jsonify, render_template, request
+
+from ..conf.config import Config
+from ..ingest.preprocess import decode_image
+from .labels import load_labels
+
+try:
+    from prometheus_client import (CONTENT_TYPE_LATEST, Counter, Histogram,
+                                   generate_latest)
+    _PROM = True
+except Exception:  # pragma: no cover
+    _PROM = False
+
+
+def create_app(cfg: Config, engine) -> Flask:
+    app = Flask(__name__)
+    labels = load_labels(cfg.labels_path, cfg.num_classes)
+
+    if _PROM:
+        req_counter = Counter("twd_requests_total", "predict requests",
+                              ["status"])
+        lat_hist = Histogram("twd_request_latency_seconds",
+                             "end-to-end predict latency")
+
+    @app.get("/")
+    def index():
+        return render_template("index.html", model=cfg.model_name or
+                               cfg.model_path)
+
+    @app.get("/healthz")
+    def healthz():
+        return jsonify({"status": "ok"})
+
+    @app.post("/predict")
+    def predict():
+        t0 = time.perf_counter()
+        data: Optional[bytes] = None
+        if "file" in request.files:
+            data = request.files["file"].read()
+        elif request.data:
+            data = request.data
+        if not data:
+            if _PROM:
+                req_counter.labels("bad_request").inc()
+            return jsonify({"error": "no image provided"}), 400
+        try:
+            arr = decode_image(data, (cfg.input_width, cfg.input_height),
+                               cfg.input_channels, cfg.normalize)
+        except Exception as e:
+            if _PROM:
+                req_counter.labels("bad_image").inc()
+            return jsonify({"error": f"cannot decode image: {e}"}), 400
+        probs = engine.infer(arr[None])[0]
+        k = min(cfg.top_k, len(probs))
+        top = np.argsort(probs)[::-1][:k]
+        dt = (time.perf_counter() - t0) * 1000.0
+        if _PROM:
+            req_counter.labels("ok").inc()
+            lat_hist.observe(dt / 1000.0)
+        return jsonify({
+            "predictions": [
+                {"label": labels[int(i)], "class_id": int(i),
+                 "prob": float(probs[int(i)])}
+                for i in top
+            ],
+            "model": cfg.model_name or cfg.model_path,
+            "latency_ms": dt,
+        })
+
+    if _PROM:
+        @app.get("/metrics")
+        def metrics():
+            return Response(generate_latest(), mimetype=CONTENT_TYPE_LATEST)
+
+    return app
+
+
+def main(argv=None):
+    import argparse
+    from ..conf.config import add_config_args, config_from_args
+    from ..engine.service import LocalEngine
+
+    p = argparse.ArgumentParser("tensorflow_web_deploy_amd server")
+    add_config_args(p)
+    args = p.parse_args(argv)
+    cfg = config_from_args(args)
+    if cfg.gpus > 1:
+        from ..parallel.replicas import ReplicatedEngine
+        engine = ReplicatedEngine(cfg)
+    else:
+        engine = LocalEngine(cfg)
+    app = create_app(cfg, engine)
+    app.run(host=cfg.host, port=cfg.port, threaded=True)
+
+
+if __name__ == "__main__":
+    main()

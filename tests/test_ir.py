@@ -1,0 +1,68 @@
+"""IR lowering-pass tests: optimized graph must match the raw graph."""
+import numpy as np
+import torch
+
+from tensorflow_web_deploy_amd.ckpt import graphdef as gd
+from tensorflow_web_deploy_amd.engine import ir
+from tensorflow_web_deploy_amd.engine.executor import Executor
+from tensorflow_web_deploy_amd.models.lenet import build_lenet
+from tensorflow_web_deploy_amd.models.resnet50 import build_resnet50
+
+
+def _run(graph, feeds):
+    return Executor(graph).run(feeds)[0]
+
+
+def test_lenet_fusion_equivalence():
+    g = build_lenet(seed=3)
+    raw = ir.toposort(ir.strip_identity(ir.from_graphdef(g, ["softmax"])))
+    opt = ir.optimize(ir.from_graphdef(g, ["softmax"]))
+    x = torch.randn(4, 28, 28, 1)
+    y_raw = _run(raw, {"input": x})
+    y_opt = _run(opt, {"input": x})
+    torch.testing.assert_close(y_raw, y_opt, rtol=1e-5, atol=1e-6)
+    ops = [n.op for n in opt.nodes]
+    assert "Conv2D" not in ops and "BiasAdd" not in ops
+    assert ops.count("ConvBlock") == 2 and ops.count("Dense") == 2
+
+
+def test_resnet_block_bn_fold_equivalence():
+    # one-bottleneck mini-resnet: BN fold + residual fusion must be exact
+    from tensorflow_web_deploy_amd.models.builder import GraphBuilder
+    from tensorflow_web_deploy_amd.models.resnet50 import _bottleneck
+    b = GraphBuilder(seed=11)
+    x = b.placeholder("input", [-1, 16, 16, 8])
+    y = _bottleneck(b, x, "blk", 8, 4, 1)
+    b.node("Softmax", "softmax", [b.global_mean(y, "gp")])
+    g = b.graphdef()
+    raw = ir.toposort(ir.strip_identity(ir.from_graphdef(g, ["softmax"])))
+    opt = ir.optimize(ir.from_graphdef(g, ["softmax"]))
+    xv = torch.randn(2, 16, 16, 8)
+    y_raw = _run(raw, {"input": xv})
+    y_opt = _run(opt, {"input": xv})
+    torch.testing.assert_close(y_raw, y_opt, rtol=1e-4, atol=1e-5)
+    ops = [n.op for n in opt.nodes]
+    assert "FusedBatchNormV3" not in ops
+    assert "AddV2" not in ops  # absorbed into the main-branch ConvBlock
+
+
+def test_resnet50_structure():
+    g = build_resnet50(seed=0)
+    opt = ir.optimize(ir.from_graphdef(g, ["softmax"]))
+    from collections import Counter
+    c = Counter(n.op for n in opt.nodes)
+    assert c["ConvBlock"] == 53
+    assert c["MaxPool"] == 1 and c["Dense"] == 1 and c["Softmax"] == 1
+    # exactly 16 residual-fused blocks
+    res = [n for n in opt.nodes if n.op == "ConvBlock" and n.attrs.get("residual")]
+    assert len(res) == 16
+
+
+def test_same_padding_asymmetric():
+    # TF SAME on even input with stride 2: pad goes bottom/right
+    from tensorflow_web_deploy_amd.engine.executor import tf_same_pad
+    x = torch.arange(16, dtype=torch.float32).reshape(1, 1, 4, 4)
+    p = tf_same_pad(x, 3, 3, 2, 2)
+    assert p.shape == (1, 1, 5, 5)
+    assert p[0, 0, 4].sum() == 0 and p[0, 0, :, 4].sum() == 0  # pad at end
+    assert p[0, 0, 0, 0] == 0.0 and p[0, 0, 0, 1] == 1.0  # no pad at start
