@@ -29,15 +29,17 @@ except Exception:  # pragma: no cover
     _PROM = False
 
 
+if _PROM:
+    # module-level singletons: create_app may be called more than once
+    # (tests, multi-app) and the default registry rejects duplicates
+    req_counter = Counter("twd_requests_total", "predict requests", ["status"])
+    lat_hist = Histogram("twd_request_latency_seconds",
+                         "end-to-end predict latency")
+
+
 def create_app(cfg: Config, engine) -> Flask:
     app = Flask(__name__)
     labels = load_labels(cfg.labels_path, cfg.num_classes)
-
-    if _PROM:
-        req_counter = Counter("twd_requests_total", "predict requests",
-                              ["status"])
-        lat_hist = Histogram("twd_request_latency_seconds",
-                             "end-to-end predict latency")
 
     @app.get("/")
     def index():
