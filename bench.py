@@ -1,0 +1,118 @@
+"""Driver benchmark contract: flagship serving step on N GPUs of one node.
+
+Measures the BASELINE.json headline metric: images/sec (whole node) for
+ResNet-50 224x224 bf16, batch=64 per GPU (weak scaling), synthetic
+images, random-init weights, through the native HIP-kernel engine
+(captured HIP graph per batch size).
+
+  python bench.py --gpus N --steps K --warmup W
+For N>1 the driver launches via torch.distributed.run (one rank per GPU,
+RCCL); rank 0 prints ONE JSON line.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, ROOT)
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--batch", type=int, default=64)
+    p.add_argument("--model", default="resnet50")
+    args = p.parse_args()
+
+    import numpy as np
+    import torch
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+        dist = dist_mod
+        torch.cuda.set_device(local_rank)
+        dist.init_process_group(backend="nccl")  # RCCL on ROCm
+
+    assert torch.cuda.is_available(), "bench needs a GPU"
+    device = f"cuda:{local_rank}"
+    torch.cuda.set_device(device)
+
+    from tensorflow_web_deploy_amd.conf.config import Config
+    from tensorflow_web_deploy_amd.engine.service import LocalEngine
+    from tensorflow_web_deploy_amd.ops.native import require_native_on_gpu
+    require_native_on_gpu()
+
+    size = 299 if args.model == "inception_v3" else 224
+    cfg = Config(model_name=args.model, device=device, input_height=size,
+                 input_width=size, input_channels=3, num_classes=1000)
+    eng = LocalEngine(cfg)
+
+    # synthetic images of the benchmark shape (no network: BASELINE.json)
+    rng = np.random.default_rng(1234 + rank)
+    x = rng.random((args.batch, size, size, 3), dtype=np.float32)
+    xt = torch.from_numpy(x).to(device)
+
+    feeds = {cfg.input_name: xt}
+
+    # warmup (also compiles + captures the HIP graph)
+    for _ in range(args.warmup):
+        eng.ex.run(feeds)
+    torch.cuda.synchronize()
+    if dist:
+        dist.barrier()
+        torch.cuda.synchronize()
+
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        eng.ex.run(feeds)
+    torch.cuda.synchronize()
+    if dist:
+        dist.barrier()
+        torch.cuda.synchronize()
+    t1 = time.perf_counter()
+
+    elapsed = t1 - t0
+    if dist:
+        t = torch.tensor([elapsed], device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    total_images = args.batch * args.steps * world
+    ips = total_images / elapsed
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "images_per_sec",
+            "value": round(ips, 2),
+            "unit": "images/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch * world,
+                "image_size": size,
+                "parallelism": f"dp{world}",
+            },
+        }))
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,37 @@
+"""In-tree build of the HIP extension (`_hipops`) for gfx950.
+
+Build:  PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+The built .so lands inside tensorflow_web_deploy_amd/ so it travels to
+GPU boxes with the repo snapshot.
+"""
+import os
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from setuptools import setup  # noqa: E402
+from torch.utils import cpp_extension  # noqa: E402
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+CSRC = os.path.join(ROOT, "tensorflow_web_deploy_amd", "csrc")
+
+ext = cpp_extension.CUDAExtension(
+    name="tensorflow_web_deploy_amd._hipops",
+    sources=[
+        os.path.join(CSRC, "bindings.cpp"),
+        os.path.join(CSRC, "gemm_conv.hip"),
+        os.path.join(CSRC, "pool_reduce.hip"),
+        os.path.join(CSRC, "elementwise.hip"),
+    ],
+    extra_compile_args={
+        "cxx": ["-O3", "-std=c++17"],
+        "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+    },
+)
+
+setup(
+    name="tensorflow_web_deploy_amd",
+    version="0.1.0",
+    packages=["tensorflow_web_deploy_amd"],
+    ext_modules=[ext],
+    cmdclass={"build_ext": cpp_extension.BuildExtension},
+)

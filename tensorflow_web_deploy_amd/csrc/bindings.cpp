@@ -1,0 +1,175 @@
+// pybind11/torch bindings for the CDNA4 HIP kernels (native path of the
+// engine). Pure HIP — no CUDA compatibility layer: streams come from
+// c10::hip directly.
+#include <torch/extension.h>
+
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include "kernels.h"
+
+namespace {
+
+hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+void check_bf16(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+// C[M][N] = act(A[M][K] @ B[N][K]^T + bias [+ residual])
+void gemm_bf16(torch::Tensor A, torch::Tensor B, torch::Tensor C,
+               c10::optional<torch::Tensor> bias,
+               c10::optional<torch::Tensor> residual, int64_t K,
+               bool relu) {
+  check_bf16(A, "A");
+  check_bf16(B, "B");
+  check_bf16(C, "C");
+  int64_t M = C.size(0), N = C.size(1);
+  int64_t Kp = B.size(1);
+  TORCH_CHECK(B.size(0) >= N, "B rows < N");
+  TORCH_CHECK(K % 8 == 0, "K must be a multiple of 8");
+  const float* bptr = nullptr;
+  if (bias) {
+    TORCH_CHECK(bias->scalar_type() == torch::kFloat32, "bias must be f32");
+    bptr = bias->data_ptr<float>();
+  }
+  const void* rptr = nullptr;
+  if (residual) {
+    check_bf16(*residual, "residual");
+    rptr = residual->data_ptr();
+  }
+  launch_gemm_bf16(A.data_ptr(), B.data_ptr(), C.data_ptr(), bptr, rptr,
+                   (int)M, (int)N, (int)K, (int)Kp,
+                   A.numel() * 2, relu, cur_stream());
+}
+
+// X [N,H,W,C] bf16, W2 [Cout][Kp] (pre-transposed HWIO, K-padded)
+void conv2d_bf16(torch::Tensor X, torch::Tensor W2, torch::Tensor Y,
+                 c10::optional<torch::Tensor> bias,
+                 c10::optional<torch::Tensor> residual, int64_t R, int64_t S,
+                 int64_t sh, int64_t sw, int64_t pt, int64_t pl, bool relu) {
+  check_bf16(X, "X");
+  check_bf16(W2, "W");
+  check_bf16(Y, "Y");
+  TORCH_CHECK(X.dim() == 4 && Y.dim() == 4);
+  int Nimg = X.size(0), H = X.size(1), W = X.size(2), Cin = X.size(3);
+  int HO = Y.size(1), WO = Y.size(2), Cout = Y.size(3);
+  TORCH_CHECK(Cin % 8 == 0, "Cin must be a multiple of 8 (pad first)");
+  const float* bptr = nullptr;
+  if (bias) bptr = bias->data_ptr<float>();
+  const void* rptr = nullptr;
+  if (residual) {
+    check_bf16(*residual, "residual");
+    rptr = residual->data_ptr();
+  }
+  launch_conv2d_bf16(X.data_ptr(), W2.data_ptr(), Y.data_ptr(), bptr, rptr,
+                     Nimg, H, W, Cin, (int)R, (int)S, Cout, (int)sh, (int)sw,
+                     (int)pt, (int)pl, HO, WO, (int)W2.size(1), relu,
+                     cur_stream());
+}
+
+void maxpool_bf16(torch::Tensor X, torch::Tensor Y, int64_t kh, int64_t kw,
+                  int64_t sh, int64_t sw, int64_t pt, int64_t pl) {
+  check_bf16(X, "X");
+  check_bf16(Y, "Y");
+  launch_maxpool_bf16(X.data_ptr(), Y.data_ptr(), X.size(0), X.size(1),
+                      X.size(2), X.size(3), kh, kw, sh, sw, pt, pl, Y.size(1),
+                      Y.size(2), cur_stream());
+}
+
+void avgpool_bf16(torch::Tensor X, torch::Tensor Y, int64_t kh, int64_t kw,
+                  int64_t sh, int64_t sw, int64_t pt, int64_t pl) {
+  check_bf16(X, "X");
+  check_bf16(Y, "Y");
+  launch_avgpool_bf16(X.data_ptr(), Y.data_ptr(), X.size(0), X.size(1),
+                      X.size(2), X.size(3), kh, kw, sh, sw, pt, pl, Y.size(1),
+                      Y.size(2), cur_stream());
+}
+
+void global_mean_bf16(torch::Tensor X, torch::Tensor Y) {
+  check_bf16(X, "X");
+  check_bf16(Y, "Y");
+  launch_global_mean_bf16(X.data_ptr(), Y.data_ptr(), X.size(0),
+                          X.size(1) * X.size(2), X.size(3), cur_stream());
+}
+
+void softmax_bf16(torch::Tensor X, torch::Tensor Y) {
+  check_bf16(X, "X");
+  TORCH_CHECK(Y.scalar_type() == torch::kFloat32);
+  launch_softmax_bf16(X.data_ptr(), Y.data_ptr<float>(), X.size(0), X.size(1),
+                      cur_stream());
+}
+
+void cast_f32_bf16(torch::Tensor X, torch::Tensor Y) {
+  TORCH_CHECK(X.scalar_type() == torch::kFloat32 && X.is_contiguous());
+  check_bf16(Y, "Y");
+  TORCH_CHECK(X.numel() % 8 == 0);
+  launch_cast_f32_bf16(X.data_ptr<float>(), Y.data_ptr(), X.numel(),
+                       cur_stream());
+}
+
+void cast_bf16_f32(torch::Tensor X, torch::Tensor Y) {
+  check_bf16(X, "X");
+  TORCH_CHECK(Y.scalar_type() == torch::kFloat32);
+  launch_cast_bf16_f32(X.data_ptr(), Y.data_ptr<float>(), X.numel(),
+                       cur_stream());
+}
+
+void channel_pad_bf16(torch::Tensor X, torch::Tensor Y) {
+  check_bf16(X, "X");
+  check_bf16(Y, "Y");
+  int64_t rows = X.numel() / X.size(-1);
+  launch_channel_pad_bf16(X.data_ptr(), Y.data_ptr(), rows, X.size(-1),
+                          Y.size(-1), cur_stream());
+}
+
+void add_relu_bf16(torch::Tensor A, torch::Tensor B, torch::Tensor Y,
+                   bool relu) {
+  check_bf16(A, "A");
+  check_bf16(B, "B");
+  check_bf16(Y, "Y");
+  launch_add_relu_bf16(A.data_ptr(), B.data_ptr(), Y.data_ptr(), A.numel(),
+                       relu, cur_stream());
+}
+
+void bias_act_bf16(torch::Tensor X, c10::optional<torch::Tensor> bias,
+                   torch::Tensor Y, int64_t act) {
+  check_bf16(X, "X");
+  check_bf16(Y, "Y");
+  const float* bptr = nullptr;
+  if (bias) bptr = bias->data_ptr<float>();
+  int C = X.size(-1);
+  launch_bias_act_bf16(X.data_ptr(), bptr, Y.data_ptr(), X.numel() / C, C,
+                       (int)act, cur_stream());
+}
+
+void scale_shift_bf16(torch::Tensor X, torch::Tensor scale,
+                      torch::Tensor shift, torch::Tensor Y, bool relu) {
+  check_bf16(X, "X");
+  check_bf16(Y, "Y");
+  int C = X.size(-1);
+  launch_scale_shift_bf16(X.data_ptr(), scale.data_ptr<float>(),
+                          shift.data_ptr<float>(), Y.data_ptr(),
+                          X.numel() / C, C, relu, cur_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("gemm_bf16", &gemm_bf16, "bf16 MFMA GEMM (A @ B^T + bias, act)");
+  m.def("conv2d_bf16", &conv2d_bf16, "bf16 implicit-GEMM conv2d NHWC");
+  m.def("maxpool_bf16", &maxpool_bf16);
+  m.def("avgpool_bf16", &avgpool_bf16);
+  m.def("global_mean_bf16", &global_mean_bf16);
+  m.def("softmax_bf16", &softmax_bf16);
+  m.def("cast_f32_bf16", &cast_f32_bf16);
+  m.def("cast_bf16_f32", &cast_bf16_f32);
+  m.def("channel_pad_bf16", &channel_pad_bf16);
+  m.def("add_relu_bf16", &add_relu_bf16);
+  m.def("bias_act_bf16", &bias_act_bf16);
+  m.def("scale_shift_bf16", &scale_shift_bf16);
+}

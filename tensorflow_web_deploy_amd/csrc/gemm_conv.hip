@@ -1,0 +1,343 @@
+// MFMA implicit-GEMM kernel for gfx950 (CDNA4) — the conv2d/dense hot path.
+//
+// Replaces the reference's TF-runtime Conv2D/MatMul kernels (cuDNN/cuBLAS
+// behind Session.run) with a hand-written CDNA4 kernel per the north star:
+// "MFMA im2col-GEMM with LDS-staged input tiles".
+//
+// Computes  C[M][N] = act( A[M][K] @ B[N][K]^T + bias [+ residual] )
+//   A: activations, NHWC-flattened rows (bf16). For conv, rows are gathered
+//      im2col-style on the fly while staging (per-lane source addresses into
+//      a buffer descriptor whose hardware OOB check supplies the zero
+//      padding — no zero-page buffers, no branches in the load path).
+//   B: weights pre-transposed to [N][K] (HWIO -> [O][R*S*C]), K padded to a
+//      multiple of BK with zeros at load time (Python side).
+//   C: bf16 [M][N] row-major = the next layer's NHWC activation.
+//
+// Structure (cdna_hip_programming.md §5 "step-3" verified shape):
+//   128x128 tile, BK=64, 256 threads = 4 waves (each computing 64x64 via
+//   4x4 fragments of v_mfma_f32_16x16x32_bf16), double-buffered LDS filled
+//   by global_load_lds (buffer variant), XOR-swizzled LDS reads
+//   (slot ^= row&7) with the inverse swizzle applied to the *source*
+//   address (guide §5.4 rule 21), XCD-aware bijective block remap (T1),
+//   epilogue staged through LDS for wide dwordx4 stores.
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <cstdint>
+
+#include "kernels.h"
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+#define BM 128
+#define BN 128
+#define BK 64
+#define NTHREADS 256
+
+// LDS: A/B tiles double buffered: [2][128][64] bf16 each = 32 KiB per op.
+// The epilogue reuses the first 32 KiB as a [128][128] bf16 C tile.
+#define LDS_BYTES (4 * BM * BK * 2)
+#define A_OFF(buf) ((buf) * (BM * BK * 2))
+#define B_OFF(buf) ((2 + (buf)) * (BM * BK * 2))
+
+extern __shared__ __attribute__((aligned(16))) char smem[];
+
+struct MagicDiv {
+  uint32_t m_lo, m_hi;  // 64-bit magic split (host packs m = ceil(2^40/d))
+  uint32_t pad0, pad1;
+};
+
+// n / d for n < 2^24, d < 2^16 with m = ceil(2^40/d)
+__device__ __forceinline__ uint32_t mdiv(uint32_t n, uint64_t m) {
+  return (uint32_t)(((uint64_t)n * m) >> 40);
+}
+
+__device__ __forceinline__ float bf2f(__bf16 v) { return (float)v; }
+
+// ---------------------------------------------------------------- kernel
+
+struct ConvParams {
+  // geometry (conv mode); unused in linear mode
+  int C;        // input channels
+  int H, W;     // input spatial
+  int R, S;     // filter spatial
+  int sh, sw;   // strides
+  int pt, pl;   // top/left pad
+  int HO, WO;   // output spatial
+  uint64_t magic_C, magic_S, magic_HW, magic_WO;
+};
+
+// MODE: 0 = linear A rows (dense / 1x1 stride-1 conv); 1 = im2col gather
+template <int MODE, bool HAS_RES>
+__global__ __launch_bounds__(NTHREADS)
+void gemm_conv_kernel(const __bf16* __restrict__ A,
+                      const __bf16* __restrict__ B,
+                      __bf16* __restrict__ Cout,
+                      const float* __restrict__ bias,
+                      const __bf16* __restrict__ residual,
+                      int M, int N, int K, int Kp,
+                      int a_bytes, int tiles_m, int tiles_n,
+                      int relu, ConvParams cp) {
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = __builtin_amdgcn_readfirstlane(tid >> 6);
+
+  // ---- XCD-aware bijective remap (guide §5: T1, bijective variant) ----
+  int nwg = tiles_m * tiles_n;
+  int bid = blockIdx.x;
+  {
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = bid & 7, off = bid >> 3;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
+  }
+  // tm fastest => consecutive blocks share the B panel (L2 reuse)
+  const int tn = bid / tiles_m;
+  const int tm = bid - tn * tiles_m;
+  const int row0 = tm * BM;
+  const int col0 = tn * BN;
+
+  const auto rsrcA = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)A, /*stride*/ 0, a_bytes, 0x00020000);
+  const auto rsrcB = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)B, 0, (int)((int64_t)N * Kp * 2), 0x00020000);
+
+  // ---- per-lane staging state ----
+  // Staging instruction j (j = wave*4 + i) covers tile rows j*8 + lane/8,
+  // lane slot = lane&7; source element k = kb + (slot ^ (row&7))*8.
+  const int srow_base = (lane >> 3);        // 0..7
+  const int sslot = lane & 7;
+  int32_t a_base[4];   // element offset of (row, k=0) or im2col base
+  int32_t a_hb[4], a_wb[4];
+  bool a_valid[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int j = wave * 4 + i;
+    int row_l = j * 8 + srow_base;
+    int row_g = row0 + row_l;
+    if (MODE == 0) {
+      a_valid[i] = row_g < M;
+      a_base[i] = row_g * K;
+    } else {
+      a_valid[i] = row_g < M;
+      uint32_t rg = (uint32_t)min(row_g, M - 1 > 0 ? M - 1 : 0);
+      uint32_t n_img = mdiv(rg, cp.magic_HW);
+      uint32_t rem = rg - n_img * (uint32_t)(cp.HO * cp.WO);
+      uint32_t ho = mdiv(rem, cp.magic_WO);
+      uint32_t wo = rem - ho * (uint32_t)cp.WO;
+      int hb = (int)ho * cp.sh - cp.pt;
+      int wb = (int)wo * cp.sw - cp.pl;
+      a_hb[i] = hb;
+      a_wb[i] = wb;
+      a_base[i] = (((int)n_img * cp.H + hb) * cp.W + wb) * cp.C;
+    }
+  }
+  // B rows (cols of C): same staging pattern
+  int32_t b_voff[4];
+  bool b_valid[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int j = wave * 4 + i;
+    int row_l = j * 8 + srow_base;
+    int col_g = col0 + row_l;
+    b_valid[i] = col_g < N;
+    b_voff[i] = col_g * Kp;
+  }
+
+  const int swz = srow_base & 7;  // row&7 of this lane's staged row
+
+  // ---- accumulators (+ free bias init) ----
+  const int wr = wave >> 1, wc = wave & 1;  // wave tile (64x64)
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      float b0 = 0.f;
+      if (bias) {
+        int col = col0 + wc * 64 + ni * 16 + (lane & 15);
+        if (col < N) b0 = bias[col];
+      }
+      acc[mi][ni] = {b0, b0, b0, b0};
+    }
+
+  const int nt = (K + BK - 1) / BK;
+
+  // ---- staging helper (as lambdas so the main loop stays readable) ----
+  auto stage = [&](int buf, int t) {
+    const int kb = t * BK;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      // A tile
+      int k = kb + ((sslot ^ swz) << 3);
+      int32_t voff;
+      if (MODE == 0) {
+        voff = (a_valid[i] && k < K) ? (a_base[i] + k) * 2 : 0x7ffffff0;
+      } else {
+        uint32_t rs = mdiv((uint32_t)k, cp.magic_C);
+        uint32_t c = (uint32_t)k - rs * (uint32_t)cp.C;
+        uint32_t r = mdiv(rs, cp.magic_S);
+        uint32_t s = rs - r * (uint32_t)cp.S;
+        int hi = a_hb[i] + (int)r;
+        int wi = a_wb[i] + (int)s;
+        bool ok = a_valid[i] && k < K && (uint32_t)hi < (uint32_t)cp.H &&
+                  (uint32_t)wi < (uint32_t)cp.W;
+        voff = ok ? (a_base[i] + ((int)r * cp.W + (int)s) * cp.C + (int)c) * 2
+                  : 0x7ffffff0;
+      }
+      __builtin_amdgcn_raw_ptr_buffer_load_lds(
+          rsrcA,
+          (__attribute__((address_space(3))) void*)(smem + A_OFF(buf) +
+                                                    (wave * 4 + i) * 1024),
+          16, voff, 0, 0, 0);
+      // B tile ([N][Kp] row-major; zero rows beyond N via OOB)
+      int kB = kb + ((sslot ^ swz) << 3);
+      int32_t voffB = b_valid[i] ? (b_voff[i] + kB) * 2 : 0x7ffffff0;
+      __builtin_amdgcn_raw_ptr_buffer_load_lds(
+          rsrcB,
+          (__attribute__((address_space(3))) void*)(smem + B_OFF(buf) +
+                                                    (wave * 4 + i) * 1024),
+          16, voffB, 0, 0, 0);
+    }
+  };
+
+  // swizzled LDS byte offset of (row, col8) in a [128][64]-bf16 tile
+  auto lds_byte = [](int row, int col8) {
+    return row * (BK * 2) + ((col8 ^ (row & 7)) << 4);
+  };
+
+  auto compute = [&](int buf) {
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 af[4], bf[4];
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) {
+        int row = wr * 64 + mi * 16 + (lane & 15);
+        int col8 = kk * 4 + (lane >> 4);
+        af[mi] = *(bf16x8*)(smem + A_OFF(buf) + lds_byte(row, col8));
+      }
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        int row = wc * 64 + ni * 16 + (lane & 15);
+        int col8 = kk * 4 + (lane >> 4);
+        bf[ni] = *(bf16x8*)(smem + B_OFF(buf) + lds_byte(row, col8));
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+  };
+
+  // ---- main loop ----
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  for (int t = 0; t < nt; ++t) {
+    if (t + 1 < nt) stage((t + 1) & 1, t + 1);
+    compute(t & 1);
+    __syncthreads();  // emits vmcnt(0) drain while glds are in flight
+  }
+
+  // ---- epilogue: stage C tile (bf16) through LDS for wide stores ----
+  // D fragment layout (16x16x32): col = lane&15, row = (lane>>4)*4 + reg.
+  __bf16* cl = (__bf16*)smem;  // [BM][BN]
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = wr * 64 + mi * 16 + (lane >> 4) * 4 + r;
+        int col = wc * 64 + ni * 16 + (lane & 15);
+        cl[row * BN + col] = (__bf16)acc[mi][ni][r];
+      }
+    }
+  __syncthreads();
+
+  // write back: 16 lanes per row, 8 bf16 (16 B) per lane
+  const int wrow = tid >> 4;          // 0..15
+  const int wcol8 = (tid & 15) * 8;   // col offset
+#pragma unroll
+  for (int p = 0; p < 8; ++p) {
+    int row_l = p * 16 + wrow;
+    int row_g = row0 + row_l;
+    int col_g = col0 + wcol8;
+    if (row_g < M && col_g + 8 <= N) {
+      bf16x8 v = *(bf16x8*)(cl + row_l * BN + wcol8);
+      if (HAS_RES) {
+        bf16x8 rv = *(const bf16x8*)(residual + (int64_t)row_g * N + col_g);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) v[e] = (__bf16)(bf2f(v[e]) + bf2f(rv[e]));
+      }
+      if (relu) {
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          v[e] = bf2f(v[e]) > 0.f ? v[e] : (__bf16)0.f;
+      }
+      *(bf16x8*)(Cout + (int64_t)row_g * N + col_g) = v;
+    }
+  }
+}
+
+// ---------------------------------------------------------------- host
+
+static uint64_t magic40(uint32_t d) {
+  if (d == 0) return 0;
+  return ((((__uint128_t)1) << 40) + d - 1) / d;
+}
+
+void launch_gemm_bf16(const void* A, const void* B, void* C,
+                      const float* bias, const void* residual, int M, int N,
+                      int K, int Kp, int64_t a_bytes, bool relu,
+                      hipStream_t stream) {
+  int tiles_m = (M + BM - 1) / BM;
+  int tiles_n = (N + BN - 1) / BN;
+  ConvParams cp{};
+  dim3 grid(tiles_m * tiles_n);
+  if (residual) {
+    hipLaunchKernelGGL((gemm_conv_kernel<0, true>), grid, dim3(NTHREADS),
+                       LDS_BYTES, stream, (const __bf16*)A, (const __bf16*)B,
+                       (__bf16*)C, bias, (const __bf16*)residual, M, N, K, Kp,
+                       (int)a_bytes, tiles_m, tiles_n, relu ? 1 : 0, cp);
+  } else {
+    hipLaunchKernelGGL((gemm_conv_kernel<0, false>), grid, dim3(NTHREADS),
+                       LDS_BYTES, stream, (const __bf16*)A, (const __bf16*)B,
+                       (__bf16*)C, bias, (const __bf16*)nullptr, M, N, K, Kp,
+                       (int)a_bytes, tiles_m, tiles_n, relu ? 1 : 0, cp);
+  }
+}
+
+void launch_conv2d_bf16(const void* X, const void* B, void* C,
+                        const float* bias, const void* residual, int Nimg,
+                        int H, int W, int Cin, int R, int S, int Cout, int sh,
+                        int sw, int pt, int pl, int HO, int WO, int Kp,
+                        bool relu, hipStream_t stream) {
+  int M = Nimg * HO * WO;
+  int K = R * S * Cin;
+  int tiles_m = (M + BM - 1) / BM;
+  int tiles_n = (Cout + BN - 1) / BN;
+  ConvParams cp;
+  cp.C = Cin; cp.H = H; cp.W = W; cp.R = R; cp.S = S;
+  cp.sh = sh; cp.sw = sw; cp.pt = pt; cp.pl = pl; cp.HO = HO; cp.WO = WO;
+  cp.magic_C = magic40(Cin);
+  cp.magic_S = magic40(S);
+  cp.magic_HW = magic40(HO * WO);
+  cp.magic_WO = magic40(WO);
+  int64_t a_bytes = (int64_t)Nimg * H * W * Cin * 2;
+  dim3 grid(tiles_m * tiles_n);
+  if (residual) {
+    hipLaunchKernelGGL((gemm_conv_kernel<1, true>), grid, dim3(NTHREADS),
+                       LDS_BYTES, stream, (const __bf16*)X, (const __bf16*)B,
+                       (__bf16*)C, bias, (const __bf16*)residual, M, Cout, K,
+                       Kp, (int)a_bytes, tiles_m, tiles_n, relu ? 1 : 0, cp);
+  } else {
+    hipLaunchKernelGGL((gemm_conv_kernel<1, false>), grid, dim3(NTHREADS),
+                       LDS_BYTES, stream, (const __bf16*)X, (const __bf16*)B,
+                       (__bf16*)C, bias, (const __bf16*)nullptr, M, Cout, K,
+                       Kp, (int)a_bytes, tiles_m, tiles_n, relu ? 1 : 0, cp);
+  }
+}

@@ -1,0 +1,42 @@
+// Host-side launch API for the CDNA4 HIP kernels (bound in bindings.cpp).
+#pragma once
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+// gemm_conv.hip — C = act(A @ B^T + bias [+ residual]), bf16 in/out, f32 acc
+void launch_gemm_bf16(const void* A, const void* B, void* C,
+                      const float* bias, const void* residual, int M, int N,
+                      int K, int Kp, int64_t a_bytes, bool relu,
+                      hipStream_t stream);
+void launch_conv2d_bf16(const void* X, const void* B, void* C,
+                        const float* bias, const void* residual, int Nimg,
+                        int H, int W, int Cin, int R, int S, int Cout, int sh,
+                        int sw, int pt, int pl, int HO, int WO, int Kp,
+                        bool relu, hipStream_t stream);
+
+// pool_reduce.hip
+void launch_maxpool_bf16(const void* X, void* Y, int Nimg, int H, int W,
+                         int C, int kh, int kw, int sh, int sw, int pt,
+                         int pl, int HO, int WO, hipStream_t stream);
+void launch_avgpool_bf16(const void* X, void* Y, int Nimg, int H, int W,
+                         int C, int kh, int kw, int sh, int sw, int pt,
+                         int pl, int HO, int WO, hipStream_t stream);
+void launch_global_mean_bf16(const void* X, void* Y, int Nimg, int HW, int C,
+                             hipStream_t stream);
+void launch_softmax_bf16(const void* X, float* Y, int M, int N,
+                         hipStream_t stream);
+
+// elementwise.hip
+void launch_cast_f32_bf16(const float* X, void* Y, int64_t n,
+                          hipStream_t stream);
+void launch_cast_bf16_f32(const void* X, float* Y, int64_t n,
+                          hipStream_t stream);
+void launch_channel_pad_bf16(const void* X, void* Y, int64_t rows, int Cin,
+                             int Cpad, hipStream_t stream);
+void launch_add_relu_bf16(const void* A, const void* B, void* Y, int64_t n,
+                          bool relu, hipStream_t stream);
+void launch_bias_act_bf16(const void* X, const float* bias, void* Y,
+                          int64_t rows, int C, int act, hipStream_t stream);
+void launch_scale_shift_bf16(const void* X, const float* scale,
+                             const float* shift, void* Y, int64_t rows, int C,
+                             bool relu, hipStream_t stream);

@@ -1,0 +1,145 @@
+// Pooling / reduction / softmax kernels, NHWC bf16 (gfx950).
+//
+// Replaces the reference TF runtime's MaxPool/AvgPool/Mean/Softmax kernels.
+// All memory-bound: vectorized bf16x8 (16 B/lane) loads per guide G13.
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <algorithm>
+#include <cstdint>
+
+#include "kernels.h"
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+
+__device__ __forceinline__ float bf2f(__bf16 v) { return (float)v; }
+
+// ------------------------------------------------------------------ pool
+
+template <bool MAX>
+__global__ void pool_kernel(const __bf16* __restrict__ X,
+                            __bf16* __restrict__ Y, int Nimg, int H, int W,
+                            int C, int kh, int kw, int sh, int sw, int pt,
+                            int pl, int HO, int WO) {
+  // one thread per (output pixel, 8-channel group)
+  int c8 = C >> 3;
+  int64_t total = (int64_t)Nimg * HO * WO * c8;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < total; idx += (int64_t)gridDim.x * blockDim.x) {
+    int cg = idx % c8;
+    int64_t p = idx / c8;
+    int wo = p % WO;
+    int64_t q = p / WO;
+    int ho = q % HO;
+    int n = q / HO;
+    int hbase = ho * sh - pt, wbase = wo * sw - pl;
+    float acc[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) acc[e] = MAX ? -3.4e38f : 0.f;
+    int cnt = 0;
+    for (int r = 0; r < kh; ++r) {
+      int hi = hbase + r;
+      if ((unsigned)hi >= (unsigned)H) continue;
+      for (int s = 0; s < kw; ++s) {
+        int wi = wbase + s;
+        if ((unsigned)wi >= (unsigned)W) continue;
+        const bf16x8 v = *(const bf16x8*)(X + ((((int64_t)n * H + hi) * W + wi) * C) + cg * 8);
+        ++cnt;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          float f = bf2f(v[e]);
+          acc[e] = MAX ? fmaxf(acc[e], f) : acc[e] + f;
+        }
+      }
+    }
+    bf16x8 o;
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      o[e] = (__bf16)(MAX ? acc[e] : acc[e] / (float)max(cnt, 1));
+    *(bf16x8*)(Y + ((((int64_t)n * HO + ho) * WO + wo) * C) + cg * 8) = o;
+  }
+}
+
+void launch_maxpool_bf16(const void* X, void* Y, int Nimg, int H, int W,
+                         int C, int kh, int kw, int sh, int sw, int pt,
+                         int pl, int HO, int WO, hipStream_t stream) {
+  int64_t total = (int64_t)Nimg * HO * WO * (C >> 3);
+  int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
+  hipLaunchKernelGGL((pool_kernel<true>), dim3(blocks), dim3(256), 0, stream,
+                     (const __bf16*)X, (__bf16*)Y, Nimg, H, W, C, kh, kw, sh,
+                     sw, pt, pl, HO, WO);
+}
+
+void launch_avgpool_bf16(const void* X, void* Y, int Nimg, int H, int W,
+                         int C, int kh, int kw, int sh, int sw, int pt,
+                         int pl, int HO, int WO, hipStream_t stream) {
+  int64_t total = (int64_t)Nimg * HO * WO * (C >> 3);
+  int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
+  hipLaunchKernelGGL((pool_kernel<false>), dim3(blocks), dim3(256), 0, stream,
+                     (const __bf16*)X, (__bf16*)Y, Nimg, H, W, C, kh, kw, sh,
+                     sw, pt, pl, HO, WO);
+}
+
+// ---------------------------------------------------------- global mean
+
+__global__ void global_mean_kernel(const __bf16* __restrict__ X,
+                                   __bf16* __restrict__ Y, int Nimg, int HW,
+                                   int C) {
+  // thread per (n, c8 group); HW is small (49/64), loop it
+  int c8 = C >> 3;
+  int64_t total = (int64_t)Nimg * c8;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < total; idx += (int64_t)gridDim.x * blockDim.x) {
+    int cg = idx % c8;
+    int n = idx / c8;
+    float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    const __bf16* base = X + (int64_t)n * HW * C + cg * 8;
+    for (int p = 0; p < HW; ++p) {
+      bf16x8 v = *(const bf16x8*)(base + (int64_t)p * C);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) acc[e] += bf2f(v[e]);
+    }
+    bf16x8 o;
+    float inv = 1.f / (float)HW;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) o[e] = (__bf16)(acc[e] * inv);
+    *(bf16x8*)(Y + (int64_t)n * C + cg * 8) = o;
+  }
+}
+
+void launch_global_mean_bf16(const void* X, void* Y, int Nimg, int HW, int C,
+                             hipStream_t stream) {
+  int64_t total = (int64_t)Nimg * (C >> 3);
+  int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
+  hipLaunchKernelGGL(global_mean_kernel, dim3(blocks), dim3(256), 0, stream,
+                     (const __bf16*)X, (__bf16*)Y, Nimg, HW, C);
+}
+
+// --------------------------------------------------------------- softmax
+
+__global__ void softmax_kernel(const __bf16* __restrict__ X,
+                               float* __restrict__ Y, int M, int N) {
+  // one wave per row (classes ~1e3); wave-wide shuffle reduce
+  int row = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  int lane = threadIdx.x & 63;
+  if (row >= M) return;
+  const __bf16* x = X + (int64_t)row * N;
+  float m = -3.4e38f;
+  for (int i = lane; i < N; i += 64) m = fmaxf(m, bf2f(x[i]));
+#pragma unroll
+  for (int off = 32; off; off >>= 1) m = fmaxf(m, __shfl_xor(m, off, 64));
+  float s = 0.f;
+  for (int i = lane; i < N; i += 64) s += __expf(bf2f(x[i]) - m);
+#pragma unroll
+  for (int off = 32; off; off >>= 1) s += __shfl_xor(s, off, 64);
+  float inv = 1.f / s;
+  float* y = Y + (int64_t)row * N;
+  for (int i = lane; i < N; i += 64) y[i] = __expf(bf2f(x[i]) - m) * inv;
+}
+
+void launch_softmax_bf16(const void* X, float* Y, int M, int N,
+                         hipStream_t stream) {
+  int wpb = 4;  // 4 waves (rows) per block
+  int blocks = (M + wpb - 1) / wpb;
+  hipLaunchKernelGGL(softmax_kernel, dim3(blocks), dim3(wpb * 64), 0, stream,
+                     (const __bf16*)X, Y, M, N);
+}

@@ -1,0 +1,345 @@
+"""MI355X GPU executor: compiles the IR into a static launch plan over the
+hand-written CDNA4 HIP kernels, optionally captured into a HIP graph.
+
+Replaces `tf.Session.run`'s C++ executor (reference dependency) with:
+  - compile(batch): shape-infer every node, preallocate bf16 NHWC
+    activation buffers, prepack weights (HWIO -> [O][Kp] bf16, BN already
+    folded by the IR passes), build a flat list of launch closures.
+  - execute: copy input, replay the plan (HIP-graph capture per batch
+    size removes per-kernel Python/launch overhead — the serving hot
+    loop, cf. SURVEY §3c "amortize launch overhead via pre-captured HIP
+    graphs per batch size").
+
+All hot ops (conv/dense/pool/softmax/elementwise) run on _hipops kernels;
+only cold glue (concat, reshape bookkeeping) touches torch ops.
+"""
+from __future__ import annotations
+
+import math
+from typing import Callable, Dict, List, Optional
+
+import numpy as np
+import torch
+
+from ..ops.native import ext, require_native_on_gpu
+from .ir import Graph, Node
+
+ACT_NONE, ACT_RELU, ACT_RELU6, ACT_TANH, ACT_SIGMOID, ACT_GELU = range(6)
+
+
+def _ceil_div(a, b):
+    return (a + b - 1) // b
+
+
+def _same_pads(h, w, kh, kw, sh, sw):
+    oh, ow = _ceil_div(h, sh), _ceil_div(w, sw)
+    ph = max((oh - 1) * sh + kh - h, 0)
+    pw = max((ow - 1) * sw + kw - w, 0)
+    return oh, ow, ph // 2, pw // 2
+
+
+def _valid_out(h, w, kh, kw, sh, sw):
+    return (h - kh) // sh + 1, (w - kw) // sw + 1
+
+
+def _pad8(n):
+    return _ceil_div(n, 8) * 8
+
+
+def _pad64(n):
+    return _ceil_div(n, 64) * 64
+
+
+class GpuExecutor:
+    def __init__(self, graph: Graph, device: str = "cuda", cfg=None):
+        require_native_on_gpu()
+        self.graph = graph
+        self.device = torch.device(device)
+        self.cfg = cfg
+        self.use_graphs = bool(getattr(cfg, "use_hip_graphs", True))
+        self._weights: Dict[str, tuple] = {}
+        self._plans: Dict[int, "_Plan"] = {}
+        self.consts = graph.consts
+
+    # ------------------------------------------------ weight prepacking
+
+    def _prep_conv_weight(self, wname: str):
+        if wname in self._weights:
+            return self._weights[wname]
+        w = self.consts[wname].astype(np.float32)  # HWIO (BN pre-folded)
+        R, S, C, O = w.shape
+        Cp = _pad8(C)
+        if Cp != C:
+            wp = np.zeros((R, S, Cp, O), np.float32)
+            wp[:, :, :C] = w
+            w = wp
+        K = R * S * Cp
+        Kp = _pad64(K)
+        wt = np.ascontiguousarray(w.transpose(3, 0, 1, 2).reshape(O, K))
+        if Kp != K:
+            wt = np.concatenate([wt, np.zeros((O, Kp - K), np.float32)], 1)
+        t = torch.from_numpy(wt).to(self.device, torch.bfloat16).contiguous()
+        self._weights[wname] = (t, R, S, C, Cp, O, Kp)
+        return self._weights[wname]
+
+    def _prep_dense_weight(self, wname: str, transpose_b: bool):
+        key = wname + ("/T" if transpose_b else "")
+        if key in self._weights:
+            return self._weights[key]
+        w = self.consts[wname].astype(np.float32)
+        if transpose_b:
+            w = w.T  # stored [N][K] already
+        K, N = w.shape
+        Kp, Np = _pad64(K), _pad8(N)
+        wt = np.zeros((Np, Kp), np.float32)
+        wt[:N, :K] = np.ascontiguousarray(w.T)
+        t = torch.from_numpy(wt).to(self.device, torch.bfloat16).contiguous()
+        self._weights[key] = (t, K, N, Kp, Np)
+        return self._weights[key]
+
+    def _bias(self, bname: Optional[str], n: int):
+        if not bname:
+            return None
+        key = bname + "/f32"
+        if key not in self._weights:
+            b = self.consts[bname].astype(np.float32).reshape(-1)
+            self._weights[key] = torch.from_numpy(b).to(self.device)
+        return self._weights[key]
+
+    # ------------------------------------------------------- compile
+
+    def _compile(self, batch: int) -> "_Plan":
+        g = self.graph
+        dev = self.device
+        E = ext()
+        steps: List[Callable[[], None]] = []
+        bufs: Dict[str, torch.Tensor] = {}
+        shapes: Dict[str, tuple] = {}
+
+        inp = g.inputs[0]
+        in_shape = None
+        if self.cfg is not None:
+            in_shape = (batch, self.cfg.input_height, self.cfg.input_width,
+                        self.cfg.input_channels)
+        else:
+            raise ValueError("GpuExecutor needs cfg for the input shape")
+        x_f32 = torch.zeros(in_shape, dtype=torch.float32, device=dev)
+        x_bf = torch.zeros(in_shape, dtype=torch.bfloat16, device=dev)
+        steps.append(lambda: E.cast_f32_bf16(x_f32, x_bf))
+        bufs[inp] = x_bf
+        shapes[inp] = in_shape
+
+        def alloc(name, shape, dtype=torch.bfloat16):
+            t = torch.zeros(shape, dtype=dtype, device=dev)
+            bufs[name] = t
+            shapes[name] = tuple(shape)
+            return t
+
+        for node in g.nodes:
+            op = node.op
+            if op in ("ConvBlock",):
+                xn = node.inputs[0]
+                X = bufs[xn]
+                N_, H, W, Cin = shapes[xn]
+                wt, R, S, C, Cp, O, Kp = self._prep_conv_weight(node.inputs[1])
+                strides = node.attrs["strides"]
+                sh, sw = int(strides[1]), int(strides[2])
+                pad = node.attrs.get("padding", "VALID")
+                if pad == "SAME":
+                    HO, WO, pt, pl = _same_pads(H, W, R, S, sh, sw)
+                else:
+                    HO, WO = _valid_out(H, W, R, S, sh, sw)
+                    pt = pl = 0
+                Y = alloc(node.name, (N_, HO, WO, O))
+                bias = self._bias(node.attrs.get("bias"), O)
+                res = bufs[node.attrs["residual"]] if node.attrs.get("residual") else None
+                relu = bool(node.attrs.get("relu"))
+                if Cp != Cin:
+                    Xp = torch.zeros((N_, H, W, Cp), dtype=torch.bfloat16,
+                                     device=dev)
+                    steps.append(lambda X=X, Xp=Xp: E.channel_pad_bf16(X, Xp))
+                    X = Xp
+                if R == 1 and S == 1 and sh == 1 and sw == 1:
+                    M = N_ * HO * WO
+                    A2 = X.view(M, Cp)
+                    Y2 = Y.view(M, O)
+                    steps.append(lambda A2=A2, wt=wt, Y2=Y2, bias=bias,
+                                 res=(res.view(M, O) if res is not None else None),
+                                 K=Cp, relu=relu:
+                                 E.gemm_bf16(A2, wt, Y2, bias, res, K, relu))
+                else:
+                    steps.append(lambda X=X, wt=wt, Y=Y, bias=bias, res=res,
+                                 R=R, S=S, sh=sh, sw=sw, pt=pt, pl=pl,
+                                 relu=relu:
+                                 E.conv2d_bf16(X, wt, Y, bias, res, R, S,
+                                               sh, sw, pt, pl, relu))
+            elif op == "Dense":
+                xn = node.inputs[0]
+                X = bufs[xn]
+                wt, K, N, Kp, Np = self._prep_dense_weight(
+                    node.inputs[1], bool(node.attrs.get("transpose_b")))
+                M = int(np.prod(shapes[xn][:-1]))
+                assert shapes[xn][-1] == K, f"dense K mismatch {node.name}"
+                bias = self._bias(node.attrs.get("bias"), N)
+                if bias is not None and Np != N:
+                    bias = torch.cat([bias, torch.full((Np - N,), -1e30,
+                                                       device=dev)])
+                relu = bool(node.attrs.get("relu"))
+                Yp = alloc(node.name + "/padded", (M, Np))
+                A2 = X.view(M, K)
+                steps.append(lambda A2=A2, wt=wt, Yp=Yp, bias=bias, K=K,
+                             relu=relu:
+                             E.gemm_bf16(A2, wt, Yp, bias, None, K, relu))
+                if Np != N:
+                    Y = alloc(node.name, (M, N))
+                    steps.append(lambda Yp=Yp, Y=Y, N=N: Y.copy_(Yp[:, :N]))
+                else:
+                    bufs[node.name] = Yp
+                    shapes[node.name] = (M, N)
+            elif op == "MaxPool" or op == "AvgPool":
+                xn = node.inputs[0]
+                X = bufs[xn]
+                N_, H, W, C = shapes[xn]
+                ks, st = node.attrs["ksize"], node.attrs["strides"]
+                kh, kw = int(ks[1]), int(ks[2])
+                sh, sw = int(st[1]), int(st[2])
+                if node.attrs.get("padding", "VALID") == "SAME":
+                    HO, WO, pt, pl = _same_pads(H, W, kh, kw, sh, sw)
+                else:
+                    HO, WO = _valid_out(H, W, kh, kw, sh, sw)
+                    pt = pl = 0
+                Y = alloc(node.name, (N_, HO, WO, C))
+                fn = E.maxpool_bf16 if op == "MaxPool" else E.avgpool_bf16
+                steps.append(lambda fn=fn, X=X, Y=Y, kh=kh, kw=kw, sh=sh,
+                             sw=sw, pt=pt, pl=pl:
+                             fn(X, Y, kh, kw, sh, sw, pt, pl))
+            elif op == "Mean":
+                xn = node.inputs[0]
+                X = bufs[xn]
+                N_, H, W, C = shapes[xn]
+                axes = self.consts[node.inputs[1]].reshape(-1).tolist()
+                assert sorted(int(a) for a in axes) == [1, 2], \
+                    f"Mean axes {axes} unsupported on GPU path"
+                keep = bool(node.attrs.get("keep_dims", False))
+                shape = (N_, 1, 1, C) if keep else (N_, C)
+                Y = alloc(node.name, shape)
+                Y2 = Y.view(N_, C)
+                steps.append(lambda X=X, Y2=Y2: E.global_mean_bf16(X, Y2))
+            elif op == "Softmax":
+                xn = node.inputs[0]
+                X = bufs[xn]
+                M, N = shapes[xn]
+                Y = alloc(node.name, (M, N), torch.float32)
+                steps.append(lambda X=X, Y=Y: E.softmax_bf16(X, Y))
+            elif op == "Reshape":
+                xn = node.inputs[0]
+                tgt = [int(v) for v in
+                       self.consts[node.inputs[1]].reshape(-1).tolist()]
+                numel = int(np.prod(shapes[xn]))
+                if -1 in tgt:
+                    known = int(np.prod([t for t in tgt if t != -1]))
+                    tgt = [numel // known if t == -1 else t for t in tgt]
+                bufs[node.name] = bufs[xn].view(tgt)
+                shapes[node.name] = tuple(tgt)
+            elif op == "Squeeze":
+                xn = node.inputs[0]
+                dims = node.attrs.get("squeeze_dims") or node.attrs.get("axis")
+                t = bufs[xn]
+                for d in sorted((int(d) for d in dims or []), reverse=True):
+                    t = t.squeeze(d)
+                if not dims:
+                    t = t.squeeze()
+                bufs[node.name] = t
+                shapes[node.name] = tuple(t.shape)
+            elif op in ("Add", "AddV2"):
+                a, b = node.inputs
+                Y = alloc(node.name, shapes[a])
+                steps.append(lambda A=bufs[a], B=bufs[b], Y=Y:
+                             E.add_relu_bf16(A, B, Y, False))
+            elif op == "Relu":
+                xn = node.inputs[0]
+                Y = alloc(node.name, shapes[xn])
+                steps.append(lambda X=bufs[xn], Y=Y:
+                             E.bias_act_bf16(X, None, Y, ACT_RELU))
+            elif op == "BiasAdd":
+                xn = node.inputs[0]
+                C = shapes[xn][-1]
+                bias = self._bias(node.inputs[1], C)
+                Y = alloc(node.name, shapes[xn])
+                steps.append(lambda X=bufs[xn], Y=Y, bias=bias:
+                             E.bias_act_bf16(X, bias, Y, ACT_NONE))
+            elif op == "BnScaleBias":
+                xn = node.inputs[0]
+                sc = self._bias(node.inputs[1], 0)
+                sh_ = self._bias(node.inputs[2], 0)
+                Y = alloc(node.name, shapes[xn])
+                steps.append(lambda X=bufs[xn], Y=Y, sc=sc, sh_=sh_:
+                             E.scale_shift_bf16(X, sc, sh_, Y, False))
+            elif op == "ConcatV2":
+                ins = node.inputs[:-1]
+                axis = int(self.consts[node.inputs[-1]].reshape(-1)[0])
+                if axis < 0:
+                    axis += len(shapes[ins[0]])
+                out_shape = list(shapes[ins[0]])
+                out_shape[axis] = sum(shapes[i][axis] for i in ins)
+                Y = alloc(node.name, out_shape)
+                srcs = [bufs[i] for i in ins]
+                steps.append(lambda Y=Y, srcs=srcs, axis=axis:
+                             torch.cat(srcs, dim=axis, out=Y))
+            else:
+                raise NotImplementedError(
+                    f"GPU path: op {op} (node {node.name})")
+
+        outs = [bufs[o] for o in g.outputs]
+        return _Plan(x_f32, steps, outs, self.use_graphs)
+
+    @torch.inference_mode()
+    def run(self, feeds: Dict[str, torch.Tensor],
+            outputs: Optional[List[str]] = None) -> List[torch.Tensor]:
+        x = feeds[self.graph.inputs[0]]
+        batch = x.shape[0]
+        plan = self._plans.get(batch)
+        if plan is None:
+            plan = self._compile(batch)
+            self._plans[batch] = plan
+        return plan.execute(x.to(self.device, torch.float32))
+
+
+class _Plan:
+    def __init__(self, x_f32, steps, outs, use_graphs):
+        self.x_f32 = x_f32
+        self.steps = steps
+        self.outs = outs
+        self.use_graphs = use_graphs
+        self.graph = None
+
+    def _maybe_capture(self):
+        if not self.use_graphs or self.graph is not None:
+            return
+        # warm-up run already happened; capture the launch sequence
+        g = torch.cuda.CUDAGraph()
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for fn in self.steps:
+                fn()
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        with torch.cuda.graph(g):
+            for fn in self.steps:
+                fn()
+        self.graph = g
+
+    def execute(self, x: torch.Tensor):
+        self.x_f32.copy_(x)
+        if self.use_graphs:
+            if self.graph is None:
+                for fn in self.steps:  # warm-up + correctness path
+                    fn()
+                self._maybe_capture()
+                return self.outs
+            self.graph.replay()
+        else:
+            for fn in self.steps:
+                fn()
+        return self.outs

@@ -1,0 +1,51 @@
+"""End-to-end GPU tests: full models through the HIP-kernel executor vs
+the CPU fp32 reference executor (gpu-marked)."""
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def _engines(model, h, w, c, nc):
+    from tensorflow_web_deploy_amd.conf.config import Config
+    from tensorflow_web_deploy_amd.engine.service import LocalEngine
+    cfg = Config(model_name=model, input_height=h, input_width=w,
+                 input_channels=c, num_classes=nc)
+    gpu = LocalEngine(cfg.replace(device="cuda"))
+    cpu = LocalEngine(cfg.replace(device="cpu"), graph=gpu.graph)
+    return gpu, cpu
+
+
+def test_lenet_gpu_vs_cpu():
+    gpu, cpu = _engines("lenet", 28, 28, 1, 10)
+    x = np.random.rand(8, 28, 28, 1).astype(np.float32)
+    pg = gpu.infer(x)
+    pc = cpu.infer(x)
+    assert pg.shape == pc.shape == (8, 10)
+    # same argmax on almost all rows; prob close
+    agree = (pg.argmax(1) == pc.argmax(1)).mean()
+    assert agree >= 0.9, f"argmax agreement {agree}"
+    np.testing.assert_allclose(pg.sum(1), 1.0, atol=1e-3)
+    np.testing.assert_allclose(pg, pc, atol=0.05)
+
+
+def test_resnet50_gpu_vs_cpu():
+    gpu, cpu = _engines("resnet50", 224, 224, 3, 1000)
+    x = (np.random.rand(2, 224, 224, 3).astype(np.float32) - 0.5)
+    pg = gpu.infer(x)
+    pc = cpu.infer(x)
+    assert pg.shape == (2, 1000)
+    np.testing.assert_allclose(pg.sum(1), 1.0, atol=1e-3)
+    # bf16 through 53 conv layers: compare top-1 and prob mass
+    assert (pg.argmax(1) == pc.argmax(1)).all(), "top-1 mismatch vs fp32"
+    assert np.abs(pg - pc).max() < 0.15
+
+
+def test_resnet50_batch_consistency():
+    """Same image in different batch slots gives the same probs."""
+    gpu, _ = _engines("resnet50", 224, 224, 3, 1000)
+    x = (np.random.rand(1, 224, 224, 3).astype(np.float32) - 0.5)
+    xb = np.repeat(x, 4, axis=0)
+    p = gpu.infer(xb)
+    np.testing.assert_allclose(p[0], p[3], atol=1e-3)

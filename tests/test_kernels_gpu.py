@@ -1,0 +1,193 @@
+"""HIP kernel numerics vs plain PyTorch fp32 references (gpu-marked).
+
+Every kernel is compared against a torch fp32 computation of the same op
+on random data; tolerances account for bf16 inputs/outputs (f32 accum).
+B operands are asymmetric random (guide §5.4 rule 16: transpose-detecting).
+"""
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def E():
+    from tensorflow_web_deploy_amd.ops.native import ext
+    return ext()
+
+
+def _bf(x):
+    return x.to("cuda", torch.bfloat16).contiguous()
+
+
+def _pad64_cols(w):
+    n, k = w.shape
+    kp = (k + 63) // 64 * 64
+    if kp != k:
+        w = torch.cat([w, torch.zeros(n, kp - k, dtype=w.dtype,
+                                      device=w.device)], 1)
+    return w
+
+
+@pytest.mark.parametrize("M,N,K", [(128, 128, 64), (256, 512, 256),
+                                   (200, 1000, 2048), (64, 64, 72),
+                                   (1, 16, 64), (3136, 2048, 512)])
+def test_gemm_bf16(E, M, N, K):
+    torch.manual_seed(0)
+    A = torch.randn(M, K) * 0.5
+    B = torch.randn(N, K) * 0.5
+    Ag, Bg = _bf(A), _pad64_cols(_bf(B))
+    C = torch.zeros(M, N, dtype=torch.bfloat16, device="cuda")
+    E.gemm_bf16(Ag, Bg, C, None, None, K, False)
+    ref = (Ag.float() @ Bg.float()[:, :K].t())
+    got = C.float()
+    torch.testing.assert_close(got, ref, rtol=2e-2, atol=2e-2 * K ** 0.5)
+
+
+def test_gemm_bias_relu_residual(E):
+    torch.manual_seed(1)
+    M, N, K = 256, 128, 128
+    A = torch.randn(M, K)
+    B = torch.randn(N, K)
+    bias = torch.randn(N)
+    res = torch.randn(M, N)
+    Ag, Bg = _bf(A), _bf(B)
+    biasg = bias.cuda()
+    resg = _bf(res)
+    C = torch.zeros(M, N, dtype=torch.bfloat16, device="cuda")
+    E.gemm_bf16(Ag, Bg, C, biasg, resg, K, True)
+    ref = torch.relu(Ag.float() @ Bg.float().t() + biasg + resg.float())
+    torch.testing.assert_close(C.float(), ref, rtol=2e-2, atol=0.3)
+
+
+def _tf_conv_ref(X, W, stride, pad_same):
+    # X NHWC f32, W HWIO f32 -> NHWC f32 (same math as engine/executor.py)
+    import torch.nn.functional as F
+    xc = X.permute(0, 3, 1, 2)
+    wc = W.permute(3, 2, 0, 1)
+    kh, kw = W.shape[0], W.shape[1]
+    if pad_same:
+        from tensorflow_web_deploy_amd.engine.executor import tf_same_pad
+        xc = tf_same_pad(xc, kh, kw, stride, stride)
+    y = F.conv2d(xc, wc, stride=stride)
+    return y.permute(0, 2, 3, 1).contiguous()
+
+
+@pytest.mark.parametrize("N,H,W,C,R,O,stride,same", [
+    (2, 16, 16, 64, 3, 64, 1, True),       # resnet 3x3 s1
+    (2, 16, 16, 64, 3, 128, 2, True),      # resnet 3x3 s2
+    (2, 14, 14, 256, 1, 64, 1, True),      # 1x1 s1 (gemm path)
+    (2, 28, 28, 128, 1, 256, 2, True),     # 1x1 s2 shortcut
+    (1, 30, 30, 32, 5, 48, 1, False),      # inception 5x5 VALID
+    (2, 17, 17, 8, 7, 64, 2, True),        # 7x7 s2 stem-like (Cpad=8)
+])
+def test_conv2d_bf16(E, N, H, W, C, R, O, stride, same):
+    torch.manual_seed(2)
+    X = torch.randn(N, H, W, C) * 0.5
+    Wt = torch.randn(R, R, C, O) * (2.0 / (R * R * C)) ** 0.5
+    ref = _tf_conv_ref(_bf(X).float().cpu(), Wt, stride, same)
+
+    Xg = _bf(X)
+    K = R * R * C
+    Kp = (K + 63) // 64 * 64
+    W2 = torch.zeros(O, Kp)
+    W2[:, :K] = Wt.permute(3, 0, 1, 2).reshape(O, K)
+    W2g = _bf(W2)
+    if same:
+        from tensorflow_web_deploy_amd.engine.gpu_executor import _same_pads
+        HO, WO, pt, pl = _same_pads(H, W, R, R, stride, stride)
+    else:
+        HO, WO = (H - R) // stride + 1, (W - R) // stride + 1
+        pt = pl = 0
+    Y = torch.zeros(N, HO, WO, O, dtype=torch.bfloat16, device="cuda")
+    if R == 1 and stride == 1:
+        E.gemm_bf16(Xg.view(-1, C), W2g, Y.view(-1, O), None, None, K, False)
+    else:
+        E.conv2d_bf16(Xg, W2g, Y, None, None, R, R, stride, stride, pt, pl,
+                      False)
+    torch.testing.assert_close(Y.float().cpu(), ref, rtol=3e-2,
+                               atol=2e-2 * K ** 0.5)
+
+
+def test_maxpool_avgpool(E):
+    torch.manual_seed(3)
+    X = torch.randn(2, 15, 15, 64)
+    Xg = _bf(X)
+    from tensorflow_web_deploy_amd.engine.ir import Node
+    from tensorflow_web_deploy_amd.engine.executor import _CPU_OPS
+    node = Node("p", "MaxPool", ["x"],
+                {"ksize": [1, 3, 3, 1], "strides": [1, 2, 2, 1],
+                 "padding": "SAME"})
+    ref = _CPU_OPS["MaxPool"](node, [Xg.float().cpu()], {})
+    from tensorflow_web_deploy_amd.engine.gpu_executor import _same_pads
+    HO, WO, pt, pl = _same_pads(15, 15, 3, 3, 2, 2)
+    Y = torch.zeros(2, HO, WO, 64, dtype=torch.bfloat16, device="cuda")
+    E.maxpool_bf16(Xg, Y, 3, 3, 2, 2, pt, pl)
+    torch.testing.assert_close(Y.float().cpu(), ref, rtol=1e-2, atol=1e-2)
+
+    node2 = Node("p", "AvgPool", ["x"],
+                 {"ksize": [1, 3, 3, 1], "strides": [1, 2, 2, 1],
+                  "padding": "SAME"})
+    ref2 = _CPU_OPS["AvgPool"](node2, [Xg.float().cpu()], {})
+    Y2 = torch.zeros(2, HO, WO, 64, dtype=torch.bfloat16, device="cuda")
+    E.avgpool_bf16(Xg, Y2, 3, 3, 2, 2, pt, pl)
+    torch.testing.assert_close(Y2.float().cpu(), ref2, rtol=1e-2, atol=1e-2)
+
+
+def test_global_mean(E):
+    X = torch.randn(4, 7, 7, 2048)
+    Xg = _bf(X)
+    Y = torch.zeros(4, 2048, dtype=torch.bfloat16, device="cuda")
+    E.global_mean_bf16(Xg, Y)
+    ref = Xg.float().mean(dim=(1, 2))
+    torch.testing.assert_close(Y.float(), ref, rtol=1e-2, atol=1e-2)
+
+
+def test_softmax(E):
+    X = torch.randn(64, 1000) * 3
+    Xg = _bf(X)
+    Y = torch.zeros(64, 1000, dtype=torch.float32, device="cuda")
+    E.softmax_bf16(Xg, Y)
+    ref = torch.softmax(Xg.float(), dim=-1)
+    torch.testing.assert_close(Y, ref, rtol=1e-3, atol=1e-4)
+    torch.testing.assert_close(Y.sum(-1), torch.ones(64, device="cuda"),
+                               rtol=1e-4, atol=1e-4)
+
+
+def test_elementwise(E):
+    X = torch.randn(3, 8, 8, 64)
+    Xg = _bf(X)
+    # cast roundtrip
+    Xf = torch.zeros_like(Xg, dtype=torch.float32)
+    E.cast_bf16_f32(Xg, Xf)
+    torch.testing.assert_close(Xf, Xg.float())
+    Xb2 = torch.zeros_like(Xg)
+    E.cast_f32_bf16(Xf, Xb2)
+    torch.testing.assert_close(Xb2.float(), Xg.float())
+    # add + relu
+    B = _bf(torch.randn(3, 8, 8, 64))
+    Y = torch.zeros_like(Xg)
+    E.add_relu_bf16(Xg, B, Y, True)
+    torch.testing.assert_close(Y.float(), torch.relu(Xg.float() + B.float()),
+                               rtol=1e-2, atol=1e-2)
+    # bias + gelu
+    bias = torch.randn(64, device="cuda")
+    Y2 = torch.zeros_like(Xg)
+    E.bias_act_bf16(Xg, bias, Y2, 5)
+    import torch.nn.functional as F
+    torch.testing.assert_close(Y2.float(), F.gelu(Xg.float() + bias),
+                               rtol=2e-2, atol=2e-2)
+    # channel pad
+    Xs = _bf(torch.randn(4, 5, 5, 3))
+    Yp = torch.zeros(4, 5, 5, 8, dtype=torch.bfloat16, device="cuda")
+    E.channel_pad_bf16(Xs, Yp)
+    torch.testing.assert_close(Yp[..., :3].float(), Xs.float())
+    assert Yp[..., 3:].abs().sum().item() == 0
+    # scale/shift (unfused BN)
+    sc = torch.rand(64, device="cuda") + 0.5
+    shf = torch.randn(64, device="cuda")
+    Y3 = torch.zeros_like(Xg)
+    E.scale_shift_bf16(Xg, sc, shf, Y3, False)
+    torch.testing.assert_close(Y3.float(), Xg.float() * sc + shf,
+                               rtol=1e-2, atol=1e-2)
