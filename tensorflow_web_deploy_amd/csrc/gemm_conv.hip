@@ -283,6 +283,247 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
   }
 }
 
+// ================================================================
+// 256x256 pipelined kernel (8 waves, BK=64, counted vmcnt, raw barriers)
+// for large layers — cdna_hip_programming.md §5 8-phase-template class.
+// LDS: [2 buf][A0,A1,B0,B1 halves][128][64] bf16 = 128 KiB; 1 block/CU.
+// Loads for K-tile t+1 stay in flight across the K-loop barriers
+// (vmcnt(2) at each iteration head, never 0 in steady state).
+// ================================================================
+
+#define BM2 256
+#define BN2 256
+#define NTHREADS2 512
+#define LDS2_BYTES (8 * 128 * BK * 2)  // 128 KiB
+// half-tile h (0=A0,1=A1,2=B0,3=B1) of buffer buf
+#define HALF_OFF(buf, h) (((buf) * 4 + (h)) * (128 * BK * 2))
+
+template <int MODE, bool HAS_RES>
+__global__ __launch_bounds__(NTHREADS2, 2)
+void gemm_conv_kernel_256(const __bf16* __restrict__ A,
+                          const __bf16* __restrict__ B,
+                          __bf16* __restrict__ Cout,
+                          const float* __restrict__ bias,
+                          const __bf16* __restrict__ residual,
+                          int M, int N, int K, int Kp,
+                          int a_bytes, int tiles_m, int tiles_n,
+                          int relu, ConvParams cp) {
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = __builtin_amdgcn_readfirstlane(tid >> 6);
+
+  int nwg = tiles_m * tiles_n;
+  int bid = blockIdx.x;
+  {
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = bid & 7, off = bid >> 3;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
+  }
+  const int tn = bid / tiles_m;
+  const int tm = bid - tn * tiles_m;
+  const int row0 = tm * BM2;
+  const int col0 = tn * BN2;
+
+  const auto rsrcA = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)A, 0, a_bytes, 0x00020000);
+  const auto rsrcB = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)B, 0, (int)((int64_t)N * Kp * 2), 0x00020000);
+
+  // staging: half-tile = 128 rows x 64 cols = 16 KiB = 2 glds per wave.
+  // glds i of half h covers rows (wave*2+i)*8 + lane/8 of that half.
+  const int srow_base = lane >> 3;
+  const int sslot = lane & 7;
+  const int swz = srow_base & 7;
+  // A halves: combo (h*2+i), rows h*128 + (wave*2+i)*8 + srow_base
+  int32_t a_base[4], a_hb[4], a_wb[4];
+  bool a_valid[4];
+  int32_t b_voff[4];
+  bool b_valid[4];
+#pragma unroll
+  for (int c2 = 0; c2 < 4; ++c2) {
+    int h = c2 >> 1, i = c2 & 1;
+    int row_l = h * 128 + (wave * 2 + i) * 8 + srow_base;
+    int row_g = row0 + row_l;
+    if (MODE == 0) {
+      a_valid[c2] = row_g < M;
+      a_base[c2] = row_g * K;
+    } else {
+      a_valid[c2] = row_g < M;
+      uint32_t rg = (uint32_t)min(row_g, max(M - 1, 0));
+      uint32_t n_img = mdiv(rg, cp.magic_HW);
+      uint32_t rem = rg - n_img * (uint32_t)(cp.HO * cp.WO);
+      uint32_t ho = mdiv(rem, cp.magic_WO);
+      uint32_t wo = rem - ho * (uint32_t)cp.WO;
+      int hb = (int)ho * cp.sh - cp.pt;
+      int wb = (int)wo * cp.sw - cp.pl;
+      a_hb[c2] = hb;
+      a_wb[c2] = wb;
+      a_base[c2] = (((int)n_img * cp.H + hb) * cp.W + wb) * cp.C;
+    }
+    int col_g = col0 + row_l;
+    b_valid[c2] = col_g < N;
+    b_voff[c2] = col_g * Kp;
+  }
+
+  const int wr = wave >> 2;  // 0..1: 128-row strip
+  const int wc = wave & 3;   // 0..3: 64-col strip
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      float b0 = 0.f;
+      if (bias) {
+        int col = col0 + wc * 64 + ni * 16 + (lane & 15);
+        if (col < N) b0 = bias[col];
+      }
+      acc[mi][ni] = {b0, b0, b0, b0};
+    }
+
+  const int nt = (K + BK - 1) / BK;
+
+  auto a_voffset = [&](int c2, int k) -> int32_t {
+    if (MODE == 0) {
+      return (a_valid[c2] && k < K) ? (a_base[c2] + k) * 2 : 0x7ffffff0;
+    }
+    uint32_t rs = mdiv((uint32_t)k, cp.magic_C);
+    uint32_t c = (uint32_t)k - rs * (uint32_t)cp.C;
+    uint32_t r = mdiv(rs, cp.magic_S);
+    uint32_t s = rs - r * (uint32_t)cp.S;
+    int hi = a_hb[c2] + (int)r;
+    int wi = a_wb[c2] + (int)s;
+    bool ok = a_valid[c2] && k < K && (uint32_t)hi < (uint32_t)cp.H &&
+              (uint32_t)wi < (uint32_t)cp.W;
+    return ok ? (a_base[c2] + ((int)r * cp.W + (int)s) * cp.C + (int)c) * 2
+              : 0x7ffffff0;
+  };
+
+  // stage one half-tile (2 glds per wave) of K-tile t into buffer t&1
+  auto stage_half = [&](int h, int t) {
+    const int kb = t * BK;
+    const int buf = t & 1;
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int32_t voff;
+      int k = kb + ((sslot ^ swz) << 3);
+      if (h < 2) {
+        voff = a_voffset(h * 2 + i, k);
+      } else {
+        int c2 = (h - 2) * 2 + i;
+        voff = b_valid[c2] ? (b_voff[c2] + k) * 2 : 0x7ffffff0;
+      }
+      __builtin_amdgcn_raw_ptr_buffer_load_lds(
+          h < 2 ? rsrcA : rsrcB,
+          (__attribute__((address_space(3))) void*)(
+              smem + HALF_OFF(buf, h) + (wave * 2 + i) * 1024),
+          16, voff, 0, 0, 0);
+    }
+  };
+
+  auto lds_byte2 = [](int row, int col8) {
+    return row * (BK * 2) + ((col8 ^ (row & 7)) << 4);
+  };
+
+  // prologue: stage all of K-tile 0
+  stage_half(0, 0); stage_half(1, 0); stage_half(2, 0); stage_half(3, 0);
+
+  for (int t = 0; t < nt; ++t) {
+    const int buf = t & 1;
+    const bool pf = (t + 1) < nt;
+    // first half of next tile staged before the wait so its loads are
+    // already in flight when we start computing
+    if (pf) {
+      stage_half(0, t + 1);
+      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();  // K-tile t fully resident for all waves
+
+    const char* Ahalf = smem + HALF_OFF(buf, wr);
+    const char* Bhalf = smem + HALF_OFF(buf, 2 + (wc >> 1));
+    const int brow0 = (wc & 1) * 64;  // 64-col strip within the B half
+
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 bfr[4];
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        int row = brow0 + ni * 16 + (lane & 15);
+        bfr[ni] = *(bf16x8*)(Bhalf + lds_byte2(row, kk * 4 + (lane >> 4)));
+      }
+#pragma unroll
+      for (int mh = 0; mh < 2; ++mh) {
+        // stage the remaining half-tiles across the phases
+        if (pf) {
+          if (kk == 0 && mh == 1) stage_half(1, t + 1);
+          if (kk == 1 && mh == 0) stage_half(2, t + 1);
+          if (kk == 1 && mh == 1) stage_half(3, t + 1);
+        }
+        bf16x8 afr[4];
+#pragma unroll
+        for (int q = 0; q < 4; ++q) {
+          int mi = mh * 4 + q;
+          int row = mi * 16 + (lane & 15);
+          afr[q] = *(bf16x8*)(Ahalf + lds_byte2(row, kk * 4 + (lane >> 4)));
+        }
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int q = 0; q < 4; ++q)
+#pragma unroll
+          for (int ni = 0; ni < 4; ++ni)
+            acc[mh * 4 + q][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afr[q], bfr[ni], acc[mh * 4 + q][ni], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+      }
+    }
+    // all reads of buf done before the next iteration stages into it
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // ---- epilogue through LDS (reuse all 128 KiB as a 256x256 bf16 tile)
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  __bf16* cl = (__bf16*)smem;
+#pragma unroll
+  for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = wr * 128 + mi * 16 + (lane >> 4) * 4 + r;
+        int col = wc * 64 + ni * 16 + (lane & 15);
+        cl[row * BN2 + col] = (__bf16)acc[mi][ni][r];
+      }
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  // 512 threads: 32 lanes per row (16 B each), 16 rows/pass, 16 passes
+  const int wrow = tid >> 5;
+  const int wcol8 = (tid & 31) * 8;
+#pragma unroll
+  for (int p = 0; p < 16; ++p) {
+    int row_l = p * 16 + wrow;
+    int row_g = row0 + row_l;
+    int col_g = col0 + wcol8;
+    if (row_g < M && col_g + 8 <= N) {
+      bf16x8 v = *(bf16x8*)(cl + row_l * BN2 + wcol8);
+      if (HAS_RES) {
+        bf16x8 rv = *(const bf16x8*)(residual + (int64_t)row_g * N + col_g);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) v[e] = (__bf16)(bf2f(v[e]) + bf2f(rv[e]));
+      }
+      if (relu) {
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          v[e] = bf2f(v[e]) > 0.f ? v[e] : (__bf16)0.f;
+      }
+      *(bf16x8*)(Cout + (int64_t)row_g * N + col_g) = v;
+    }
+  }
+}
+
 // ---------------------------------------------------------------- host
 
 static uint64_t magic40(uint32_t d) {
@@ -290,25 +531,58 @@ static uint64_t magic40(uint32_t d) {
   return ((((__uint128_t)1) << 40) + d - 1) / d;
 }
 
+// Tile dispatch: 256^2 pipelined kernel when the grid still fills the
+// 256-CU chip and the tile isn't mostly padding; else the 128^2 kernel.
+static bool use_256(int M, int N) {
+  long b256 = ((M + BM2 - 1) / BM2) * (long)((N + BN2 - 1) / BN2);
+  return M >= 192 && N >= 192 && b256 >= 256;
+}
+
+template <int MODE>
+static void dispatch(const void* A, const void* B, void* C, const float* bias,
+                     const void* residual, int M, int N, int K, int Kp,
+                     int64_t a_bytes, bool relu, const ConvParams& cp,
+                     hipStream_t stream) {
+  if (use_256(M, N)) {
+    int tiles_m = (M + BM2 - 1) / BM2;
+    int tiles_n = (N + BN2 - 1) / BN2;
+    dim3 grid(tiles_m * tiles_n);
+    if (residual)
+      hipLaunchKernelGGL((gemm_conv_kernel_256<MODE, true>), grid,
+                         dim3(NTHREADS2), LDS2_BYTES, stream,
+                         (const __bf16*)A, (const __bf16*)B, (__bf16*)C, bias,
+                         (const __bf16*)residual, M, N, K, Kp, (int)a_bytes,
+                         tiles_m, tiles_n, relu ? 1 : 0, cp);
+    else
+      hipLaunchKernelGGL((gemm_conv_kernel_256<MODE, false>), grid,
+                         dim3(NTHREADS2), LDS2_BYTES, stream,
+                         (const __bf16*)A, (const __bf16*)B, (__bf16*)C, bias,
+                         (const __bf16*)nullptr, M, N, K, Kp, (int)a_bytes,
+                         tiles_m, tiles_n, relu ? 1 : 0, cp);
+    return;
+  }
+  int tiles_m = (M + BM - 1) / BM;
+  int tiles_n = (N + BN - 1) / BN;
+  dim3 grid(tiles_m * tiles_n);
+  if (residual)
+    hipLaunchKernelGGL((gemm_conv_kernel<MODE, true>), grid, dim3(NTHREADS),
+                       LDS_BYTES, stream, (const __bf16*)A, (const __bf16*)B,
+                       (__bf16*)C, bias, (const __bf16*)residual, M, N, K, Kp,
+                       (int)a_bytes, tiles_m, tiles_n, relu ? 1 : 0, cp);
+  else
+    hipLaunchKernelGGL((gemm_conv_kernel<MODE, false>), grid, dim3(NTHREADS),
+                       LDS_BYTES, stream, (const __bf16*)A, (const __bf16*)B,
+                       (__bf16*)C, bias, (const __bf16*)nullptr, M, N, K, Kp,
+                       (int)a_bytes, tiles_m, tiles_n, relu ? 1 : 0, cp);
+}
+
 void launch_gemm_bf16(const void* A, const void* B, void* C,
                       const float* bias, const void* residual, int M, int N,
                       int K, int Kp, int64_t a_bytes, bool relu,
                       hipStream_t stream) {
-  int tiles_m = (M + BM - 1) / BM;
-  int tiles_n = (N + BN - 1) / BN;
   ConvParams cp{};
-  dim3 grid(tiles_m * tiles_n);
-  if (residual) {
-    hipLaunchKernelGGL((gemm_conv_kernel<0, true>), grid, dim3(NTHREADS),
-                       LDS_BYTES, stream, (const __bf16*)A, (const __bf16*)B,
-                       (__bf16*)C, bias, (const __bf16*)residual, M, N, K, Kp,
-                       (int)a_bytes, tiles_m, tiles_n, relu ? 1 : 0, cp);
-  } else {
-    hipLaunchKernelGGL((gemm_conv_kernel<0, false>), grid, dim3(NTHREADS),
-                       LDS_BYTES, stream, (const __bf16*)A, (const __bf16*)B,
-                       (__bf16*)C, bias, (const __bf16*)nullptr, M, N, K, Kp,
-                       (int)a_bytes, tiles_m, tiles_n, relu ? 1 : 0, cp);
-  }
+  dispatch<0>(A, B, C, bias, residual, M, N, K, Kp, a_bytes, relu, cp,
+              stream);
 }
 
 void launch_conv2d_bf16(const void* X, const void* B, void* C,
@@ -318,8 +592,6 @@ void launch_conv2d_bf16(const void* X, const void* B, void* C,
                         bool relu, hipStream_t stream) {
   int M = Nimg * HO * WO;
   int K = R * S * Cin;
-  int tiles_m = (M + BM - 1) / BM;
-  int tiles_n = (Cout + BN - 1) / BN;
   ConvParams cp;
   cp.C = Cin; cp.H = H; cp.W = W; cp.R = R; cp.S = S;
   cp.sh = sh; cp.sw = sw; cp.pt = pt; cp.pl = pl; cp.HO = HO; cp.WO = WO;
@@ -328,16 +600,6 @@ void launch_conv2d_bf16(const void* X, const void* B, void* C,
   cp.magic_HW = magic40(HO * WO);
   cp.magic_WO = magic40(WO);
   int64_t a_bytes = (int64_t)Nimg * H * W * Cin * 2;
-  dim3 grid(tiles_m * tiles_n);
-  if (residual) {
-    hipLaunchKernelGGL((gemm_conv_kernel<1, true>), grid, dim3(NTHREADS),
-                       LDS_BYTES, stream, (const __bf16*)X, (const __bf16*)B,
-                       (__bf16*)C, bias, (const __bf16*)residual, M, Cout, K,
-                       Kp, (int)a_bytes, tiles_m, tiles_n, relu ? 1 : 0, cp);
-  } else {
-    hipLaunchKernelGGL((gemm_conv_kernel<1, false>), grid, dim3(NTHREADS),
-                       LDS_BYTES, stream, (const __bf16*)X, (const __bf16*)B,
-                       (__bf16*)C, bias, (const __bf16*)nullptr, M, Cout, K,
-                       Kp, (int)a_bytes, tiles_m, tiles_n, relu ? 1 : 0, cp);
-  }
+  dispatch<1>(X, B, C, bias, residual, M, Cout, K, Kp, a_bytes, relu, cp,
+              stream);
 }

@@ -32,7 +32,10 @@ def _pad64_cols(w):
 
 @pytest.mark.parametrize("M,N,K", [(128, 128, 64), (256, 512, 256),
                                    (200, 1000, 2048), (64, 64, 72),
-                                   (1, 16, 64), (3136, 2048, 512)])
+                                   (1, 16, 64), (3136, 2048, 512),
+                                   # 256^2 pipelined-kernel shapes
+                                   (8192, 2048, 128), (4096, 4096, 192),
+                                   (8200, 2048, 64)])
 def test_gemm_bf16(E, M, N, K):
     torch.manual_seed(0)
     A = torch.randn(M, K) * 0.5
@@ -81,6 +84,8 @@ def _tf_conv_ref(X, W, stride, pad_same):
     (2, 28, 28, 128, 1, 256, 2, True),     # 1x1 s2 shortcut
     (1, 30, 30, 32, 5, 48, 1, False),      # inception 5x5 VALID
     (2, 17, 17, 8, 7, 64, 2, True),        # 7x7 s2 stem-like (Cpad=8)
+    (64, 56, 56, 64, 1, 256, 1, True),     # 256^2-kernel 1x1 (gemm path)
+    (64, 56, 56, 64, 3, 256, 1, True),     # 256^2-kernel im2col path
 ])
 def test_conv2d_bf16(E, N, H, W, C, R, O, stride, same):
     torch.manual_seed(2)
