@@ -101,13 +101,23 @@ def main(argv=None):
 
     p = argparse.ArgumentParser("tensorflow_web_deploy_amd server")
     add_config_args(p)
+    p.add_argument("--no-batcher", action="store_true",
+                   help="serve without dynamic batching")
     args = p.parse_args(argv)
     cfg = config_from_args(args)
     if cfg.gpus > 1:
         from ..parallel.replicas import ReplicatedEngine
-        engine = ReplicatedEngine(cfg)
+        backend_engine = ReplicatedEngine(cfg)
     else:
-        engine = LocalEngine(cfg)
+        backend_engine = LocalEngine(cfg)
+    if args.no_batcher:
+        engine = backend_engine
+    else:
+        from ..sched.batcher import BatchedEngine, DynamicBatcher
+        batcher = DynamicBatcher(backend_engine, max_batch=cfg.max_batch,
+                                 max_wait_ms=cfg.max_wait_ms)
+        batcher.start()
+        engine = BatchedEngine(batcher)
     app = create_app(cfg, engine)
     app.run(host=cfg.host, port=cfg.port, threaded=True)
 
