@@ -157,6 +157,35 @@ void scale_shift_bf16(torch::Tensor X, torch::Tensor scale,
                           X.numel() / C, C, relu, cur_stream());
 }
 
+void layernorm_bf16(torch::Tensor X, c10::optional<torch::Tensor> gamma,
+                    c10::optional<torch::Tensor> beta, torch::Tensor Y,
+                    double eps) {
+  check_bf16(X, "X");
+  check_bf16(Y, "Y");
+  int C = X.size(-1);
+  TORCH_CHECK(C % 8 == 0 && C <= 2048, "layernorm C must be %8 and <=2048");
+  const float* g = gamma ? gamma->data_ptr<float>() : nullptr;
+  const float* b = beta ? beta->data_ptr<float>() : nullptr;
+  launch_layernorm_bf16(X.data_ptr(), g, b, Y.data_ptr(), X.numel() / C, C,
+                        (float)eps, cur_stream());
+}
+
+// batched C[b] = A[b] @ B[b]^T  (A [batch,M,K], B [batch,N,K], C [batch,M,N])
+void gemm_batched_bf16(torch::Tensor A, torch::Tensor B, torch::Tensor C) {
+  check_bf16(A, "A");
+  check_bf16(B, "B");
+  check_bf16(C, "C");
+  TORCH_CHECK(A.dim() == 3 && B.dim() == 3 && C.dim() == 3);
+  int batch = A.size(0), M = A.size(1), K = A.size(2);
+  int N = B.size(1);
+  TORCH_CHECK(B.size(0) == batch && C.size(0) == batch);
+  TORCH_CHECK(B.size(2) == K && C.size(1) == M && C.size(2) == N);
+  TORCH_CHECK(K % 8 == 0, "batched gemm K must be a multiple of 8");
+  launch_gemm_batched_bf16(A.data_ptr(), B.data_ptr(), C.data_ptr(), batch,
+                           M, N, K, K, (int64_t)M * K, (int64_t)N * K,
+                           (int64_t)M * N, cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -172,4 +201,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("add_relu_bf16", &add_relu_bf16);
   m.def("bias_act_bf16", &bias_act_bf16);
   m.def("scale_shift_bf16", &scale_shift_bf16);
+  m.def("layernorm_bf16", &layernorm_bf16);
+  m.def("gemm_batched_bf16", &gemm_batched_bf16);
 }

@@ -77,7 +77,16 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
                       const __bf16* __restrict__ residual,
                       int M, int N, int K, int Kp,
                       int a_bytes, int tiles_m, int tiles_n,
-                      int relu, ConvParams cp) {
+                      int relu, ConvParams cp,
+                      int64_t sA = 0, int64_t sB = 0, int64_t sC = 0) {
+  // strided-batch support (ViT attention: blockIdx.y = batch index)
+  {
+    const int64_t bz = blockIdx.y;
+    A += bz * sA;
+    B += bz * sB;
+    Cout += bz * sC;
+    if (HAS_RES) residual += bz * sC;
+  }
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = __builtin_amdgcn_readfirstlane(tid >> 6);
@@ -574,6 +583,21 @@ static void dispatch(const void* A, const void* B, void* C, const float* bias,
                        LDS_BYTES, stream, (const __bf16*)A, (const __bf16*)B,
                        (__bf16*)C, bias, (const __bf16*)nullptr, M, N, K, Kp,
                        (int)a_bytes, tiles_m, tiles_n, relu ? 1 : 0, cp);
+}
+
+void launch_gemm_batched_bf16(const void* A, const void* B, void* C,
+                              int batch, int M, int N, int K, int Kp,
+                              int64_t sA, int64_t sB, int64_t sC,
+                              hipStream_t stream) {
+  ConvParams cp{};
+  int tiles_m = (M + BM - 1) / BM;
+  int tiles_n = (N + BN - 1) / BN;
+  dim3 grid(tiles_m * tiles_n, batch);
+  hipLaunchKernelGGL((gemm_conv_kernel<0, false>), grid, dim3(NTHREADS),
+                     LDS_BYTES, stream, (const __bf16*)A, (const __bf16*)B,
+                     (__bf16*)C, nullptr, (const __bf16*)nullptr, M, N, K,
+                     Kp, (int)(M * (int64_t)K * 2), tiles_m, tiles_n, 0, cp,
+                     sA, sB, sC);
 }
 
 void launch_gemm_bf16(const void* A, const void* B, void* C,

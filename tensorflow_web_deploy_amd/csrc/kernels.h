@@ -26,6 +26,17 @@ void launch_global_mean_bf16(const void* X, void* Y, int Nimg, int HW, int C,
 void launch_softmax_bf16(const void* X, float* Y, int M, int N,
                          hipStream_t stream);
 
+// layernorm.hip
+void launch_layernorm_bf16(const void* X, const float* gamma,
+                           const float* beta, void* Y, int64_t rows, int C,
+                           float eps, hipStream_t stream);
+
+// gemm_conv.hip (strided-batch, MODE 0, no epilogue fusion)
+void launch_gemm_batched_bf16(const void* A, const void* B, void* C,
+                              int batch, int M, int N, int K, int Kp,
+                              int64_t sA, int64_t sB, int64_t sC,
+                              hipStream_t stream);
+
 // elementwise.hip
 void launch_cast_f32_bf16(const float* X, void* Y, int64_t n,
                           hipStream_t stream);

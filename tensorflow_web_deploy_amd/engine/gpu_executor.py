@@ -216,21 +216,132 @@ class GpuExecutor:
             elif op == "Mean":
                 xn = node.inputs[0]
                 X = bufs[xn]
-                N_, H, W, C = shapes[xn]
-                axes = self.consts[node.inputs[1]].reshape(-1).tolist()
-                assert sorted(int(a) for a in axes) == [1, 2], \
-                    f"Mean axes {axes} unsupported on GPU path"
+                sh_in = shapes[xn]
+                axes = sorted(int(a) for a in
+                              self.consts[node.inputs[1]].reshape(-1))
                 keep = bool(node.attrs.get("keep_dims", False))
-                shape = (N_, 1, 1, C) if keep else (N_, C)
-                Y = alloc(node.name, shape)
-                Y2 = Y.view(N_, C)
-                steps.append(lambda X=X, Y2=Y2: E.global_mean_bf16(X, Y2))
+                if len(sh_in) == 4 and axes == [1, 2]:
+                    N_, H, W, C = sh_in
+                    shape = (N_, 1, 1, C) if keep else (N_, C)
+                    Y = alloc(node.name, shape)
+                    Y2 = Y.view(N_, C)
+                    steps.append(lambda Xv=bufs[xn], Y2=Y2:
+                                 E.global_mean_bf16(Xv, Y2))
+                elif len(sh_in) == 3 and axes == [1]:
+                    N_, T, C = sh_in
+                    shape = (N_, 1, C) if keep else (N_, C)
+                    Y = alloc(node.name, shape)
+                    Y2 = Y.view(N_, C)
+                    Xv = bufs[xn].view(N_, T, 1, C)
+                    steps.append(lambda Xv=Xv, Y2=Y2:
+                                 E.global_mean_bf16(Xv, Y2))
+                else:
+                    raise NotImplementedError(
+                        f"Mean axes {axes} on shape {sh_in} (node {node.name})")
             elif op == "Softmax":
                 xn = node.inputs[0]
                 X = bufs[xn]
-                M, N = shapes[xn]
-                Y = alloc(node.name, (M, N), torch.float32)
-                steps.append(lambda X=X, Y=Y: E.softmax_bf16(X, Y))
+                sh_in = shapes[xn]
+                N = sh_in[-1]
+                M = int(np.prod(sh_in[:-1]))
+                if len(sh_in) == 2:
+                    # classifier head: emit probabilities in f32
+                    Y = alloc(node.name, (M, N), torch.float32)
+                    steps.append(lambda X=X, Y=Y: E.softmax_bf16(X, Y))
+                else:
+                    # attention probs: keep bf16 for the downstream GEMM
+                    Y = alloc(node.name, sh_in)
+                    Yf = torch.zeros((M, N), dtype=torch.float32, device=dev)
+                    Yv = Y.view(M, N)
+                    Xv = X.view(M, N)
+                    steps.append(lambda Xv=Xv, Yf=Yf, Yv=Yv:
+                                 (E.softmax_bf16(Xv, Yf),
+                                  E.cast_f32_bf16(Yf, Yv)))
+            elif op == "LayerNorm":
+                xn = node.inputs[0]
+                gamma = self._bias(node.inputs[1], 0)
+                beta = self._bias(node.inputs[2], 0)
+                eps = float(node.attrs.get("epsilon", 1e-6) or 1e-6)
+                Y = alloc(node.name, shapes[xn])
+                steps.append(lambda X=bufs[xn], Y=Y, g=gamma, b=beta, e=eps:
+                             E.layernorm_bf16(X, g, b, Y, e))
+            elif op == "Gelu":
+                xn = node.inputs[0]
+                Y = alloc(node.name, shapes[xn])
+                steps.append(lambda X=bufs[xn], Y=Y:
+                             E.bias_act_bf16(X, None, Y, ACT_GELU))
+            elif op in ("BatchMatMul", "BatchMatMulV2"):
+                an, bn = node.inputs
+                sa = shapes[an]
+                if bn in self.consts and self.consts[bn].ndim == 2:
+                    # rows @ weight: plain GEMM over flattened rows
+                    wt, K, N, Kp, Np = self._prep_dense_weight(bn, False)
+                    assert Np == N, "BatchMatMul weight N must be %8"
+                    M = int(np.prod(sa[:-1]))
+                    Y = alloc(node.name, (*sa[:-1], N))
+                    A2 = bufs[an].view(M, K)
+                    Y2 = Y.view(M, N)
+                    steps.append(lambda A2=A2, wt=wt, Y2=Y2, K=K:
+                                 E.gemm_bf16(A2, wt, Y2, None, None, K,
+                                             False))
+                else:
+                    sb = shapes[bn]
+                    batch = int(np.prod(sa[:-2]))
+                    M, K = sa[-2], sa[-1]
+                    adj_y = bool(node.attrs.get("adj_y"))
+                    A3 = bufs[an].view(batch, M, K)
+                    if adj_y:
+                        # scores = Q @ Kmat^T: B rows are [N][K] already
+                        Nn = sb[-2]
+                        B3 = bufs[bn].view(batch, Nn, K)
+                        Kp8 = K
+                        if K % 8:
+                            raise NotImplementedError("adj_y K%8")
+                        Y = alloc(node.name, (*sa[:-2], M, Nn))
+                        Y3 = Y.view(batch, M, Nn)
+                        steps.append(lambda A3=A3, B3=B3, Y3=Y3:
+                                     E.gemm_batched_bf16(A3, B3, Y3))
+                    else:
+                        # P @ V: transpose V to [N][K] layout and zero-pad
+                        # K to a multiple of 8 (exact: pads contribute 0)
+                        Nn = sb[-1]
+                        K8 = _pad8(K)
+                        Bt = torch.zeros((batch, Nn, K8),
+                                         dtype=torch.bfloat16, device=dev)
+                        Ap = (A3 if K8 == K else
+                              torch.zeros((batch, M, K8),
+                                          dtype=torch.bfloat16, device=dev))
+                        Bsrc = bufs[bn].view(batch, K, Nn)
+                        Y = alloc(node.name, (*sa[:-2], M, Nn))
+                        Y3 = Y.view(batch, M, Nn)
+
+                        def _pv(A3=A3, Ap=Ap, Bt=Bt, Bsrc=Bsrc, Y3=Y3, K=K,
+                                K8=K8):
+                            Bt[:, :, :K].copy_(Bsrc.transpose(1, 2))
+                            if K8 != K:
+                                Ap[:, :, :K].copy_(A3)
+                            E.gemm_batched_bf16(Ap, Bt, Y3)
+                        steps.append(_pv)
+            elif op == "Transpose":
+                xn = node.inputs[0]
+                perm = [int(p) for p in
+                        self.consts[node.inputs[1]].reshape(-1)]
+                out_shape = tuple(shapes[xn][p] for p in perm)
+                Y = alloc(node.name, out_shape)
+                steps.append(lambda X=bufs[xn], Y=Y, perm=perm:
+                             Y.copy_(X.permute(perm)))
+            elif op == "Mul":
+                an, bn = node.inputs
+                cb = self.consts.get(bn)
+                if cb is not None and cb.size == 1:
+                    s = float(cb.reshape(-1)[0])
+                    Y = alloc(node.name, shapes[an])
+                    steps.append(lambda X=bufs[an], Y=Y, s=s:
+                                 torch.mul(X, s, out=Y))
+                else:
+                    Y = alloc(node.name, shapes[an])
+                    steps.append(lambda A=bufs[an], B=bufs[bn], Y=Y:
+                                 torch.mul(A, B, out=Y))
             elif op == "Reshape":
                 xn = node.inputs[0]
                 tgt = [int(v) for v in
@@ -253,9 +364,36 @@ class GpuExecutor:
                 shapes[node.name] = tuple(t.shape)
             elif op in ("Add", "AddV2"):
                 a, b = node.inputs
-                Y = alloc(node.name, shapes[a])
-                steps.append(lambda A=bufs[a], B=bufs[b], Y=Y:
-                             E.add_relu_bf16(A, B, Y, False))
+                # normalize: tensor operand first, const operand second
+                if a in self.consts and b not in self.consts:
+                    a, b = b, a
+                if b in self.consts:
+                    cb = self.consts[b].astype(np.float32)
+                    sh_a = shapes[a]
+                    Y = alloc(node.name, sh_a)
+                    if cb.ndim <= 1 or cb.size == sh_a[-1]:
+                        bias = self._bias(b, cb.size)
+                        steps.append(lambda X=bufs[a], Y=Y, bias=bias:
+                                     E.bias_act_bf16(X, bias, Y, ACT_NONE))
+                    else:
+                        # broadcast const like [1,T,C] (positional embed):
+                        # flatten to a per-row bias of size T*C
+                        flat = cb.reshape(-1)
+                        assert flat.size == int(np.prod(sh_a[1:])), \
+                            f"Add broadcast shape {cb.shape} vs {sh_a}"
+                        key = b + "/flatf32"
+                        if key not in self._weights:
+                            self._weights[key] = torch.from_numpy(
+                                np.ascontiguousarray(flat)).to(dev)
+                        bias = self._weights[key]
+                        Xv = bufs[a].view(sh_a[0], flat.size)
+                        Yv = Y.view(sh_a[0], flat.size)
+                        steps.append(lambda Xv=Xv, Yv=Yv, bias=bias:
+                                     E.bias_act_bf16(Xv, bias, Yv, ACT_NONE))
+                else:
+                    Y = alloc(node.name, shapes[a])
+                    steps.append(lambda A=bufs[a], B=bufs[b], Y=Y:
+                                 E.add_relu_bf16(A, B, Y, False))
             elif op == "Relu":
                 xn = node.inputs[0]
                 Y = alloc(node.name, shapes[xn])

@@ -262,6 +262,115 @@ def fuse_conv_blocks(graph: Graph) -> Graph:
     return Graph(new_nodes, graph.consts, graph.inputs, graph.outputs)
 
 
+def fuse_layernorm(graph: Graph) -> Graph:
+    """Recognize the canonical TF LayerNorm decomposition and fuse it:
+
+        m    = Mean(x, [-1], keep_dims=True)
+        sq   = SquaredDifference(x, m)
+        v    = Mean(sq, [-1], keep_dims=True)
+        veps = AddV2(v, eps)          # eps scalar const
+        rstd = Rsqrt(veps)
+        xm   = Sub(x, m)
+        norm = Mul(xm, rstd)
+        scl  = Mul(norm, gamma)
+        out  = AddV2(scl, beta)
+
+    -> LayerNorm(x, gamma, beta) {epsilon}.
+    """
+    nmap = graph.node_map()
+
+    def producer(name):
+        return nmap.get(name)
+
+    drop = set()
+    new_nodes: List[Node] = []
+    for n in graph.nodes:
+        if n.name in drop:
+            pass
+        matched = False
+        if n.op in ("Add", "AddV2"):
+            # out = AddV2(scl, beta)
+            scl = producer(n.inputs[0])
+            beta = n.inputs[1]
+            if (scl is not None and scl.op == "Mul"
+                    and beta in graph.consts):
+                norm = producer(scl.inputs[0])
+                gamma = scl.inputs[1]
+                if (norm is not None and norm.op == "Mul"
+                        and gamma in graph.consts):
+                    xm = producer(norm.inputs[0])
+                    rstd = producer(norm.inputs[1])
+                    if (xm is not None and xm.op == "Sub"
+                            and rstd is not None and rstd.op == "Rsqrt"):
+                        x_name, m_name = xm.inputs
+                        veps = producer(rstd.inputs[0])
+                        if veps is not None and veps.op in ("Add", "AddV2"):
+                            v = producer(veps.inputs[0])
+                            eps_c = veps.inputs[1]
+                            if (v is not None and v.op == "Mean"
+                                    and eps_c in graph.consts):
+                                sq = producer(v.inputs[0])
+                                if (sq is not None
+                                        and sq.op == "SquaredDifference"
+                                        and sq.inputs[0] == x_name
+                                        and sq.inputs[1] == m_name):
+                                    m = producer(m_name)
+                                    if m is not None and m.op == "Mean" \
+                                            and m.inputs[0] == x_name:
+                                        eps = float(
+                                            graph.consts[eps_c].reshape(-1)[0])
+                                        drop.update({scl.name, norm.name,
+                                                     xm.name, rstd.name,
+                                                     veps.name, v.name,
+                                                     sq.name, m.name})
+                                        new_nodes.append(Node(
+                                            n.name, "LayerNorm",
+                                            [x_name, gamma, beta],
+                                            {"epsilon": eps}))
+                                        matched = True
+        if not matched:
+            new_nodes.append(n)
+    new_nodes = [x for x in new_nodes if x.name not in drop]
+    return Graph(new_nodes, graph.consts, graph.inputs, graph.outputs)
+
+
+def fuse_gelu(graph: Graph) -> Graph:
+    """Recognize erf-GELU:  out = Mul(Mul(x, AddV2(Erf(Mul(x, 1/sqrt2)), 1)),
+    0.5)  (and the 0.5-first variant) -> Gelu(x)."""
+    nmap = graph.node_map()
+    drop = set()
+    new_nodes: List[Node] = []
+    for n in graph.nodes:
+        matched = False
+        if n.op == "Mul":
+            a, b_name = n.inputs
+            half = graph.consts.get(b_name)
+            inner = nmap.get(a)
+            if (half is not None and half.size == 1
+                    and abs(float(half.reshape(-1)[0]) - 0.5) < 1e-6
+                    and inner is not None and inner.op == "Mul"):
+                x_name, e1_name = inner.inputs
+                e1 = nmap.get(e1_name)
+                if e1 is not None and e1.op in ("Add", "AddV2"):
+                    erf = nmap.get(e1.inputs[0])
+                    one = graph.consts.get(e1.inputs[1])
+                    if (erf is not None and erf.op == "Erf"
+                            and one is not None and one.size == 1
+                            and abs(float(one.reshape(-1)[0]) - 1.0) < 1e-6):
+                        scale = nmap.get(erf.inputs[0])
+                        if (scale is not None and scale.op == "Mul"
+                                and scale.inputs[0] == x_name):
+                            drop.update({inner.name, e1.name, erf.name,
+                                         scale.name})
+                            new_nodes.append(Node(n.name, "Gelu", [x_name],
+                                                  {"approximate": False}))
+                            matched = True
+        if not matched:
+            new_nodes.append(n)
+    new_nodes = [x for x in new_nodes if x.name not in drop]
+    return Graph(new_nodes, graph.consts, graph.inputs, graph.outputs)
+
+
 def toposort(graph: Graph) -> Graph:
     nmap = graph.node_map()
     order: List[Node] = []
@@ -296,6 +405,8 @@ def toposort(graph: Graph) -> Graph:
 def optimize(graph: Graph) -> Graph:
     g = strip_identity(graph)
     g = fold_batchnorm(g)
+    g = fuse_layernorm(g)
+    g = fuse_gelu(g)
     g = fuse_conv_blocks(g)
     g = toposort(g)
     return g

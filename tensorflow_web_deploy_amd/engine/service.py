@@ -29,8 +29,12 @@ def build_graph(cfg: Config) -> Graph:
         "resnet50": lambda: M.resnet50.build_resnet50(),
     }
     try:
-        from ..models import inception_v3, vit
+        from ..models import inception_v3
         builders["inception_v3"] = lambda: inception_v3.build_inception_v3()
+    except ImportError:
+        pass
+    try:
+        from ..models import vit
         builders["vit_b16"] = lambda: vit.build_vit_b16()
     except ImportError:
         pass

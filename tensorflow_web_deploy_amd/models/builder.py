@@ -94,6 +94,15 @@ class GraphBuilder:
                          strides=[1, stride, stride, 1], padding=padding,
                          data_format="NHWC", T=gd.AttrValue(type=gd.DT_FLOAT))
 
+    def conv2d_hw(self, x: str, name: str, kh: int, kw: int, cin: int,
+                  cout: int, stride: int = 1, padding: str = "SAME") -> str:
+        """Asymmetric-kernel conv (Inception 1x7/7x1 factorizations)."""
+        return self.conv2d(x, name, kh, kw, cin, cout, stride, padding)
+
+    def concat(self, xs, name: str, axis: int = 3) -> str:
+        ax = self.const(f"{name}/axis", np.array(axis, np.int32))
+        return self.node("ConcatV2", name, list(xs) + [ax], N=len(xs))
+
     def bias_add(self, x: str, name: str, c: int, zero: bool = False) -> str:
         b = self.const(f"{name}/bias",
                        np.zeros(c, np.float32) if zero else

@@ -50,3 +50,23 @@ def test_resnet50_batch_consistency():
     xb = np.repeat(x, 4, axis=0)
     p = gpu.infer(xb)
     np.testing.assert_allclose(p[0], p[3], atol=1e-3)
+
+
+def test_inception_v3_gpu_vs_cpu():
+    gpu, cpu = _engines("inception_v3", 299, 299, 3, 1000)
+    x = (np.random.rand(2, 299, 299, 3).astype(np.float32) - 0.5)
+    pg = gpu.infer(x)
+    pc = cpu.infer(x)
+    np.testing.assert_allclose(pg.sum(1), 1.0, atol=1e-3)
+    assert (pg.argmax(1) == pc.argmax(1)).all()
+    assert np.abs(pg - pc).max() < 0.35
+
+
+def test_vit_b16_gpu_vs_cpu():
+    gpu, cpu = _engines("vit_b16", 224, 224, 3, 1000)
+    x = (np.random.rand(2, 224, 224, 3).astype(np.float32) - 0.5)
+    pg = gpu.infer(x)
+    pc = cpu.infer(x)
+    np.testing.assert_allclose(pg.sum(1), 1.0, atol=1e-3)
+    assert (pg.argmax(1) == pc.argmax(1)).all()
+    assert np.abs(pg - pc).max() < 0.35

@@ -196,3 +196,35 @@ def test_elementwise(E):
     E.scale_shift_bf16(Xg, sc, shf, Y3, False)
     torch.testing.assert_close(Y3.float(), Xg.float() * sc + shf,
                                rtol=1e-2, atol=1e-2)
+
+
+def test_layernorm(E):
+    import torch.nn.functional as F
+    for rows, C in [(196 * 2, 768), (64, 2048), (33, 64)]:
+        X = torch.randn(rows, C) * 2 + 0.5
+        Xg = _bf(X)
+        g = torch.rand(C, device="cuda") + 0.5
+        b = torch.randn(C, device="cuda")
+        Y = torch.zeros_like(Xg)
+        E.layernorm_bf16(Xg, g, b, Y, 1e-6)
+        ref = F.layer_norm(Xg.float(), (C,), g, b, 1e-6)
+        torch.testing.assert_close(Y.float(), ref, rtol=2e-2, atol=2e-2)
+
+
+def test_gemm_batched(E):
+    torch.manual_seed(7)
+    batch, M, N, K = 24, 196, 196, 64
+    A = _bf(torch.randn(batch, M, K))
+    B = _bf(torch.randn(batch, N, K))
+    C = torch.zeros(batch, M, N, dtype=torch.bfloat16, device="cuda")
+    E.gemm_batched_bf16(A, B, C)
+    ref = torch.matmul(A.float(), B.float().transpose(1, 2))
+    torch.testing.assert_close(C.float(), ref, rtol=2e-2, atol=0.2)
+    # P@V shape with padded K (196 -> 200)
+    batch, M, N, K = 24, 196, 64, 200
+    A = _bf(torch.randn(batch, M, K))
+    B = _bf(torch.randn(batch, N, K))
+    C = torch.zeros(batch, M, N, dtype=torch.bfloat16, device="cuda")
+    E.gemm_batched_bf16(A, B, C)
+    ref = torch.matmul(A.float(), B.float().transpose(1, 2))
+    torch.testing.assert_close(C.float(), ref, rtol=2e-2, atol=0.3)

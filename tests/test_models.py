@@ -1,0 +1,56 @@
+"""Model-family builders: structure + CPU forward sanity."""
+import numpy as np
+import torch
+from collections import Counter
+
+from tensorflow_web_deploy_amd.conf.config import Config
+from tensorflow_web_deploy_amd.engine.service import LocalEngine
+
+
+def _run(model, h, w, c, nc, batch=1):
+    cfg = Config(model_name=model, device="cpu", input_height=h,
+                 input_width=w, input_channels=c, num_classes=nc)
+    eng = LocalEngine(cfg)
+    x = np.random.rand(batch, h, w, c).astype(np.float32)
+    p = eng.infer(x)
+    assert p.shape == (batch, nc)
+    np.testing.assert_allclose(p.sum(1), 1.0, atol=1e-4)
+    return eng, Counter(n.op for n in eng.graph.nodes)
+
+
+def test_inception_v3():
+    eng, ops = _run("inception_v3", 299, 299, 3, 1000)
+    assert ops["ConvBlock"] == 94
+    assert ops["ConcatV2"] == 11
+    assert ops["AvgPool"] == 9 and ops["MaxPool"] == 4
+
+
+def test_vit_b16():
+    eng, ops = _run("vit_b16", 224, 224, 3, 1000)
+    # the primitive-op graph must fully fuse
+    assert ops["LayerNorm"] == 25   # 2/layer + final
+    assert ops["Gelu"] == 12
+    assert ops["Softmax"] == 13     # 12 attention + classifier
+    assert "SquaredDifference" not in ops and "Erf" not in ops
+
+
+def test_vit_layernorm_fusion_equivalence():
+    """Fused LayerNorm/Gelu must match the raw primitive graph."""
+    from tensorflow_web_deploy_amd.engine import ir
+    from tensorflow_web_deploy_amd.engine.executor import Executor
+    from tensorflow_web_deploy_amd.models.builder import GraphBuilder
+    from tensorflow_web_deploy_amd.models.vit import _gelu, _layernorm
+    b = GraphBuilder(seed=4)
+    x = b.placeholder("input", [-1, 6, 32])
+    y = _layernorm(b, x, "ln", 32)
+    y = _gelu(b, y, "gelu")
+    b.node("Identity", "out", [y])
+    g = b.graphdef()
+    raw = ir.toposort(ir.strip_identity(ir.from_graphdef(g, ["out"])))
+    opt = ir.optimize(ir.from_graphdef(g, ["out"]))
+    ops = [n.op for n in opt.nodes]
+    assert ops == ["LayerNorm", "Gelu"], ops
+    xv = torch.randn(3, 6, 32)
+    y_raw = Executor(raw).run({"input": xv})[0]
+    y_opt = Executor(opt).run({"input": xv})[0]
+    torch.testing.assert_close(y_raw, y_opt, rtol=1e-5, atol=1e-6)
