@@ -275,19 +275,28 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
     int row_l = p * 16 + wrow;
     int row_g = row0 + row_l;
     int col_g = col0 + wcol8;
-    if (row_g < M && col_g + 8 <= N) {
+    if (row_g < M && col_g < N) {
       bf16x8 v = *(bf16x8*)(cl + row_l * BN + wcol8);
+      const int nvalid = min(8, N - col_g);
       if (HAS_RES) {
-        bf16x8 rv = *(const bf16x8*)(residual + (int64_t)row_g * N + col_g);
 #pragma unroll
-        for (int e = 0; e < 8; ++e) v[e] = (__bf16)(bf2f(v[e]) + bf2f(rv[e]));
+        for (int e = 0; e < 8; ++e)
+          if (e < nvalid)
+            v[e] = (__bf16)(bf2f(v[e]) +
+                            bf2f(residual[(int64_t)row_g * N + col_g + e]));
       }
       if (relu) {
 #pragma unroll
         for (int e = 0; e < 8; ++e)
           v[e] = bf2f(v[e]) > 0.f ? v[e] : (__bf16)0.f;
       }
-      *(bf16x8*)(Cout + (int64_t)row_g * N + col_g) = v;
+      if (nvalid == 8) {
+        *(bf16x8*)(Cout + (int64_t)row_g * N + col_g) = v;
+      } else {
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          if (e < nvalid) Cout[(int64_t)row_g * N + col_g + e] = v[e];
+      }
     }
   }
 }
@@ -516,19 +525,28 @@ void gemm_conv_kernel_256(const __bf16* __restrict__ A,
     int row_l = p * 16 + wrow;
     int row_g = row0 + row_l;
     int col_g = col0 + wcol8;
-    if (row_g < M && col_g + 8 <= N) {
+    if (row_g < M && col_g < N) {
       bf16x8 v = *(bf16x8*)(cl + row_l * BN2 + wcol8);
+      const int nvalid = min(8, N - col_g);
       if (HAS_RES) {
-        bf16x8 rv = *(const bf16x8*)(residual + (int64_t)row_g * N + col_g);
 #pragma unroll
-        for (int e = 0; e < 8; ++e) v[e] = (__bf16)(bf2f(v[e]) + bf2f(rv[e]));
+        for (int e = 0; e < 8; ++e)
+          if (e < nvalid)
+            v[e] = (__bf16)(bf2f(v[e]) +
+                            bf2f(residual[(int64_t)row_g * N + col_g + e]));
       }
       if (relu) {
 #pragma unroll
         for (int e = 0; e < 8; ++e)
           v[e] = bf2f(v[e]) > 0.f ? v[e] : (__bf16)0.f;
       }
-      *(bf16x8*)(Cout + (int64_t)row_g * N + col_g) = v;
+      if (nvalid == 8) {
+        *(bf16x8*)(Cout + (int64_t)row_g * N + col_g) = v;
+      } else {
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          if (e < nvalid) Cout[(int64_t)row_g * N + col_g + e] = v[e];
+      }
     }
   }
 }
