@@ -19,7 +19,10 @@ static inline int grid_for(int64_t work) {
 // -------------------------------------------------------------- casts
 
 __global__ void cast_f32_bf16_kernel(const float* __restrict__ X,
-                                     __bf16* __restrict__ Y, int64_t n8) {
+                                     __bf16* __restrict__ Y, int64_t n8,
+                                     int64_t n) {
+  if (blockIdx.x == 0 && threadIdx.x == 0)
+    for (int64_t i = n8 * 8; i < n; ++i) Y[i] = (__bf16)X[i];
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
        i += (int64_t)gridDim.x * blockDim.x) {
     f32x4 a = *(const f32x4*)(X + i * 8);
@@ -34,7 +37,10 @@ __global__ void cast_f32_bf16_kernel(const float* __restrict__ X,
 }
 
 __global__ void cast_bf16_f32_kernel(const __bf16* __restrict__ X,
-                                     float* __restrict__ Y, int64_t n8) {
+                                     float* __restrict__ Y, int64_t n8,
+                                     int64_t n) {
+  if (blockIdx.x == 0 && threadIdx.x == 0)
+    for (int64_t i = n8 * 8; i < n; ++i) Y[i] = bf2f(X[i]);
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
        i += (int64_t)gridDim.x * blockDim.x) {
     bf16x8 v = *(const bf16x8*)(X + i * 8);
@@ -50,16 +56,16 @@ __global__ void cast_bf16_f32_kernel(const __bf16* __restrict__ X,
 
 void launch_cast_f32_bf16(const float* X, void* Y, int64_t n,
                           hipStream_t stream) {
-  int64_t n8 = n / 8;  // caller guarantees n % 8 == 0
-  hipLaunchKernelGGL(cast_f32_bf16_kernel, dim3(grid_for(n8)), dim3(256), 0,
-                     stream, X, (__bf16*)Y, n8);
+  int64_t n8 = n / 8;
+  hipLaunchKernelGGL(cast_f32_bf16_kernel, dim3(grid_for(n8 + 1)), dim3(256),
+                     0, stream, X, (__bf16*)Y, n8, n);
 }
 
 void launch_cast_bf16_f32(const void* X, float* Y, int64_t n,
                           hipStream_t stream) {
   int64_t n8 = n / 8;
-  hipLaunchKernelGGL(cast_bf16_f32_kernel, dim3(grid_for(n8)), dim3(256), 0,
-                     stream, (const __bf16*)X, Y, n8);
+  hipLaunchKernelGGL(cast_bf16_f32_kernel, dim3(grid_for(n8 + 1)), dim3(256),
+                     0, stream, (const __bf16*)X, Y, n8, n);
 }
 
 // ------------------------------------------------- channel pad (stem)
