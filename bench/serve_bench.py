@@ -1,0 +1,163 @@
+"""HTTP serving benchmark: images/sec (closed-loop) + p50/p99 latency
+(open-loop) through the real Flask stack — the second half of the
+BASELINE metric ("images/sec (whole node) + p50 HTTP latency").
+
+Runs the server in-process on a real TCP port, drives it with
+concurrent clients posting PNG images.
+
+  python bench/serve_bench.py --model resnet50 --duration 10 \
+      --clients 32 [--qps 500] [--gpus N]
+"""
+import argparse
+import io
+import json
+import os
+import sys
+import threading
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import numpy as np  # noqa: E402
+
+
+def _png_bytes(h, w, gray=False):
+    from PIL import Image
+    arr = (np.random.rand(h, w, 3) * 255).astype("uint8")
+    img = Image.fromarray(arr[..., 0] if gray else arr, "L" if gray else "RGB")
+    buf = io.BytesIO()
+    img.save(buf, "PNG")
+    return buf.getvalue()
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--model", default="resnet50")
+    p.add_argument("--duration", type=float, default=10.0)
+    p.add_argument("--clients", type=int, default=32)
+    p.add_argument("--qps", type=float, default=0.0,
+                   help=">0: open-loop at this rate (latency mode)")
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--port", type=int, default=5123)
+    p.add_argument("--max-batch", type=int, default=64)
+    p.add_argument("--max-wait-ms", type=float, default=2.0)
+    args = p.parse_args()
+
+    import torch
+    from tensorflow_web_deploy_amd.conf.config import Config
+    from tensorflow_web_deploy_amd.engine.service import LocalEngine
+    from tensorflow_web_deploy_amd.sched.batcher import (BatchedEngine,
+                                                         DynamicBatcher)
+    from tensorflow_web_deploy_amd.server.app import create_app
+
+    size = 299 if args.model == "inception_v3" else 224
+    gray = args.model == "lenet"
+    h = w = 28 if gray else size
+    cfg = Config(model_name=args.model,
+                 device="cuda" if torch.cuda.is_available() else "cpu",
+                 input_height=h, input_width=w,
+                 input_channels=1 if gray else 3,
+                 num_classes=10 if gray else 1000,
+                 max_batch=args.max_batch, max_wait_ms=args.max_wait_ms,
+                 gpus=args.gpus, port=args.port)
+    if args.gpus > 1:
+        from tensorflow_web_deploy_amd.parallel.replicas import \
+            ReplicatedEngine
+        backend = ReplicatedEngine(cfg)
+    else:
+        backend = LocalEngine(cfg)
+    # warm the engine (compile + HIP-graph capture for common batches)
+    for b in (1, args.max_batch):
+        backend.infer(np.zeros((b, h, w, cfg.input_channels), np.float32))
+    batcher = DynamicBatcher(backend, max_batch=cfg.max_batch,
+                             max_wait_ms=cfg.max_wait_ms)
+    batcher.start()
+    engine = BatchedEngine(batcher)
+    app = create_app(cfg, engine)
+
+    th = threading.Thread(
+        target=lambda: app.run(host="127.0.0.1", port=args.port,
+                               threaded=True, use_reloader=False),
+        daemon=True)
+    th.start()
+    time.sleep(1.0)
+
+    import urllib.request
+
+    img = _png_bytes(h, w, gray)
+    url = f"http://127.0.0.1:{args.port}/predict"
+
+    lat = []
+    lat_lock = threading.Lock()
+    stop = threading.Event()
+    count = [0]
+
+    def post_once():
+        t0 = time.perf_counter()
+        req = urllib.request.Request(url, data=img, headers={
+            "Content-Type": "application/octet-stream"})
+        with urllib.request.urlopen(req, timeout=30) as r:
+            r.read()
+        dt = time.perf_counter() - t0
+        with lat_lock:
+            lat.append(dt)
+            count[0] += 1
+
+    def closed_loop():
+        while not stop.is_set():
+            post_once()
+
+    def open_loop(rate_per_client):
+        interval = 1.0 / rate_per_client
+        nxt = time.perf_counter()
+        while not stop.is_set():
+            now = time.perf_counter()
+            if now < nxt:
+                time.sleep(min(interval, nxt - now))
+                continue
+            nxt += interval
+            post_once()
+
+    # warmup requests
+    for _ in range(4):
+        post_once()
+    with lat_lock:
+        lat.clear()
+        count[0] = 0
+
+    threads = []
+    for _ in range(args.clients):
+        if args.qps > 0:
+            t = threading.Thread(target=open_loop,
+                                 args=(args.qps / args.clients,), daemon=True)
+        else:
+            t = threading.Thread(target=closed_loop, daemon=True)
+        t.start()
+        threads.append(t)
+    t0 = time.time()
+    time.sleep(args.duration)
+    stop.set()
+    elapsed = time.time() - t0
+    for t in threads:
+        t.join(timeout=5.0)
+
+    with lat_lock:
+        ls = sorted(lat)
+        n = count[0]
+    batcher.stop()
+    out = {
+        "metric": "serve_images_per_sec" if args.qps <= 0 else "serve_p50_ms",
+        "mode": "closed" if args.qps <= 0 else f"open@{args.qps}qps",
+        "model": args.model,
+        "clients": args.clients,
+        "images_per_sec": round(n / elapsed, 2),
+        "p50_ms": round(1000 * ls[len(ls) // 2], 2) if ls else None,
+        "p99_ms": round(1000 * ls[int(len(ls) * 0.99)], 2) if ls else None,
+        "requests": n,
+        "batch_trace_tail": batcher.trace[-3:],
+    }
+    print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()

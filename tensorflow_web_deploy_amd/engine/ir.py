@@ -417,8 +417,9 @@ def load_model(path: str, outputs: Optional[List[str]] = None) -> Graph:
     import os
     if os.path.isdir(path):
         from ..ckpt.savedmodel import load_savedmodel
-        graph, sig_out = load_savedmodel(path)
-        return optimize(graph if outputs is None else
-                        Graph(graph.nodes, graph.consts, graph.inputs, outputs))
+        g, sig = load_savedmodel(path)
+        if outputs is None and sig is not None and sig.outputs:
+            outputs = [_canon(n) for n in sig.outputs.values()]
+        return optimize(from_graphdef(g, outputs))
     g = gd.load_frozen_graph(path)
     return optimize(from_graphdef(g, outputs))
