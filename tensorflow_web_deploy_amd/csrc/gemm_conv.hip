@@ -279,11 +279,18 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
       bf16x8 v = *(bf16x8*)(cl + row_l * BN + wcol8);
       const int nvalid = min(8, N - col_g);
       if (HAS_RES) {
+        if (nvalid == 8) {
+          bf16x8 rv = *(const bf16x8*)(residual + (int64_t)row_g * N + col_g);
 #pragma unroll
-        for (int e = 0; e < 8; ++e)
-          if (e < nvalid)
-            v[e] = (__bf16)(bf2f(v[e]) +
-                            bf2f(residual[(int64_t)row_g * N + col_g + e]));
+          for (int e = 0; e < 8; ++e)
+            v[e] = (__bf16)(bf2f(v[e]) + bf2f(rv[e]));
+        } else {
+#pragma unroll
+          for (int e = 0; e < 8; ++e)
+            if (e < nvalid)
+              v[e] = (__bf16)(bf2f(v[e]) +
+                              bf2f(residual[(int64_t)row_g * N + col_g + e]));
+        }
       }
       if (relu) {
 #pragma unroll
@@ -529,11 +536,18 @@ void gemm_conv_kernel_256(const __bf16* __restrict__ A,
       bf16x8 v = *(bf16x8*)(cl + row_l * BN2 + wcol8);
       const int nvalid = min(8, N - col_g);
       if (HAS_RES) {
+        if (nvalid == 8) {
+          bf16x8 rv = *(const bf16x8*)(residual + (int64_t)row_g * N + col_g);
 #pragma unroll
-        for (int e = 0; e < 8; ++e)
-          if (e < nvalid)
-            v[e] = (__bf16)(bf2f(v[e]) +
-                            bf2f(residual[(int64_t)row_g * N + col_g + e]));
+          for (int e = 0; e < 8; ++e)
+            v[e] = (__bf16)(bf2f(v[e]) + bf2f(rv[e]));
+        } else {
+#pragma unroll
+          for (int e = 0; e < 8; ++e)
+            if (e < nvalid)
+              v[e] = (__bf16)(bf2f(v[e]) +
+                              bf2f(residual[(int64_t)row_g * N + col_g + e]));
+        }
       }
       if (relu) {
 #pragma unroll
