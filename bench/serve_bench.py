@@ -30,6 +30,53 @@ def _png_bytes(h, w, gray=False):
     return buf.getvalue()
 
 
+def _client_proc(url, img, nthreads, per_thread_qps, duration, q):
+    import threading
+    import time
+    import urllib.request
+    lat = []
+    lock = threading.Lock()
+    stop_at = time.perf_counter() + duration
+
+    def post_once():
+        t0 = time.perf_counter()
+        req = urllib.request.Request(url, data=img, headers={
+            "Content-Type": "application/octet-stream"})
+        with urllib.request.urlopen(req, timeout=30) as r:
+            r.read()
+        dt = time.perf_counter() - t0
+        with lock:
+            lat.append(dt)
+
+    def run():
+        if per_thread_qps > 0:
+            interval = 1.0 / per_thread_qps
+            nxt = time.perf_counter()
+            while time.perf_counter() < stop_at:
+                now = time.perf_counter()
+                if now < nxt:
+                    time.sleep(min(interval, nxt - now))
+                    continue
+                nxt += interval
+                try:
+                    post_once()
+                except Exception:
+                    pass
+        else:
+            while time.perf_counter() < stop_at:
+                try:
+                    post_once()
+                except Exception:
+                    pass
+
+    ts = [threading.Thread(target=run, daemon=True) for _ in range(nthreads)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join(timeout=duration + 30)
+    q.put(lat)
+
+
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--model", default="resnet50")
@@ -82,68 +129,40 @@ def main():
     th.start()
     time.sleep(1.0)
 
-    import urllib.request
+    # Load generation runs in SEPARATE PROCESSES so the measurement is not
+    # GIL-coupled to the server (each worker runs a few client threads).
+    import multiprocessing as mp
 
     img = _png_bytes(h, w, gray)
     url = f"http://127.0.0.1:{args.port}/predict"
 
-    lat = []
-    lat_lock = threading.Lock()
-    stop = threading.Event()
-    count = [0]
-
-    def post_once():
-        t0 = time.perf_counter()
+    # warmup through the full stack
+    import urllib.request
+    for _ in range(4):
         req = urllib.request.Request(url, data=img, headers={
             "Content-Type": "application/octet-stream"})
-        with urllib.request.urlopen(req, timeout=30) as r:
-            r.read()
-        dt = time.perf_counter() - t0
-        with lat_lock:
-            lat.append(dt)
-            count[0] += 1
+        urllib.request.urlopen(req, timeout=60).read()
 
-    def closed_loop():
-        while not stop.is_set():
-            post_once()
-
-    def open_loop(rate_per_client):
-        interval = 1.0 / rate_per_client
-        nxt = time.perf_counter()
-        while not stop.is_set():
-            now = time.perf_counter()
-            if now < nxt:
-                time.sleep(min(interval, nxt - now))
-                continue
-            nxt += interval
-            post_once()
-
-    # warmup requests
-    for _ in range(4):
-        post_once()
-    with lat_lock:
-        lat.clear()
-        count[0] = 0
-
-    threads = []
-    for _ in range(args.clients):
-        if args.qps > 0:
-            t = threading.Thread(target=open_loop,
-                                 args=(args.qps / args.clients,), daemon=True)
-        else:
-            t = threading.Thread(target=closed_loop, daemon=True)
-        t.start()
-        threads.append(t)
-    t0 = time.time()
-    time.sleep(args.duration)
-    stop.set()
-    elapsed = time.time() - t0
-    for t in threads:
-        t.join(timeout=5.0)
-
-    with lat_lock:
-        ls = sorted(lat)
-        n = count[0]
+    nproc = min(args.clients, 16)
+    per_proc_threads = max(1, args.clients // nproc)
+    ctx = mp.get_context("fork")
+    q = ctx.Queue()
+    per_thread_qps = (args.qps / args.clients) if args.qps > 0 else 0.0
+    procs = []
+    for _ in range(nproc):
+        pr = ctx.Process(target=_client_proc,
+                         args=(url, img, per_proc_threads, per_thread_qps,
+                               args.duration, q), daemon=True)
+        pr.start()
+        procs.append(pr)
+    elapsed = args.duration
+    lat = []
+    for _ in range(nproc):
+        lat.extend(q.get(timeout=args.duration + 60))
+    for pr in procs:
+        pr.join(timeout=10)
+    ls = sorted(lat)
+    n = len(ls)
     batcher.stop()
     out = {
         "metric": "serve_images_per_sec" if args.qps <= 0 else "serve_p50_ms",

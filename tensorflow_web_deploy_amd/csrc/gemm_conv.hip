@@ -67,9 +67,13 @@ struct ConvParams {
   uint64_t magic_C, magic_S, magic_HW, magic_WO;
 };
 
+// Templated tile kernel. TBM x TBN output tile, BK=64 K-steps, wave tile
+// 64x64 => NWAVES = (TBM/64)*(TBN/64). Instantiated as (128,128) for
+// general shapes and (256,64) for narrow-N layers (ResNet stage-1 1x1/3x3
+// with Cout=64) where a 128-wide tile would waste half the MFMA work.
 // MODE: 0 = linear A rows (dense / 1x1 stride-1 conv); 1 = im2col gather
-template <int MODE, bool HAS_RES>
-__global__ __launch_bounds__(NTHREADS)
+template <int TBM, int TBN, int MODE, bool HAS_RES>
+__global__ __launch_bounds__((TBM / 64) * (TBN / 64) * 64)
 void gemm_conv_kernel(const __bf16* __restrict__ A,
                       const __bf16* __restrict__ B,
                       __bf16* __restrict__ Cout,
@@ -79,6 +83,12 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
                       int a_bytes, int tiles_m, int tiles_n,
                       int relu, ConvParams cp,
                       int64_t sA = 0, int64_t sB = 0, int64_t sC = 0) {
+  constexpr int NWAVES = (TBM / 64) * (TBN / 64);
+  constexpr int WC_N = TBN / 64;            // wave columns
+  constexpr int CA = TBM / 8 / NWAVES;      // A-staging glds per wave
+  constexpr int CB = TBN / 8 / NWAVES;      // B-staging glds per wave
+  constexpr int ATILE = TBM * BK * 2;       // bytes
+  constexpr int BTILE = TBN * BK * 2;
   // strided-batch support (ViT attention: blockIdx.y = batch index)
   {
     const int64_t bz = blockIdx.y;
@@ -102,8 +112,8 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
   // tm fastest => consecutive blocks share the B panel (L2 reuse)
   const int tn = bid / tiles_m;
   const int tm = bid - tn * tiles_m;
-  const int row0 = tm * BM;
-  const int col0 = tn * BN;
+  const int row0 = tm * TBM;
+  const int col0 = tn * TBN;
 
   const auto rsrcA = __builtin_amdgcn_make_buffer_rsrc(
       (void*)A, /*stride*/ 0, a_bytes, 0x00020000);
@@ -111,16 +121,17 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
       (void*)B, 0, (int)((int64_t)N * Kp * 2), 0x00020000);
 
   // ---- per-lane staging state ----
-  // Staging instruction j (j = wave*4 + i) covers tile rows j*8 + lane/8,
-  // lane slot = lane&7; source element k = kb + (slot ^ (row&7))*8.
+  // Staging instruction j covers tile rows j*8 + lane/8, lane slot =
+  // lane&7; source element k = kb + (slot ^ (row&7))*8 (rule 21: the
+  // XOR swizzle moves to the source address; LDS image stays linear).
   const int srow_base = (lane >> 3);        // 0..7
   const int sslot = lane & 7;
-  int32_t a_base[4];   // element offset of (row, k=0) or im2col base
-  int32_t a_hb[4], a_wb[4];
-  bool a_valid[4];
+  int32_t a_base[CA];   // element offset of (row, k=0) or im2col base
+  int32_t a_hb[CA], a_wb[CA];
+  bool a_valid[CA];
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    int j = wave * 4 + i;
+  for (int i = 0; i < CA; ++i) {
+    int j = wave * CA + i;
     int row_l = j * 8 + srow_base;
     int row_g = row0 + row_l;
     if (MODE == 0) {
@@ -128,7 +139,7 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
       a_base[i] = row_g * K;
     } else {
       a_valid[i] = row_g < M;
-      uint32_t rg = (uint32_t)min(row_g, M - 1 > 0 ? M - 1 : 0);
+      uint32_t rg = (uint32_t)min(row_g, max(M - 1, 0));
       uint32_t n_img = mdiv(rg, cp.magic_HW);
       uint32_t rem = rg - n_img * (uint32_t)(cp.HO * cp.WO);
       uint32_t ho = mdiv(rem, cp.magic_WO);
@@ -141,11 +152,11 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
     }
   }
   // B rows (cols of C): same staging pattern
-  int32_t b_voff[4];
-  bool b_valid[4];
+  int32_t b_voff[CB];
+  bool b_valid[CB];
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    int j = wave * 4 + i;
+  for (int i = 0; i < CB; ++i) {
+    int j = wave * CB + i;
     int row_l = j * 8 + srow_base;
     int col_g = col0 + row_l;
     b_valid[i] = col_g < N;
@@ -155,7 +166,7 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
   const int swz = srow_base & 7;  // row&7 of this lane's staged row
 
   // ---- accumulators (+ free bias init) ----
-  const int wr = wave >> 1, wc = wave & 1;  // wave tile (64x64)
+  const int wr = wave / WC_N, wc = wave % WC_N;  // wave tile (64x64)
   f32x4 acc[4][4];
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi)
@@ -170,14 +181,15 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
     }
 
   const int nt = (K + BK - 1) / BK;
+  auto a_off = [&](int buf) { return buf * ATILE; };
+  auto b_off = [&](int buf) { return 2 * ATILE + buf * BTILE; };
 
   // ---- staging helper (as lambdas so the main loop stays readable) ----
   auto stage = [&](int buf, int t) {
     const int kb = t * BK;
+    const int k = kb + ((sslot ^ swz) << 3);
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      // A tile
-      int k = kb + ((sslot ^ swz) << 3);
+    for (int i = 0; i < CA; ++i) {
       int32_t voff;
       if (MODE == 0) {
         voff = (a_valid[i] && k < K) ? (a_base[i] + k) * 2 : 0x7ffffff0;
@@ -185,31 +197,33 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
         uint32_t rs = mdiv((uint32_t)k, cp.magic_C);
         uint32_t c = (uint32_t)k - rs * (uint32_t)cp.C;
         uint32_t r = mdiv(rs, cp.magic_S);
-        uint32_t s = rs - r * (uint32_t)cp.S;
+        uint32_t sx = rs - r * (uint32_t)cp.S;
         int hi = a_hb[i] + (int)r;
-        int wi = a_wb[i] + (int)s;
+        int wi = a_wb[i] + (int)sx;
         bool ok = a_valid[i] && k < K && (uint32_t)hi < (uint32_t)cp.H &&
                   (uint32_t)wi < (uint32_t)cp.W;
-        voff = ok ? (a_base[i] + ((int)r * cp.W + (int)s) * cp.C + (int)c) * 2
+        voff = ok ? (a_base[i] + ((int)r * cp.W + (int)sx) * cp.C + (int)c) * 2
                   : 0x7ffffff0;
       }
       __builtin_amdgcn_raw_ptr_buffer_load_lds(
           rsrcA,
-          (__attribute__((address_space(3))) void*)(smem + A_OFF(buf) +
-                                                    (wave * 4 + i) * 1024),
+          (__attribute__((address_space(3))) void*)(smem + a_off(buf) +
+                                                    (wave * CA + i) * 1024),
           16, voff, 0, 0, 0);
+    }
+#pragma unroll
+    for (int i = 0; i < CB; ++i) {
       // B tile ([N][Kp] row-major; zero rows beyond N via OOB)
-      int kB = kb + ((sslot ^ swz) << 3);
-      int32_t voffB = b_valid[i] ? (b_voff[i] + kB) * 2 : 0x7ffffff0;
+      int32_t voffB = b_valid[i] ? (b_voff[i] + k) * 2 : 0x7ffffff0;
       __builtin_amdgcn_raw_ptr_buffer_load_lds(
           rsrcB,
-          (__attribute__((address_space(3))) void*)(smem + B_OFF(buf) +
-                                                    (wave * 4 + i) * 1024),
+          (__attribute__((address_space(3))) void*)(smem + b_off(buf) +
+                                                    (wave * CB + i) * 1024),
           16, voffB, 0, 0, 0);
     }
   };
 
-  // swizzled LDS byte offset of (row, col8) in a [128][64]-bf16 tile
+  // swizzled LDS byte offset of (row, col8) in a [rows][64]-bf16 tile
   auto lds_byte = [](int row, int col8) {
     return row * (BK * 2) + ((col8 ^ (row & 7)) << 4);
   };
@@ -217,18 +231,18 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
   auto compute = [&](int buf) {
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      bf16x8 af[4], bf[4];
+      bf16x8 af[4], bfr[4];
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi) {
         int row = wr * 64 + mi * 16 + (lane & 15);
         int col8 = kk * 4 + (lane >> 4);
-        af[mi] = *(bf16x8*)(smem + A_OFF(buf) + lds_byte(row, col8));
+        af[mi] = *(bf16x8*)(smem + a_off(buf) + lds_byte(row, col8));
       }
 #pragma unroll
       for (int ni = 0; ni < 4; ++ni) {
         int row = wc * 64 + ni * 16 + (lane & 15);
         int col8 = kk * 4 + (lane >> 4);
-        bf[ni] = *(bf16x8*)(smem + B_OFF(buf) + lds_byte(row, col8));
+        bfr[ni] = *(bf16x8*)(smem + b_off(buf) + lds_byte(row, col8));
       }
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -236,12 +250,12 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
 #pragma unroll
         for (int ni = 0; ni < 4; ++ni)
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+              af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
     }
   };
 
-  // ---- main loop ----
+  // ---- main loop (2-buffer overlap; glds drained by __syncthreads) ----
   stage(0, 0);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
@@ -253,7 +267,7 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
 
   // ---- epilogue: stage C tile (bf16) through LDS for wide stores ----
   // D fragment layout (16x16x32): col = lane&15, row = (lane>>4)*4 + reg.
-  __bf16* cl = (__bf16*)smem;  // [BM][BN]
+  __bf16* cl = (__bf16*)smem;  // [TBM][TBN]
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
@@ -262,21 +276,23 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
       for (int r = 0; r < 4; ++r) {
         int row = wr * 64 + mi * 16 + (lane >> 4) * 4 + r;
         int col = wc * 64 + ni * 16 + (lane & 15);
-        cl[row * BN + col] = (__bf16)acc[mi][ni][r];
+        cl[row * TBN + col] = (__bf16)acc[mi][ni][r];
       }
     }
   __syncthreads();
 
-  // write back: 16 lanes per row, 8 bf16 (16 B) per lane
-  const int wrow = tid >> 4;          // 0..15
-  const int wcol8 = (tid & 15) * 8;   // col offset
+  // write back: TBN/8 lanes per row, 8 bf16 (16 B) per lane
+  constexpr int LPR = TBN / 8;                  // lanes per row
+  constexpr int RPP = NWAVES * 64 / LPR;        // rows per pass
+  const int wrow = tid / LPR;
+  const int wcol8 = (tid % LPR) * 8;
 #pragma unroll
-  for (int p = 0; p < 8; ++p) {
-    int row_l = p * 16 + wrow;
+  for (int p = 0; p < TBM / RPP; ++p) {
+    int row_l = p * RPP + wrow;
     int row_g = row0 + row_l;
     int col_g = col0 + wcol8;
     if (row_g < M && col_g < N) {
-      bf16x8 v = *(bf16x8*)(cl + row_l * BN + wcol8);
+      bf16x8 v = *(bf16x8*)(cl + row_l * TBN + wcol8);
       const int nvalid = min(8, N - col_g);
       if (HAS_RES) {
         if (nvalid == 8) {
@@ -573,10 +589,38 @@ static uint64_t magic40(uint32_t d) {
 }
 
 // Tile dispatch: 256^2 pipelined kernel when the grid still fills the
-// 256-CU chip and the tile isn't mostly padding; else the 128^2 kernel.
+// 256-CU chip and the tile isn't mostly padding; a narrow (256,64) tile
+// for N <= 96 layers; else the general (128,128) tile.
 static bool use_256(int M, int N) {
   long b256 = ((M + BM2 - 1) / BM2) * (long)((N + BN2 - 1) / BN2);
   return M >= 192 && N >= 192 && b256 >= 256;
+}
+
+template <int TBM, int TBN, int MODE>
+static void launch_tile(const void* A, const void* B, void* C,
+                        const float* bias, const void* residual, int M,
+                        int N, int K, int Kp, int64_t a_bytes, bool relu,
+                        const ConvParams& cp, hipStream_t stream) {
+  constexpr int THREADS = (TBM / 64) * (TBN / 64) * 64;
+  constexpr int LDS = 2 * (TBM + TBN) * BK * 2;
+  constexpr int LDS_TOTAL = LDS > TBM * TBN * 2 ? LDS : TBM * TBN * 2;
+  int tiles_m = (M + TBM - 1) / TBM;
+  int tiles_n = (N + TBN - 1) / TBN;
+  dim3 grid(tiles_m * tiles_n);
+  if (residual)
+    hipLaunchKernelGGL((gemm_conv_kernel<TBM, TBN, MODE, true>), grid,
+                       dim3(THREADS), LDS_TOTAL, stream, (const __bf16*)A,
+                       (const __bf16*)B, (__bf16*)C, bias,
+                       (const __bf16*)residual, M, N, K, Kp, (int)a_bytes,
+                       tiles_m, tiles_n, relu ? 1 : 0, cp, (int64_t)0,
+                       (int64_t)0, (int64_t)0);
+  else
+    hipLaunchKernelGGL((gemm_conv_kernel<TBM, TBN, MODE, false>), grid,
+                       dim3(THREADS), LDS_TOTAL, stream, (const __bf16*)A,
+                       (const __bf16*)B, (__bf16*)C, bias,
+                       (const __bf16*)nullptr, M, N, K, Kp, (int)a_bytes,
+                       tiles_m, tiles_n, relu ? 1 : 0, cp, (int64_t)0,
+                       (int64_t)0, (int64_t)0);
 }
 
 template <int MODE>
@@ -602,34 +646,12 @@ static void dispatch(const void* A, const void* B, void* C, const float* bias,
                          tiles_m, tiles_n, relu ? 1 : 0, cp);
     return;
   }
-  int tiles_m = (M + BM - 1) / BM;
-  int tiles_n = (N + BN - 1) / BN;
-  dim3 grid(tiles_m * tiles_n);
-  if (residual)
-    hipLaunchKernelGGL((gemm_conv_kernel<MODE, true>), grid, dim3(NTHREADS),
-                       LDS_BYTES, stream, (const __bf16*)A, (const __bf16*)B,
-                       (__bf16*)C, bias, (const __bf16*)residual, M, N, K, Kp,
-                       (int)a_bytes, tiles_m, tiles_n, relu ? 1 : 0, cp);
+  if (N <= 96)
+    launch_tile<256, 64, MODE>(A, B, C, bias, residual, M, N, K, Kp,
+                               a_bytes, relu, cp, stream);
   else
-    hipLaunchKernelGGL((gemm_conv_kernel<MODE, false>), grid, dim3(NTHREADS),
-                       LDS_BYTES, stream, (const __bf16*)A, (const __bf16*)B,
-                       (__bf16*)C, bias, (const __bf16*)nullptr, M, N, K, Kp,
-                       (int)a_bytes, tiles_m, tiles_n, relu ? 1 : 0, cp);
-}
-
-void launch_gemm_batched_bf16(const void* A, const void* B, void* C,
-                              int batch, int M, int N, int K, int Kp,
-                              int64_t sA, int64_t sB, int64_t sC,
-                              hipStream_t stream) {
-  ConvParams cp{};
-  int tiles_m = (M + BM - 1) / BM;
-  int tiles_n = (N + BN - 1) / BN;
-  dim3 grid(tiles_m * tiles_n, batch);
-  hipLaunchKernelGGL((gemm_conv_kernel<0, false>), grid, dim3(NTHREADS),
-                     LDS_BYTES, stream, (const __bf16*)A, (const __bf16*)B,
-                     (__bf16*)C, nullptr, (const __bf16*)nullptr, M, N, K,
-                     Kp, (int)(M * (int64_t)K * 2), tiles_m, tiles_n, 0, cp,
-                     sA, sB, sC);
+    launch_tile<128, 128, MODE>(A, B, C, bias, residual, M, N, K, Kp,
+                                a_bytes, relu, cp, stream);
 }
 
 void launch_gemm_bf16(const void* A, const void* B, void* C,
