@@ -654,6 +654,22 @@ static void dispatch(const void* A, const void* B, void* C, const float* bias,
                                 a_bytes, relu, cp, stream);
 }
 
+void launch_gemm_batched_bf16(const void* A, const void* B, void* C,
+                              int batch, int M, int N, int K, int Kp,
+                              int64_t sA, int64_t sB, int64_t sC,
+                              hipStream_t stream) {
+  ConvParams cp{};
+  int tiles_m = (M + BM - 1) / BM;
+  int tiles_n = (N + BN - 1) / BN;
+  dim3 grid(tiles_m * tiles_n, batch);
+  hipLaunchKernelGGL((gemm_conv_kernel<128, 128, 0, false>), grid,
+                     dim3(NTHREADS), LDS_BYTES, stream, (const __bf16*)A,
+                     (const __bf16*)B, (__bf16*)C, nullptr,
+                     (const __bf16*)nullptr, M, N, K, Kp,
+                     (int)(M * (int64_t)K * 2), tiles_m, tiles_n, 0, cp,
+                     sA, sB, sC);
+}
+
 void launch_gemm_bf16(const void* A, const void* B, void* C,
                       const float* bias, const void* residual, int M, int N,
                       int K, int Kp, int64_t a_bytes, bool relu,
