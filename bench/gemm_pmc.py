@@ -1,0 +1,30 @@
+"""Single-shape GEMM loop for clean PMC counter collection.
+   python bench/gemm_pmc.py [--n 4096] [--iters 20] [--shape conv]"""
+import argparse
+import os
+import sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import torch
+
+p = argparse.ArgumentParser()
+p.add_argument("--n", type=int, default=4096)
+p.add_argument("--iters", type=int, default=20)
+p.add_argument("--shape", default="square")  # square | s43x3
+args = p.parse_args()
+from tensorflow_web_deploy_amd.ops.native import ext
+E = ext()
+if args.shape == "square":
+    n = args.n
+    A = torch.randn(n, n, device="cuda").bfloat16()
+    B = torch.randn(n, n, device="cuda").bfloat16()
+    C = torch.zeros(n, n, dtype=torch.bfloat16, device="cuda")
+    for _ in range(args.iters):
+        E.gemm_bf16(A, B, C, None, None, n, False)
+else:
+    X = torch.randn(64, 7, 7, 512, device="cuda").bfloat16()
+    W2 = torch.randn(512, 4608, device="cuda").bfloat16()
+    Y = torch.zeros(64, 7, 7, 512, dtype=torch.bfloat16, device="cuda")
+    for _ in range(args.iters):
+        E.conv2d_bf16(X, W2, Y, None, None, 3, 3, 1, 1, 1, 1, False)
+torch.cuda.synchronize()
+print("done")
