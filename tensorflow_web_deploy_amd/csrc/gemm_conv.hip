@@ -465,22 +465,25 @@ void gemm_conv_kernel_256(const __bf16* __restrict__ A,
     return row * (BK * 2) + ((col8 ^ (row & 7)) << 4);
   };
 
-  // prologue: stage all of K-tile 0
+  // prologue: stage all of K-tile 0 (8 glds per wave)
   stage_half(0, 0); stage_half(1, 0); stage_half(2, 0); stage_half(3, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
 
+  // Schedule: stage the WHOLE next tile at the top of each iteration so
+  // its 8 glds have a full K-tile of MFMA work (~4k SIMD cycles) as
+  // latency cover; one barrier + one counted drain per K-tile (the
+  // previous schedule staged the last halves only ~1 phase ahead:
+  // MfmaUtil 25%, see profiles/).
   for (int t = 0; t < nt; ++t) {
     const int buf = t & 1;
     const bool pf = (t + 1) < nt;
-    // first half of next tile staged before the wait so its loads are
-    // already in flight when we start computing
     if (pf) {
-      stage_half(0, t + 1);
-      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
-    } else {
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      // overwrite-safe: buf^1 was consumed in iteration t-1 and every
+      // wave passed the barrier above
+      stage_half(0, t + 1); stage_half(1, t + 1);
+      stage_half(2, t + 1); stage_half(3, t + 1);
     }
-    __builtin_amdgcn_s_barrier();  // K-tile t fully resident for all waves
-
     const char* Ahalf = smem + HALF_OFF(buf, wr);
     const char* Bhalf = smem + HALF_OFF(buf, 2 + (wc >> 1));
     const int brow0 = (wc & 1) * 64;  // 64-col strip within the B half
@@ -495,12 +498,6 @@ void gemm_conv_kernel_256(const __bf16* __restrict__ A,
       }
 #pragma unroll
       for (int mh = 0; mh < 2; ++mh) {
-        // stage the remaining half-tiles across the phases
-        if (pf) {
-          if (kk == 0 && mh == 1) stage_half(1, t + 1);
-          if (kk == 1 && mh == 0) stage_half(2, t + 1);
-          if (kk == 1 && mh == 1) stage_half(3, t + 1);
-        }
         bf16x8 afr[4];
 #pragma unroll
         for (int q = 0; q < 4; ++q) {
@@ -518,8 +515,9 @@ void gemm_conv_kernel_256(const __bf16* __restrict__ A,
         __builtin_amdgcn_s_setprio(0);
       }
     }
-    // all reads of buf done before the next iteration stages into it
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    // drain tile t+1's glds (covered by the MFMAs above) + LDS reads,
+    // then sync so the next iteration may read/overwrite
+    asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
   }
 
