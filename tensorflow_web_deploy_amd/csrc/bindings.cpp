@@ -42,9 +42,17 @@ void gemm_bf16(torch::Tensor A, torch::Tensor B, torch::Tensor C,
     check_bf16(*residual, "residual");
     rptr = residual->data_ptr();
   }
+  int sk = splitk_plan((int)M, (int)N, (int)K);
+  torch::Tensor ws;
+  float* wptr = nullptr;
+  if (sk > 1) {
+    ws = torch::empty({(int64_t)sk * M * N},
+                      torch::dtype(torch::kFloat32).device(A.device()));
+    wptr = ws.data_ptr<float>();
+  }
   launch_gemm_bf16(A.data_ptr(), B.data_ptr(), C.data_ptr(), bptr, rptr,
                    (int)M, (int)N, (int)K, (int)Kp,
-                   A.numel() * 2, relu, cur_stream());
+                   A.numel() * 2, relu, cur_stream(), wptr, sk);
 }
 
 // X [N,H,W,C] bf16, W2 [Cout][Kp] (pre-transposed HWIO, K-padded)
@@ -66,10 +74,20 @@ void conv2d_bf16(torch::Tensor X, torch::Tensor W2, torch::Tensor Y,
     check_bf16(*residual, "residual");
     rptr = residual->data_ptr();
   }
+  int M = Nimg * HO * WO;
+  int K = (int)(R * S) * Cin;
+  int sk = splitk_plan(M, Cout, K);
+  torch::Tensor ws;
+  float* wptr = nullptr;
+  if (sk > 1) {
+    ws = torch::empty({(int64_t)sk * M * Cout},
+                      torch::dtype(torch::kFloat32).device(X.device()));
+    wptr = ws.data_ptr<float>();
+  }
   launch_conv2d_bf16(X.data_ptr(), W2.data_ptr(), Y.data_ptr(), bptr, rptr,
                      Nimg, H, W, Cin, (int)R, (int)S, Cout, (int)sh, (int)sw,
                      (int)pt, (int)pl, HO, WO, (int)W2.size(1), relu,
-                     cur_stream());
+                     cur_stream(), wptr, sk);
 }
 
 void maxpool_bf16(torch::Tensor X, torch::Tensor Y, int64_t kh, int64_t kw,

@@ -22,6 +22,7 @@
 //   epilogue staged through LDS for wide dwordx4 stores.
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
+#include <algorithm>
 #include <cstdint>
 
 #include "kernels.h"
@@ -82,7 +83,9 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
                       int M, int N, int K, int Kp,
                       int a_bytes, int tiles_m, int tiles_n,
                       int relu, ConvParams cp,
-                      int64_t sA = 0, int64_t sB = 0, int64_t sC = 0) {
+                      int64_t sA = 0, int64_t sB = 0, int64_t sC = 0,
+                      float* __restrict__ Cpart = nullptr,
+                      int nt_slice = 1 << 28) {
   constexpr int NWAVES = (TBM / 64) * (TBN / 64);
   constexpr int WC_N = TBN / 64;            // wave columns
   constexpr int CA = TBM / 8 / NWAVES;      // A-staging glds per wave
@@ -180,7 +183,10 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
       acc[mi][ni] = {b0, b0, b0, b0};
     }
 
-  const int nt = (K + BK - 1) / BK;
+  // split-K: blockIdx.z selects a contiguous K-step slice
+  const int nt_all = (K + BK - 1) / BK;
+  const int kt0 = blockIdx.z * nt_slice;
+  const int nt = min(nt_all - kt0, nt_slice);
   auto a_off = [&](int buf) { return buf * ATILE; };
   auto b_off = [&](int buf) { return 2 * ATILE + buf * BTILE; };
 
@@ -256,13 +262,31 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
   };
 
   // ---- main loop (2-buffer overlap; glds drained by __syncthreads) ----
-  stage(0, 0);
+  stage(0, kt0);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
   for (int t = 0; t < nt; ++t) {
-    if (t + 1 < nt) stage((t + 1) & 1, t + 1);
+    if (t + 1 < nt) stage((t + 1) & 1, kt0 + t + 1);
     compute(t & 1);
     __syncthreads();  // emits vmcnt(0) drain while glds are in flight
+  }
+
+  if (Cpart != nullptr) {
+    // split-K partial: scattered f32 stores to the slice slab (combined
+    // with bias/residual/act by splitk_combine_kernel)
+    float* slab = Cpart + (int64_t)blockIdx.z * M * N;
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int row_g = row0 + wr * 64 + mi * 16 + (lane >> 4) * 4 + r;
+          int col_g = col0 + wc * 64 + ni * 16 + (lane & 15);
+          if (row_g < M && col_g < N)
+            slab[(int64_t)row_g * N + col_g] = acc[mi][ni][r];
+        }
+    return;
   }
 
   // ---- epilogue: stage C tile (bf16) through LDS for wide stores ----
@@ -586,6 +610,43 @@ static uint64_t magic40(uint32_t d) {
   return ((((__uint128_t)1) << 40) + d - 1) / d;
 }
 
+// ------------------------------- split-K combine (elementwise) ---------
+__global__ void splitk_combine_kernel(const float* __restrict__ Cpart,
+                                      const float* __restrict__ bias,
+                                      const __bf16* __restrict__ residual,
+                                      __bf16* __restrict__ Cout, int64_t MN,
+                                      int N, int splitk, int relu) {
+  for (int64_t i8 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i8 * 8 < MN; i8 += (int64_t)gridDim.x * blockDim.x) {
+    int64_t i = i8 * 8;
+    int rem = (int)min((int64_t)8, MN - i);
+    float v[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) v[e] = 0.f;
+    for (int sidx = 0; sidx < splitk; ++sidx) {
+      const float* slab = Cpart + (int64_t)sidx * MN + i;
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        if (e < rem) v[e] += slab[e];
+    }
+    if (bias) {
+      int col = (int)(i % N);  // N % 8 == 0 when this path is used
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        if (e < rem) v[e] += bias[col + e];
+    }
+    if (residual) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        if (e < rem) v[e] += bf2f(residual[i + e]);
+    }
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      if (e < rem)
+        Cout[i + e] = (__bf16)(relu && v[e] < 0.f ? 0.f : v[e]);
+  }
+}
+
 // Tile dispatch: 256^2 pipelined kernel when the grid still fills the
 // 256-CU chip and the tile isn't mostly padding; a narrow (256,64) tile
 // for N <= 96 layers; else the general (128,128) tile.
@@ -598,12 +659,30 @@ template <int TBM, int TBN, int MODE>
 static void launch_tile(const void* A, const void* B, void* C,
                         const float* bias, const void* residual, int M,
                         int N, int K, int Kp, int64_t a_bytes, bool relu,
-                        const ConvParams& cp, hipStream_t stream) {
+                        const ConvParams& cp, hipStream_t stream,
+                        float* ws = nullptr, int splitk = 1) {
   constexpr int THREADS = (TBM / 64) * (TBN / 64) * 64;
   constexpr int LDS = 2 * (TBM + TBN) * BK * 2;
   constexpr int LDS_TOTAL = LDS > TBM * TBN * 2 ? LDS : TBM * TBN * 2;
   int tiles_m = (M + TBM - 1) / TBM;
   int tiles_n = (N + TBN - 1) / TBN;
+  if (splitk > 1 && ws) {
+    int nt = (K + BK - 1) / BK;
+    int nt_slice = (nt + splitk - 1) / splitk;
+    dim3 grid(tiles_m * tiles_n, 1, splitk);
+    hipLaunchKernelGGL((gemm_conv_kernel<TBM, TBN, MODE, false>), grid,
+                       dim3(THREADS), LDS_TOTAL, stream, (const __bf16*)A,
+                       (const __bf16*)B, (__bf16*)C, nullptr,
+                       (const __bf16*)nullptr, M, N, K, Kp, (int)a_bytes,
+                       tiles_m, tiles_n, 0, cp, (int64_t)0, (int64_t)0,
+                       (int64_t)0, ws, nt_slice);
+    int64_t MN = (int64_t)M * N;
+    int blocks = (int)std::min<int64_t>((MN / 8 + 255) / 256, 2048);
+    hipLaunchKernelGGL(splitk_combine_kernel, dim3(blocks), dim3(256), 0,
+                       stream, ws, bias, (const __bf16*)residual,
+                       (__bf16*)C, MN, N, splitk, relu ? 1 : 0);
+    return;
+  }
   dim3 grid(tiles_m * tiles_n);
   if (residual)
     hipLaunchKernelGGL((gemm_conv_kernel<TBM, TBN, MODE, true>), grid,
@@ -621,11 +700,26 @@ static void launch_tile(const void* A, const void* B, void* C,
                        (int64_t)0, (int64_t)0);
 }
 
+// How many K-slices should this shape use? (1 = no split)
+int splitk_plan(int M, int N, int K) {
+  if (use_256(M, N)) return 1;
+  int TBN = N <= 96 ? 64 : 128;
+  int TBM = N <= 96 ? 256 : 128;
+  long blocks = ((M + TBM - 1) / TBM) * (long)((N + TBN - 1) / TBN);
+  int nt = (K + BK - 1) / BK;
+  if (blocks >= 224 || nt < 4) return 1;
+  int sk = (int)((224 + blocks - 1) / blocks);
+  if (sk > 8) sk = 8;
+  while (sk > 1 && nt / sk < 2) --sk;
+  return sk;
+}
+
 template <int MODE>
 static void dispatch(const void* A, const void* B, void* C, const float* bias,
                      const void* residual, int M, int N, int K, int Kp,
                      int64_t a_bytes, bool relu, const ConvParams& cp,
-                     hipStream_t stream) {
+                     hipStream_t stream, float* ws = nullptr,
+                     int splitk = 1) {
   if (use_256(M, N)) {
     int tiles_m = (M + BM2 - 1) / BM2;
     int tiles_n = (N + BN2 - 1) / BN2;
@@ -646,10 +740,10 @@ static void dispatch(const void* A, const void* B, void* C, const float* bias,
   }
   if (N <= 96)
     launch_tile<256, 64, MODE>(A, B, C, bias, residual, M, N, K, Kp,
-                               a_bytes, relu, cp, stream);
+                               a_bytes, relu, cp, stream, ws, splitk);
   else
     launch_tile<128, 128, MODE>(A, B, C, bias, residual, M, N, K, Kp,
-                                a_bytes, relu, cp, stream);
+                                a_bytes, relu, cp, stream, ws, splitk);
 }
 
 void launch_gemm_batched_bf16(const void* A, const void* B, void* C,
@@ -671,17 +765,18 @@ void launch_gemm_batched_bf16(const void* A, const void* B, void* C,
 void launch_gemm_bf16(const void* A, const void* B, void* C,
                       const float* bias, const void* residual, int M, int N,
                       int K, int Kp, int64_t a_bytes, bool relu,
-                      hipStream_t stream) {
+                      hipStream_t stream, float* ws, int splitk) {
   ConvParams cp{};
   dispatch<0>(A, B, C, bias, residual, M, N, K, Kp, a_bytes, relu, cp,
-              stream);
+              stream, ws, splitk);
 }
 
 void launch_conv2d_bf16(const void* X, const void* B, void* C,
                         const float* bias, const void* residual, int Nimg,
                         int H, int W, int Cin, int R, int S, int Cout, int sh,
                         int sw, int pt, int pl, int HO, int WO, int Kp,
-                        bool relu, hipStream_t stream) {
+                        bool relu, hipStream_t stream, float* ws,
+                        int splitk) {
   int M = Nimg * HO * WO;
   int K = R * S * Cin;
   ConvParams cp;
@@ -693,5 +788,5 @@ void launch_conv2d_bf16(const void* X, const void* B, void* C,
   cp.magic_WO = magic40(WO);
   int64_t a_bytes = (int64_t)Nimg * H * W * Cin * 2;
   dispatch<1>(X, B, C, bias, residual, M, Cout, K, Kp, a_bytes, relu, cp,
-              stream);
+              stream, ws, splitk);
 }

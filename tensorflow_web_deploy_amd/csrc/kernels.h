@@ -4,15 +4,18 @@
 #include <cstdint>
 
 // gemm_conv.hip — C = act(A @ B^T + bias [+ residual]), bf16 in/out, f32 acc
+int splitk_plan(int M, int N, int K);
 void launch_gemm_bf16(const void* A, const void* B, void* C,
                       const float* bias, const void* residual, int M, int N,
                       int K, int Kp, int64_t a_bytes, bool relu,
-                      hipStream_t stream);
+                      hipStream_t stream, float* ws = nullptr,
+                      int splitk = 1);
 void launch_conv2d_bf16(const void* X, const void* B, void* C,
                         const float* bias, const void* residual, int Nimg,
                         int H, int W, int Cin, int R, int S, int Cout, int sh,
                         int sw, int pt, int pl, int HO, int WO, int Kp,
-                        bool relu, hipStream_t stream);
+                        bool relu, hipStream_t stream, float* ws = nullptr,
+                        int splitk = 1);
 
 // pool_reduce.hip
 void launch_maxpool_bf16(const void* X, void* Y, int Nimg, int H, int W,
