@@ -707,7 +707,12 @@ int splitk_plan(int M, int N, int K) {
   int TBM = N <= 96 ? 256 : 128;
   long blocks = ((M + TBM - 1) / TBM) * (long)((N + TBN - 1) / TBN);
   int nt = (K + BK - 1) / BK;
-  if (blocks >= 224 || nt < 4) return 1;
+  // Split only when the chip is badly underfilled: the f32 slab
+  // round-trip costs ~bytes/6TBps, so moderate underfill (>=128 blocks,
+  // or >=64 blocks at small K) is cheaper unsplit (measured: s3 3x3
+  // regressed 30->45us at 196 blocks split 2).
+  if (blocks >= 128 || nt < 4) return 1;
+  if (blocks >= 64 && K < 3072) return 1;
   int sk = (int)((224 + blocks - 1) / blocks);
   if (sk > 8) sk = 8;
   while (sk > 1 && nt / sk < 2) --sk;

@@ -43,7 +43,10 @@ def test_resnet50_gpu_vs_cpu():
     top5 = np.argsort(pc, 1)[:, -5:]
     for i in range(pg.shape[0]):
         assert pg[i].argmax() in top5[i], (pg[i].argmax(), top5[i])
-    assert np.abs(pg - pc).max() < 0.35
+    # distribution agreement (random-init nets have near-flat probs, so
+    # absolute prob bounds are flaky; correlation is stable)
+    corr = np.corrcoef(pg.ravel(), pc.ravel())[0, 1]
+    assert corr > 0.95, corr
 
 
 def test_resnet50_batch_consistency():
