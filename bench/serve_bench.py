@@ -111,6 +111,9 @@ def main():
         from tensorflow_web_deploy_amd.parallel.replicas import \
             ReplicatedEngine
         backend = ReplicatedEngine(cfg)
+    elif torch.cuda.is_available():
+        from tensorflow_web_deploy_amd.engine.service import GpuServingEngine
+        backend = GpuServingEngine(cfg)
     else:
         backend = LocalEngine(cfg)
     # warm the engine (compile + HIP-graph capture for common batches)
@@ -119,7 +122,7 @@ def main():
     batcher = DynamicBatcher(backend, max_batch=cfg.max_batch,
                              max_wait_ms=cfg.max_wait_ms)
     batcher.start()
-    engine = BatchedEngine(batcher)
+    engine = BatchedEngine(batcher, raw=hasattr(backend, "infer_blobs"))
     app = create_app(cfg, engine)
 
     th = threading.Thread(

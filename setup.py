@@ -22,6 +22,7 @@ ext = cpp_extension.CUDAExtension(
         os.path.join(CSRC, "pool_reduce.hip"),
         os.path.join(CSRC, "elementwise.hip"),
         os.path.join(CSRC, "layernorm.hip"),
+        os.path.join(CSRC, "preprocess.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

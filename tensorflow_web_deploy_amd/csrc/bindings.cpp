@@ -203,6 +203,17 @@ void gemm_batched_bf16(torch::Tensor A, torch::Tensor B, torch::Tensor C) {
                            (int64_t)M * N, cur_stream());
 }
 
+// src: uint8 [sh, sw, sc]; dst: f32 [dh, dw, dc] on the given stream
+void resize_norm(torch::Tensor src, torch::Tensor dst, int64_t norm) {
+  TORCH_CHECK(src.is_cuda() && src.scalar_type() == torch::kUInt8
+              && src.is_contiguous());
+  TORCH_CHECK(dst.is_cuda() && dst.scalar_type() == torch::kFloat32
+              && dst.is_contiguous());
+  launch_resize_norm(src.data_ptr(), src.size(0), src.size(1), src.size(2),
+                     dst.data_ptr<float>(), dst.size(0), dst.size(1),
+                     dst.size(2), (int)norm, cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -220,4 +231,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scale_shift_bf16", &scale_shift_bf16);
   m.def("layernorm_bf16", &layernorm_bf16);
   m.def("gemm_batched_bf16", &gemm_batched_bf16);
+  m.def("resize_norm", &resize_norm);
 }

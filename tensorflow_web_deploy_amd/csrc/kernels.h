@@ -54,3 +54,8 @@ void launch_bias_act_bf16(const void* X, const float* bias, void* Y,
 void launch_scale_shift_bf16(const void* X, const float* scale,
                              const float* shift, void* Y, int64_t rows, int C,
                              bool relu, hipStream_t stream);
+
+// preprocess.hip
+void launch_resize_norm(const void* src, int sh, int sw, int sc, float* dst,
+                        int dh, int dw, int dc, int norm,
+                        hipStream_t stream);

@@ -65,6 +65,24 @@ class LocalEngine:
         return out.float().cpu().numpy()
 
 
+class GpuServingEngine(LocalEngine):
+    """LocalEngine + GPU ingest: raw image bytes -> side-stream HIP
+    preprocess -> engine, no CPU resize in the request path."""
+
+    def __init__(self, cfg: Config, graph=None):
+        super().__init__(cfg, graph)
+        from ..ingest.gpu_pipeline import GpuPreprocessor
+        self.pre = GpuPreprocessor(cfg.input_height, cfg.input_width,
+                                   cfg.input_channels, cfg.normalize,
+                                   device=self.device)
+
+    @torch.inference_mode()
+    def infer_blobs(self, blobs) -> np.ndarray:
+        batch = self.pre.process(blobs)
+        out = self.ex.run({self.cfg.input_name: batch})[0]
+        return out.float().cpu().numpy()
+
+
 class FakeEngine:
     """Deterministic engine for HTTP-layer tests: prob peak at a hash of
     the image bytes so tests can assert routing without a model."""

@@ -89,9 +89,14 @@ class DynamicBatcher:
             if not items:
                 continue
             t_close = self.clock()
-            batch = np.stack([im for im, _ in items], axis=0)
+            raw = isinstance(items[0][0], (bytes, bytearray))
             try:
-                probs = self.engine.infer(batch)
+                if raw:
+                    probs = self.engine.infer_blobs(
+                        [im for im, _ in items])
+                else:
+                    batch = np.stack([im for im, _ in items], axis=0)
+                    probs = self.engine.infer(batch)
                 t_done = self.clock()
                 for i, (_, fut) in enumerate(items):
                     fut.set_result(probs[i])
@@ -111,9 +116,15 @@ class BatchedEngine:
     """Adapter giving the batcher the single-image `infer` the HTTP layer
     expects: enqueue one row, wait for the demuxed result."""
 
-    def __init__(self, batcher: DynamicBatcher):
+    def __init__(self, batcher: DynamicBatcher, raw: bool = False):
         self.batcher = batcher
+        if raw:
+            self.infer_blobs = self._infer_blobs
 
     def infer(self, batch: np.ndarray) -> np.ndarray:
         futs = [self.batcher.submit(batch[i]) for i in range(batch.shape[0])]
+        return np.stack([f.result(timeout=60.0) for f in futs], axis=0)
+
+    def _infer_blobs(self, blobs) -> np.ndarray:
+        futs = [self.batcher.submit(b) for b in blobs]
         return np.stack([f.result(timeout=60.0) for f in futs], axis=0)

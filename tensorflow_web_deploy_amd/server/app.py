@@ -62,14 +62,28 @@ def create_app(cfg: Config, engine) -> Flask:
             if _PROM:
                 req_counter.labels("bad_request").inc()
             return jsonify({"error": "no image provided"}), 400
-        try:
-            arr = decode_image(data, (cfg.input_width, cfg.input_height),
-                               cfg.input_channels, cfg.normalize)
-        except Exception as e:
-            if _PROM:
-                req_counter.labels("bad_image").inc()
-            return jsonify({"error": f"cannot decode image: {e}"}), 400
-        probs = engine.infer(arr[None])[0]
+        if hasattr(engine, "infer_blobs"):
+            # GPU-ingest path: cheap header validation here, decode +
+            # side-stream HIP preprocess inside the batch
+            try:
+                import io as _io
+
+                from PIL import Image
+                Image.open(_io.BytesIO(data)).verify()
+            except Exception as e:
+                if _PROM:
+                    req_counter.labels("bad_image").inc()
+                return jsonify({"error": f"cannot decode image: {e}"}), 400
+            probs = engine.infer_blobs([data])[0]
+        else:
+            try:
+                arr = decode_image(data, (cfg.input_width, cfg.input_height),
+                                   cfg.input_channels, cfg.normalize)
+            except Exception as e:
+                if _PROM:
+                    req_counter.labels("bad_image").inc()
+                return jsonify({"error": f"cannot decode image: {e}"}), 400
+            probs = engine.infer(arr[None])[0]
         k = min(cfg.top_k, len(probs))
         top = np.argsort(probs)[::-1][:k]
         dt = (time.perf_counter() - t0) * 1000.0
@@ -105,9 +119,13 @@ def main(argv=None):
                    help="serve without dynamic batching")
     args = p.parse_args(argv)
     cfg = config_from_args(args)
+    import torch
     if cfg.gpus > 1:
         from ..parallel.replicas import ReplicatedEngine
         backend_engine = ReplicatedEngine(cfg)
+    elif torch.cuda.is_available():
+        from ..engine.service import GpuServingEngine
+        backend_engine = GpuServingEngine(cfg)
     else:
         backend_engine = LocalEngine(cfg)
     if args.no_batcher:
@@ -117,7 +135,8 @@ def main(argv=None):
         batcher = DynamicBatcher(backend_engine, max_batch=cfg.max_batch,
                                  max_wait_ms=cfg.max_wait_ms)
         batcher.start()
-        engine = BatchedEngine(batcher)
+        engine = BatchedEngine(batcher,
+                               raw=hasattr(backend_engine, "infer_blobs"))
     app = create_app(cfg, engine)
     app.run(host=cfg.host, port=cfg.port, threaded=True)
 
