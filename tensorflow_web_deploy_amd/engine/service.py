@@ -44,6 +44,15 @@ def build_graph(cfg: Config) -> Graph:
     return ir.optimize(ir.from_graphdef(g, outputs=[cfg.output_name]))
 
 
+def _bucket(n: int, max_batch: int = 512) -> int:
+    """Round a batch up to the next captured size so a serving engine
+    replays a handful of HIP graphs instead of compiling per size."""
+    b = 1
+    while b < n and b < max_batch:
+        b <<= 1
+    return b
+
+
 class LocalEngine:
     def __init__(self, cfg: Config, graph: Optional[Graph] = None):
         self.cfg = cfg
@@ -61,8 +70,14 @@ class LocalEngine:
     @torch.inference_mode()
     def infer(self, batch: np.ndarray) -> np.ndarray:
         x = torch.from_numpy(np.ascontiguousarray(batch, dtype=np.float32))
+        n = x.shape[0]
+        nb = _bucket(n)
+        if nb != n and self.device != "cpu":
+            xp = torch.zeros((nb, *x.shape[1:]), dtype=x.dtype)
+            xp[:n] = x
+            x = xp
         out = self.ex.run({self.cfg.input_name: x})[0]
-        return out.float().cpu().numpy()
+        return out.float().cpu().numpy()[:n]
 
 
 class GpuServingEngine(LocalEngine):
@@ -78,9 +93,17 @@ class GpuServingEngine(LocalEngine):
 
     @torch.inference_mode()
     def infer_blobs(self, blobs) -> np.ndarray:
+        n = len(blobs)
         batch = self.pre.process(blobs)
+        nb = _bucket(n)
+        if nb != n:
+            # the preprocessor's output buffer has headroom; zero-fill
+            # the pad rows so the bucketed graph sees a stable shape
+            full = self.pre._out[:nb]
+            full[n:].zero_()
+            batch = full
         out = self.ex.run({self.cfg.input_name: batch})[0]
-        return out.float().cpu().numpy()
+        return out.float().cpu().numpy()[:n]
 
 
 class FakeEngine:

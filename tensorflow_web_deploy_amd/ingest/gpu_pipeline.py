@@ -16,6 +16,7 @@ tests/test_ingest_gpu.py.
 from __future__ import annotations
 
 import io
+from concurrent.futures import ThreadPoolExecutor
 from typing import List, Optional, Sequence
 
 import numpy as np
@@ -36,6 +37,8 @@ class GpuPreprocessor:
         self._out = torch.empty((max_batch, height, width, channels),
                                 dtype=torch.float32, device=self.device)
         self._pin: dict = {}
+        # PIL decode releases the GIL in its C core: pool the decodes
+        self._pool = ThreadPoolExecutor(max_workers=16)
 
     def _decode_cpu(self, data: bytes) -> np.ndarray:
         from PIL import Image
@@ -56,7 +59,7 @@ class GpuPreprocessor:
         """
         n = len(blobs)
         assert n <= self.max_batch
-        arrs = [self._decode_cpu(b) for b in blobs]
+        arrs = list(self._pool.map(self._decode_cpu, blobs))
         out = self._out[:n]
         with torch.cuda.stream(self.stream):
             for i, arr in enumerate(arrs):
