@@ -26,7 +26,7 @@ def _png_bytes(h, w, gray=False):
     arr = (np.random.rand(h, w, 3) * 255).astype("uint8")
     img = Image.fromarray(arr[..., 0] if gray else arr, "L" if gray else "RGB")
     buf = io.BytesIO()
-    img.save(buf, "PNG")
+    img.save(buf, "JPEG", quality=90)  # realistic upload payload
     return buf.getvalue()
 
 

@@ -61,18 +61,24 @@ class GpuPreprocessor:
         assert n <= self.max_batch
         arrs = list(self._pool.map(self._decode_cpu, blobs))
         out = self._out[:n]
+        # pack all images into one pinned arena (distinct offsets: safe
+        # with the async H2D copies still in flight)
+        total = sum(a.nbytes for a in arrs)
+        arena = self._pin.get("arena")
+        if arena is None or arena.numel() < total:
+            arena = torch.empty(max(total, 1 << 22), dtype=torch.uint8,
+                                pin_memory=True)
+            self._pin["arena"] = arena
+        off = 0
+        offs = []
+        for a in arrs:
+            arena[off:off + a.nbytes].copy_(torch.from_numpy(a.reshape(-1)))
+            offs.append(off)
+            off += a.nbytes
         with torch.cuda.stream(self.stream):
             for i, arr in enumerate(arrs):
-                key = arr.shape
-                pin = self._pin.get(key)
-                if pin is None or pin.shape[0] < arr.nbytes:
-                    pin = torch.empty(arr.nbytes, dtype=torch.uint8,
-                                      pin_memory=True)
-                    self._pin[key] = pin
-                pin[:arr.nbytes].copy_(
-                    torch.from_numpy(arr.reshape(-1)))
-                src = pin[:arr.nbytes].to(self.device, non_blocking=True)
-                src = src.view(arr.shape)
+                src = arena[offs[i]:offs[i] + arr.nbytes].to(
+                    self.device, non_blocking=True).view(arr.shape)
                 self.E.resize_norm(src, out[i], self.norm)
         ev = torch.cuda.Event()
         ev.record(self.stream)
