@@ -122,7 +122,7 @@ def main():
     batcher = DynamicBatcher(backend, max_batch=cfg.max_batch,
                              max_wait_ms=cfg.max_wait_ms)
     batcher.start()
-    engine = BatchedEngine(batcher, raw=hasattr(backend, "infer_blobs"))
+    engine = BatchedEngine(batcher, arrays=hasattr(backend, "infer_arrays"))
     app = create_app(cfg, engine)
 
     th = threading.Thread(

@@ -91,6 +91,23 @@ class GpuServingEngine(LocalEngine):
                                    cfg.input_channels, cfg.normalize,
                                    device=self.device)
 
+    def decode(self, blob: bytes) -> np.ndarray:
+        """CPU entropy decode only (runs in the HTTP request thread);
+        resize/normalize happen on the GPU side stream in the batch."""
+        return self.pre._decode_cpu(blob)
+
+    @torch.inference_mode()
+    def infer_arrays(self, arrs) -> np.ndarray:
+        n = len(arrs)
+        batch = self.pre.process_arrays(arrs)
+        nb = _bucket(n)
+        if nb != n:
+            full = self.pre._out[:nb]
+            full[n:].zero_()
+            batch = full
+        out = self.ex.run({self.cfg.input_name: batch})[0]
+        return out.float().cpu().numpy()[:n]
+
     @torch.inference_mode()
     def infer_blobs(self, blobs) -> np.ndarray:
         n = len(blobs)

@@ -57,9 +57,15 @@ class GpuPreprocessor:
         Upload + resize/normalize run on the side stream; the returned
         tensor is safe to consume on `compute_stream` (event-synced).
         """
-        n = len(blobs)
-        assert n <= self.max_batch
         arrs = list(self._pool.map(self._decode_cpu, blobs))
+        return self.process_arrays(arrs, compute_stream)
+
+    def process_arrays(self, arrs: Sequence[np.ndarray],
+                       compute_stream: Optional[torch.cuda.Stream] = None
+                       ) -> torch.Tensor:
+        """decoded uint8 HWC arrays (any sizes) -> f32 NHWC batch."""
+        n = len(arrs)
+        assert n <= self.max_batch
         out = self._out[:n]
         # pack all images into one pinned arena (distinct offsets: safe
         # with the async H2D copies still in flight)

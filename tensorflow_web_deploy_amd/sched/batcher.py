@@ -89,10 +89,15 @@ class DynamicBatcher:
             if not items:
                 continue
             t_close = self.clock()
-            raw = isinstance(items[0][0], (bytes, bytearray))
+            first = items[0][0]
+            raw = isinstance(first, (bytes, bytearray))
+            u8 = isinstance(first, np.ndarray) and first.dtype == np.uint8
             try:
                 if raw:
                     probs = self.engine.infer_blobs(
+                        [im for im, _ in items])
+                elif u8:
+                    probs = self.engine.infer_arrays(
                         [im for im, _ in items])
                 else:
                     batch = np.stack([im for im, _ in items], axis=0)
@@ -116,10 +121,14 @@ class BatchedEngine:
     """Adapter giving the batcher the single-image `infer` the HTTP layer
     expects: enqueue one row, wait for the demuxed result."""
 
-    def __init__(self, batcher: DynamicBatcher, raw: bool = False):
+    def __init__(self, batcher: DynamicBatcher, raw: bool = False,
+                 arrays: bool = False):
         self.batcher = batcher
         if raw:
             self.infer_blobs = self._infer_blobs
+        if arrays:
+            self.infer_arrays = self._infer_arrays
+            self.decode = batcher.engine.decode
 
     def infer(self, batch: np.ndarray) -> np.ndarray:
         futs = [self.batcher.submit(batch[i]) for i in range(batch.shape[0])]
@@ -127,4 +136,8 @@ class BatchedEngine:
 
     def _infer_blobs(self, blobs) -> np.ndarray:
         futs = [self.batcher.submit(b) for b in blobs]
+        return np.stack([f.result(timeout=60.0) for f in futs], axis=0)
+
+    def _infer_arrays(self, arrs) -> np.ndarray:
+        futs = [self.batcher.submit(a) for a in arrs]
         return np.stack([f.result(timeout=60.0) for f in futs], axis=0)
