@@ -78,6 +78,8 @@ def _client_proc(url, img, nthreads, per_thread_qps, duration, q):
 
 
 def main():
+    import sys as _sys
+    _sys.setswitchinterval(0.002)  # fewer GIL handoffs under thread load
     p = argparse.ArgumentParser()
     p.add_argument("--model", default="resnet50")
     p.add_argument("--duration", type=float, default=10.0)

@@ -214,6 +214,20 @@ void resize_norm(torch::Tensor src, torch::Tensor dst, int64_t norm) {
                      dst.size(2), (int)norm, cur_stream());
 }
 
+// arena u8 (packed images), meta i32 [n,4] = {off, sh, sw, sc} on GPU,
+// dst f32 [n, H, W, C]
+void resize_norm_batch(torch::Tensor arena, torch::Tensor meta,
+                       torch::Tensor dst, int64_t norm) {
+  TORCH_CHECK(arena.is_cuda() && arena.scalar_type() == torch::kUInt8);
+  TORCH_CHECK(meta.is_cuda() && meta.scalar_type() == torch::kInt32);
+  TORCH_CHECK(dst.is_cuda() && dst.scalar_type() == torch::kFloat32
+              && dst.is_contiguous());
+  launch_resize_norm_batch(arena.data_ptr(), meta.data_ptr<int>(),
+                           dst.data_ptr<float>(), dst.size(0), dst.size(1),
+                           dst.size(2), dst.size(3), (int)norm,
+                           cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -232,4 +246,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_bf16", &layernorm_bf16);
   m.def("gemm_batched_bf16", &gemm_batched_bf16);
   m.def("resize_norm", &resize_norm);
+  m.def("resize_norm_batch", &resize_norm_batch);
 }

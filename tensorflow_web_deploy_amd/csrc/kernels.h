@@ -59,3 +59,6 @@ void launch_scale_shift_bf16(const void* X, const float* scale,
 void launch_resize_norm(const void* src, int sh, int sw, int sc, float* dst,
                         int dh, int dw, int dc, int norm,
                         hipStream_t stream);
+void launch_resize_norm_batch(const void* arena, const int* meta,
+                              float* dst, int n, int dh, int dw, int dc,
+                              int norm, hipStream_t stream);

@@ -75,17 +75,19 @@ class GpuPreprocessor:
             arena = torch.empty(max(total, 1 << 22), dtype=torch.uint8,
                                 pin_memory=True)
             self._pin["arena"] = arena
+        meta = np.empty((n, 4), np.int32)
         off = 0
-        offs = []
-        for a in arrs:
+        for i, a in enumerate(arrs):
             arena[off:off + a.nbytes].copy_(torch.from_numpy(a.reshape(-1)))
-            offs.append(off)
+            meta[i] = (off, a.shape[0], a.shape[1], a.shape[2])
             off += a.nbytes
         with torch.cuda.stream(self.stream):
-            for i, arr in enumerate(arrs):
-                src = arena[offs[i]:offs[i] + arr.nbytes].to(
-                    self.device, non_blocking=True).view(arr.shape)
-                self.E.resize_norm(src, out[i], self.norm)
+            # ONE H2D + ONE kernel for the whole super-batch (a per-image
+            # Python loop here gets GIL-starved under request load)
+            arena_gpu = arena[:off].to(self.device, non_blocking=True)
+            meta_gpu = torch.from_numpy(meta).to(self.device,
+                                                 non_blocking=True)
+            self.E.resize_norm_batch(arena_gpu, meta_gpu, out, self.norm)
         ev = torch.cuda.Event()
         ev.record(self.stream)
         target = compute_stream or torch.cuda.current_stream()

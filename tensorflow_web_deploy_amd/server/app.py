@@ -107,6 +107,8 @@ def create_app(cfg: Config, engine) -> Flask:
 
 
 def main(argv=None):
+    import sys as _sys
+    _sys.setswitchinterval(0.002)  # fewer GIL handoffs under thread load
     import argparse
     from ..conf.config import add_config_args, config_from_args
     from ..engine.service import LocalEngine
