@@ -118,13 +118,19 @@ def main():
         backend = GpuServingEngine(cfg)
     else:
         backend = LocalEngine(cfg)
-    # warm the engine (compile + HIP-graph capture for common batches)
-    for b in (1, args.max_batch):
+    # warm the engine (compile + HIP-graph capture for every bucket)
+    b = 1
+    while True:
         backend.infer(np.zeros((b, h, w, cfg.input_channels), np.float32))
+        if b >= args.max_batch:
+            break
+        b = min(b * 2, args.max_batch)
     batcher = DynamicBatcher(backend, max_batch=cfg.max_batch,
                              max_wait_ms=cfg.max_wait_ms)
     batcher.start()
-    engine = BatchedEngine(batcher, arrays=hasattr(backend, "infer_arrays"))
+    engine = BatchedEngine(batcher, raw=hasattr(backend, "infer_blobs"))
+    if hasattr(backend, "pre"):
+        backend.pre.warm_pool()
     app = create_app(cfg, engine)
 
     th = threading.Thread(

@@ -62,17 +62,20 @@ def create_app(cfg: Config, engine) -> Flask:
             if _PROM:
                 req_counter.labels("bad_request").inc()
             return jsonify({"error": "no image provided"}), 400
-        if hasattr(engine, "infer_arrays"):
-            # GPU-ingest path: entropy decode in this request thread
-            # (parallel across HTTP workers), resize/normalize on the
-            # GPU side stream inside the coalesced batch
+        if hasattr(engine, "infer_blobs"):
+            # GPU-ingest path: decode in worker processes + side-stream
+            # HIP preprocess inside the coalesced batch; request thread
+            # only validates the header
             try:
-                arr = engine.decode(data)
+                import io as _io
+
+                from PIL import Image
+                Image.open(_io.BytesIO(data)).verify()
             except Exception as e:
                 if _PROM:
                     req_counter.labels("bad_image").inc()
                 return jsonify({"error": f"cannot decode image: {e}"}), 400
-            probs = engine.infer_arrays([arr])[0]
+            probs = engine.infer_blobs([data])[0]
         else:
             try:
                 arr = decode_image(data, (cfg.input_width, cfg.input_height),
@@ -136,8 +139,7 @@ def main(argv=None):
                                  max_wait_ms=cfg.max_wait_ms)
         batcher.start()
         engine = BatchedEngine(batcher,
-                               arrays=hasattr(backend_engine,
-                                              "infer_arrays"))
+                               raw=hasattr(backend_engine, "infer_blobs"))
     app = create_app(cfg, engine)
     app.run(host=cfg.host, port=cfg.port, threaded=True)
 
