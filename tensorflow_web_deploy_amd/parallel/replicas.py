@@ -125,15 +125,23 @@ TorchDistComm._scatter_known = _scatter_known
 
 
 class ReplicatedEngine:
-    """`infer(batch) -> probs` over N DP replica processes."""
+    """`infer(batch) -> probs` over N DP replica processes.
+
+    Failure policy (SURVEY §5 "failure detection"): a dead replica is
+    detected on the next call; with shrink_on_failure the group is torn
+    down and respawned with world-1 (8 -> 7 -> ...) and the in-flight
+    batch is retried on the survivors."""
 
     def __init__(self, cfg: Config, world: Optional[int] = None,
-                 backend: Optional[str] = None, start_timeout: float = 600.0):
+                 backend: Optional[str] = None, start_timeout: float = 600.0,
+                 shrink_on_failure: bool = True):
         self.cfg = cfg
         self.world = world or cfg.gpus
         if backend is None:
             backend = "nccl" if torch.cuda.is_available() else "gloo"
         self.backend = backend
+        self.shrink_on_failure = shrink_on_failure
+        self._start_timeout = start_timeout
         self._spawn(start_timeout)
 
     def _spawn(self, timeout: float):
@@ -166,17 +174,37 @@ class ReplicatedEngine:
     def _check_health(self):
         try:
             rank, tb = self.err_q.get_nowait()
-            self.shutdown()
+            self._teardown()
             raise RuntimeError(f"replica {rank} died:\n{tb}")
         except queue.Empty:
             pass
         for r, p in enumerate(self.procs):
             if not p.is_alive():
-                self.shutdown()
+                self._teardown()
                 raise RuntimeError(f"replica {r} exited unexpectedly")
 
+    def _teardown(self):
+        for p in self.procs:
+            if p.is_alive():
+                p.terminate()
+        for p in self.procs:
+            p.join(timeout=5.0)
+
     def infer(self, batch: np.ndarray) -> np.ndarray:
-        self.in_q.put(np.ascontiguousarray(batch, dtype=np.float32))
+        batch = np.ascontiguousarray(batch, dtype=np.float32)
+        try:
+            return self._infer_once(batch)
+        except RuntimeError:
+            if not self.shrink_on_failure or self.world <= 1:
+                raise
+            # shrink-serve: respawn with the survivors and retry
+            self.shutdown()
+            self.world -= 1
+            self._spawn(self._start_timeout)
+            return self._infer_once(batch)
+
+    def _infer_once(self, batch: np.ndarray) -> np.ndarray:
+        self.in_q.put(batch)
         while True:
             try:
                 tag, payload = self.out_q.get(timeout=5.0)

@@ -37,9 +37,11 @@ if _PROM:
                          "end-to-end predict latency")
 
 
-def create_app(cfg: Config, engine) -> Flask:
+def create_app(cfg: Config, engine, batcher=None) -> Flask:
     app = Flask(__name__)
     labels = load_labels(cfg.labels_path, cfg.num_classes)
+    if batcher is None:
+        batcher = getattr(engine, "batcher", None)
 
     @app.get("/")
     def index():
@@ -105,6 +107,18 @@ def create_app(cfg: Config, engine) -> Flask:
         @app.get("/metrics")
         def metrics():
             return Response(generate_latest(), mimetype=CONTENT_TYPE_LATEST)
+
+    @app.get("/debug/trace")
+    def debug_trace():
+        """Batching/latency spans ring buffer (SURVEY §5 tracing)."""
+        spans = list(batcher.trace) if batcher is not None else []
+        return jsonify({
+            "batches": spans[-64:],
+            "config": {"max_batch": cfg.max_batch,
+                       "max_wait_ms": cfg.max_wait_ms,
+                       "model": cfg.model_name or cfg.model_path,
+                       "gpus": cfg.gpus},
+        })
 
     return app
 

@@ -49,11 +49,35 @@ def test_replicated_engine_gloo_matches_local():
 def test_replica_death_detected():
     cfg = Config(model_name="lenet", device="cpu", gpus=2)
     from tensorflow_web_deploy_amd.parallel.replicas import ReplicatedEngine
-    eng = ReplicatedEngine(cfg, world=2, backend="gloo")
+    eng = ReplicatedEngine(cfg, world=2, backend="gloo",
+                           shrink_on_failure=False)
     try:
         eng.procs[1].terminate()
         eng.procs[1].join(timeout=10)
         with pytest.raises(RuntimeError):
             eng.infer(np.zeros((2, 28, 28, 1), np.float32))
+    finally:
+        eng.shutdown()
+
+
+@pytest.mark.timeout(300)
+def test_replica_shrink_serves_on_survivors():
+    """Fault injection: kill a replica mid-service; the engine shrinks
+    the group and the in-flight request is retried on the survivors."""
+    cfg = Config(model_name="lenet", device="cpu", gpus=2)
+    from tensorflow_web_deploy_amd.engine.service import LocalEngine
+    from tensorflow_web_deploy_amd.parallel.replicas import ReplicatedEngine
+    local = LocalEngine(cfg.replace(gpus=1))
+    x = np.random.default_rng(8).random((4, 28, 28, 1), dtype=np.float32)
+    want = local.infer(x)
+    eng = ReplicatedEngine(cfg, world=2, backend="gloo",
+                           shrink_on_failure=True)
+    try:
+        np.testing.assert_allclose(eng.infer(x), want, rtol=1e-5, atol=1e-6)
+        eng.procs[1].terminate()
+        eng.procs[1].join(timeout=10)
+        got = eng.infer(x)  # retried on the shrunken (world=1) group
+        assert eng.world == 1
+        np.testing.assert_allclose(got, want, rtol=1e-5, atol=1e-6)
     finally:
         eng.shutdown()
