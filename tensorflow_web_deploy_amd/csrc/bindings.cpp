@@ -136,6 +136,14 @@ void cast_bf16_f32(torch::Tensor X, torch::Tensor Y) {
                        cur_stream());
 }
 
+void cast_f32_pad_bf16(torch::Tensor X, torch::Tensor Y) {
+  TORCH_CHECK(X.scalar_type() == torch::kFloat32 && X.is_contiguous());
+  check_bf16(Y, "Y");
+  int64_t rows = X.numel() / X.size(-1);
+  launch_cast_f32_pad_bf16(X.data_ptr<float>(), Y.data_ptr(), rows,
+                           X.size(-1), Y.size(-1), cur_stream());
+}
+
 void channel_pad_bf16(torch::Tensor X, torch::Tensor Y) {
   check_bf16(X, "X");
   check_bf16(Y, "Y");
@@ -240,6 +248,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cast_f32_bf16", &cast_f32_bf16);
   m.def("cast_bf16_f32", &cast_bf16_f32);
   m.def("channel_pad_bf16", &channel_pad_bf16);
+  m.def("cast_f32_pad_bf16", &cast_f32_pad_bf16);
   m.def("add_relu_bf16", &add_relu_bf16);
   m.def("bias_act_bf16", &bias_act_bf16);
   m.def("scale_shift_bf16", &scale_shift_bf16);

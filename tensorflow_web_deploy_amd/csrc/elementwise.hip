@@ -91,6 +91,27 @@ void launch_channel_pad_bf16(const void* X, void* Y, int64_t rows, int Cin,
                      stream, (const __bf16*)X, (__bf16*)Y, rows, Cin, Cpad);
 }
 
+// f32 [rows][Cin] -> bf16 [rows][Cpad] zero-padded (fused input cast +
+// stem channel pad: saves one full activation pass)
+__global__ void cast_pad_kernel(const float* __restrict__ X,
+                                __bf16* __restrict__ Y, int64_t rows,
+                                int Cin, int Cpad) {
+  int64_t total = rows * Cpad;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int c = (int)(i % Cpad);
+    int64_t r = i / Cpad;
+    Y[i] = (c < Cin) ? (__bf16)X[r * Cin + c] : (__bf16)0.f;
+  }
+}
+
+void launch_cast_f32_pad_bf16(const float* X, void* Y, int64_t rows,
+                              int Cin, int Cpad, hipStream_t stream) {
+  int64_t total = rows * Cpad;
+  hipLaunchKernelGGL(cast_pad_kernel, dim3(grid_for(total)), dim3(256), 0,
+                     stream, X, (__bf16*)Y, rows, Cin, Cpad);
+}
+
 // --------------------------------------------------------- add (+relu)
 
 __global__ void add_relu_kernel(const __bf16* __restrict__ A,

@@ -650,9 +650,11 @@ __global__ void splitk_combine_kernel(const float* __restrict__ Cpart,
 // Tile dispatch: 256^2 pipelined kernel when the grid still fills the
 // 256-CU chip and the tile isn't mostly padding; a narrow (256,64) tile
 // for N <= 96 layers; else the general (128,128) tile.
-static bool use_256(int M, int N) {
+static bool use_256(int M, int N, int K) {
   long b256 = ((M + BM2 - 1) / BM2) * (long)((N + BN2 - 1) / BN2);
-  return M >= 192 && N >= 192 && b256 >= 256;
+  // At K <= 128 (1-2 K-steps) the 256^2 tile is epilogue-dominated
+  // (128 KiB LDS restage per block); the 128^2 tile wins there.
+  return M >= 192 && N >= 192 && K >= 192 && b256 >= 256;
 }
 
 template <int TBM, int TBN, int MODE>
@@ -702,7 +704,7 @@ static void launch_tile(const void* A, const void* B, void* C,
 
 // How many K-slices should this shape use? (1 = no split)
 int splitk_plan(int M, int N, int K) {
-  if (use_256(M, N)) return 1;
+  if (use_256(M, N, K)) return 1;
   int TBN = N <= 96 ? 64 : 128;
   int TBM = N <= 96 ? 256 : 128;
   long blocks = ((M + TBM - 1) / TBM) * (long)((N + TBN - 1) / TBN);
@@ -725,7 +727,7 @@ static void dispatch(const void* A, const void* B, void* C, const float* bias,
                      int64_t a_bytes, bool relu, const ConvParams& cp,
                      hipStream_t stream, float* ws = nullptr,
                      int splitk = 1) {
-  if (use_256(M, N)) {
+  if (use_256(M, N, K)) {
     int tiles_m = (M + BM2 - 1) / BM2;
     int tiles_n = (N + BN2 - 1) / BN2;
     dim3 grid(tiles_m * tiles_n);

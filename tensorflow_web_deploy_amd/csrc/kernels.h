@@ -45,6 +45,8 @@ void launch_cast_f32_bf16(const float* X, void* Y, int64_t n,
                           hipStream_t stream);
 void launch_cast_bf16_f32(const void* X, float* Y, int64_t n,
                           hipStream_t stream);
+void launch_cast_f32_pad_bf16(const float* X, void* Y, int64_t rows,
+                              int Cin, int Cpad, hipStream_t stream);
 void launch_channel_pad_bf16(const void* X, void* Y, int64_t rows, int Cin,
                              int Cpad, hipStream_t stream);
 void launch_add_relu_bf16(const void* A, const void* B, void* Y, int64_t n,

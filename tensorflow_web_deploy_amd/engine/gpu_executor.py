@@ -124,10 +124,25 @@ class GpuExecutor:
         else:
             raise ValueError("GpuExecutor needs cfg for the input shape")
         x_f32 = torch.zeros(in_shape, dtype=torch.float32, device=dev)
-        x_bf = torch.zeros(in_shape, dtype=torch.bfloat16, device=dev)
-        steps.append(lambda: E.cast_f32_bf16(x_f32, x_bf))
-        bufs[inp] = x_bf
-        shapes[inp] = in_shape
+        # If the input feeds exactly one conv that needs channel padding
+        # (C=1/3 stems), fuse the f32->bf16 cast with the pad: one pass.
+        in_consumers = [n for n in g.nodes if inp in n.inputs]
+        pad_fused = False
+        if (len(in_consumers) == 1 and in_consumers[0].op == "ConvBlock"
+                and in_shape[-1] % 8 != 0):
+            Cp = _pad8(in_shape[-1])
+            pshape = (*in_shape[:-1], Cp)
+            x_bf = torch.zeros(pshape, dtype=torch.bfloat16, device=dev)
+            steps.append(lambda: E.cast_f32_pad_bf16(x_f32, x_bf))
+            bufs[inp] = x_bf
+            shapes[inp] = (*in_shape[:-1], in_shape[-1])  # logical C
+            self._input_padded = True
+            pad_fused = True
+        if not pad_fused:
+            x_bf = torch.zeros(in_shape, dtype=torch.bfloat16, device=dev)
+            steps.append(lambda: E.cast_f32_bf16(x_f32, x_bf))
+            bufs[inp] = x_bf
+            shapes[inp] = in_shape
 
         def alloc(name, shape, dtype=torch.bfloat16):
             t = torch.zeros(shape, dtype=dtype, device=dev)
@@ -154,7 +169,7 @@ class GpuExecutor:
                 bias = self._bias(node.attrs.get("bias"), O)
                 res = bufs[node.attrs["residual"]] if node.attrs.get("residual") else None
                 relu = bool(node.attrs.get("relu"))
-                if Cp != Cin:
+                if Cp != Cin and X.shape[-1] != Cp:
                     Xp = torch.zeros((N_, H, W, Cp), dtype=torch.bfloat16,
                                      device=dev)
                     steps.append(lambda X=X, Xp=Xp: E.channel_pad_bf16(X, Xp))
