@@ -652,9 +652,10 @@ __global__ void splitk_combine_kernel(const float* __restrict__ Cpart,
 // for N <= 96 layers; else the general (128,128) tile.
 static bool use_256(int M, int N, int K) {
   long b256 = ((M + BM2 - 1) / BM2) * (long)((N + BN2 - 1) / BN2);
-  // At K <= 128 (1-2 K-steps) the 256^2 tile is epilogue-dominated
-  // (128 KiB LDS restage per block); the 128^2 tile wins there.
-  return M >= 192 && N >= 192 && K >= 192 && b256 >= 256;
+  // At K = 64 (one K-step) the 256^2 tile is epilogue-dominated
+  // (128 KiB LDS restage per block); the 128^2 tile wins there
+  // (measured: 1x1 C64->256 189 -> 209 TF; K=128 already favors 256^2).
+  return M >= 192 && N >= 192 && K >= 128 && b256 >= 256;
 }
 
 template <int TBM, int TBN, int MODE>
