@@ -20,6 +20,12 @@ if args.shape == "square":
     C = torch.zeros(n, n, dtype=torch.bfloat16, device="cuda")
     for _ in range(args.iters):
         E.gemm_bf16(A, B, C, None, None, n, False)
+elif args.shape == "s13x3":
+    X = torch.randn(64, 56, 56, 64, device="cuda").bfloat16()
+    W2 = torch.randn(64, 576, device="cuda").bfloat16()
+    Y = torch.zeros(64, 56, 56, 64, dtype=torch.bfloat16, device="cuda")
+    for _ in range(args.iters):
+        E.conv2d_bf16(X, W2, Y, None, None, 3, 3, 1, 1, 1, 1, False)
 else:
     X = torch.randn(64, 7, 7, 512, device="cuda").bfloat16()
     W2 = torch.randn(512, 4608, device="cuda").bfloat16()
