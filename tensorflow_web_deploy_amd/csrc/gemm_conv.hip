@@ -73,7 +73,7 @@ struct ConvParams {
 // general shapes and (256,64) for narrow-N layers (ResNet stage-1 1x1/3x3
 // with Cout=64) where a 128-wide tile would waste half the MFMA work.
 // MODE: 0 = linear A rows (dense / 1x1 stride-1 conv); 1 = im2col gather
-template <int TBM, int TBN, int MODE, bool HAS_RES>
+template <int TBM, int TBN, int MODE, bool HAS_RES, int DEPTH = 2>
 __global__ __launch_bounds__((TBM / 64) * (TBN / 64) * 64)
 void gemm_conv_kernel(const __bf16* __restrict__ A,
                       const __bf16* __restrict__ B,
@@ -188,7 +188,7 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
   const int kt0 = blockIdx.z * nt_slice;
   const int nt = min(nt_all - kt0, nt_slice);
   auto a_off = [&](int buf) { return buf * ATILE; };
-  auto b_off = [&](int buf) { return 2 * ATILE + buf * BTILE; };
+  auto b_off = [&](int buf) { return DEPTH * ATILE + buf * BTILE; };
 
   // ---- staging helper (as lambdas so the main loop stays readable) ----
   auto stage = [&](int buf, int t) {
@@ -261,15 +261,37 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
     }
   };
 
-  // ---- main loop (2-buffer overlap; glds drained by __syncthreads) ----
-  stage(0, kt0);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  __syncthreads();
+  // ---- main loop: DEPTH-deep pipeline ----
+  // Prefetch distance PF = DEPTH-1 K-tiles; per-iteration:
+  //   counted wait (tile t landed, newer stages stay in flight)
+  //   -> barrier -> stage tile t+PF -> compute tile t.
+  // DEPTH=3 gives the narrow (256,64) tile (544 MFMA cycles/K-step)
+  // two full iterations of latency cover (PMC: 44% barrier-parked at
+  // DEPTH=2 on s1 3x3 — see profiles/).
+  constexpr int PF = DEPTH - 1;
+  constexpr int GPS = CA + CB;  // glds per stage per wave
+#pragma unroll
+  for (int i = 0; i < PF; ++i)
+    if (i < nt) stage(i % DEPTH, kt0 + i);
   for (int t = 0; t < nt; ++t) {
-    if (t + 1 < nt) stage((t + 1) & 1, kt0 + t + 1);
-    compute(t & 1);
-    __syncthreads();  // emits vmcnt(0) drain while glds are in flight
+    // stages newer than tile t still in flight at this point: the
+    // prologue/earlier iterations staged through tile t+PF-1
+    const int pending = min(nt - 1 - t, PF - 1);
+    if (pending >= 2)
+      asm volatile("s_waitcnt vmcnt(%0)" ::"i"(2 * GPS) : "memory");
+    else if (pending == 1)
+      asm volatile("s_waitcnt vmcnt(%0)" ::"i"(GPS) : "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    if (t + PF < nt) stage((t + PF) % DEPTH, kt0 + t + PF);
+    compute(t % DEPTH);
+    // reads of buf t%DEPTH complete before the next barrier (compiler
+    // lgkm waits precede the MFMAs); ordering for overwrite is the
+    // head barrier of iteration t+1
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
   }
+  __builtin_amdgcn_s_barrier();
 
   if (Cpart != nullptr) {
     // split-K partial: scattered f32 stores to the slice slab (combined
@@ -658,14 +680,14 @@ static bool use_256(int M, int N, int K) {
   return M >= 192 && N >= 192 && K >= 128 && b256 >= 256;
 }
 
-template <int TBM, int TBN, int MODE>
+template <int TBM, int TBN, int MODE, int DEPTH = 2>
 static void launch_tile(const void* A, const void* B, void* C,
                         const float* bias, const void* residual, int M,
                         int N, int K, int Kp, int64_t a_bytes, bool relu,
                         const ConvParams& cp, hipStream_t stream,
                         float* ws = nullptr, int splitk = 1) {
   constexpr int THREADS = (TBM / 64) * (TBN / 64) * 64;
-  constexpr int LDS = 2 * (TBM + TBN) * BK * 2;
+  constexpr int LDS = DEPTH * (TBM + TBN) * BK * 2;
   constexpr int LDS_TOTAL = LDS > TBM * TBN * 2 ? LDS : TBM * TBN * 2;
   int tiles_m = (M + TBM - 1) / TBM;
   int tiles_n = (N + TBN - 1) / TBN;
@@ -673,7 +695,7 @@ static void launch_tile(const void* A, const void* B, void* C,
     int nt = (K + BK - 1) / BK;
     int nt_slice = (nt + splitk - 1) / splitk;
     dim3 grid(tiles_m * tiles_n, 1, splitk);
-    hipLaunchKernelGGL((gemm_conv_kernel<TBM, TBN, MODE, false>), grid,
+    hipLaunchKernelGGL((gemm_conv_kernel<TBM, TBN, MODE, false, DEPTH>), grid,
                        dim3(THREADS), LDS_TOTAL, stream, (const __bf16*)A,
                        (const __bf16*)B, (__bf16*)C, nullptr,
                        (const __bf16*)nullptr, M, N, K, Kp, (int)a_bytes,
@@ -688,14 +710,14 @@ static void launch_tile(const void* A, const void* B, void* C,
   }
   dim3 grid(tiles_m * tiles_n);
   if (residual)
-    hipLaunchKernelGGL((gemm_conv_kernel<TBM, TBN, MODE, true>), grid,
+    hipLaunchKernelGGL((gemm_conv_kernel<TBM, TBN, MODE, true, DEPTH>), grid,
                        dim3(THREADS), LDS_TOTAL, stream, (const __bf16*)A,
                        (const __bf16*)B, (__bf16*)C, bias,
                        (const __bf16*)residual, M, N, K, Kp, (int)a_bytes,
                        tiles_m, tiles_n, relu ? 1 : 0, cp, (int64_t)0,
                        (int64_t)0, (int64_t)0);
   else
-    hipLaunchKernelGGL((gemm_conv_kernel<TBM, TBN, MODE, false>), grid,
+    hipLaunchKernelGGL((gemm_conv_kernel<TBM, TBN, MODE, false, DEPTH>), grid,
                        dim3(THREADS), LDS_TOTAL, stream, (const __bf16*)A,
                        (const __bf16*)B, (__bf16*)C, bias,
                        (const __bf16*)nullptr, M, N, K, Kp, (int)a_bytes,
@@ -747,11 +769,11 @@ static void dispatch(const void* A, const void* B, void* C, const float* bias,
     return;
   }
   if (N <= 96)
-    launch_tile<256, 64, MODE>(A, B, C, bias, residual, M, N, K, Kp,
-                               a_bytes, relu, cp, stream, ws, splitk);
+    launch_tile<256, 64, MODE, 3>(A, B, C, bias, residual, M, N, K, Kp,
+                                  a_bytes, relu, cp, stream, ws, splitk);
   else
-    launch_tile<128, 128, MODE>(A, B, C, bias, residual, M, N, K, Kp,
-                                a_bytes, relu, cp, stream, ws, splitk);
+    launch_tile<128, 128, MODE, 2>(A, B, C, bias, residual, M, N, K, Kp,
+                                   a_bytes, relu, cp, stream, ws, splitk);
 }
 
 void launch_gemm_batched_bf16(const void* A, const void* B, void* C,
@@ -762,7 +784,7 @@ void launch_gemm_batched_bf16(const void* A, const void* B, void* C,
   int tiles_m = (M + BM - 1) / BM;
   int tiles_n = (N + BN - 1) / BN;
   dim3 grid(tiles_m * tiles_n, batch);
-  hipLaunchKernelGGL((gemm_conv_kernel<128, 128, 0, false>), grid,
+  hipLaunchKernelGGL((gemm_conv_kernel<128, 128, 0, false, 2>), grid,
                      dim3(NTHREADS), LDS_BYTES, stream, (const __bf16*)A,
                      (const __bf16*)B, (__bf16*)C, nullptr,
                      (const __bf16*)nullptr, M, N, K, Kp,
