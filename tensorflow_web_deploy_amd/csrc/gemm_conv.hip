@@ -768,8 +768,11 @@ static void dispatch(const void* A, const void* B, void* C, const float* bias,
                          tiles_m, tiles_n, relu ? 1 : 0, cp);
     return;
   }
+  // DEPTH=3 measured WORSE here (120 KiB LDS -> 1 block/CU, losing the
+  // cross-block overlap that was hiding half the staging latency):
+  // stem 537->351 TF, s1 3x3 500->325. Keep DEPTH=2 + 2 blocks/CU.
   if (N <= 96)
-    launch_tile<256, 64, MODE, 3>(A, B, C, bias, residual, M, N, K, Kp,
+    launch_tile<256, 64, MODE, 2>(A, B, C, bias, residual, M, N, K, Kp,
                                   a_bytes, relu, cp, stream, ws, splitk);
   else
     launch_tile<128, 128, MODE, 2>(A, B, C, bias, residual, M, N, K, Kp,
