@@ -20,14 +20,23 @@ void check_bf16(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
 }
 
+// 2D output that may be a column-slice view: [M][N] with row stride ldc
+int check_out2d(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.dim() == 2 && t.stride(1) == 1, name,
+              " must be 2D with contiguous rows");
+  return (int)t.stride(0);
+}
+
 // C[M][N] = act(A[M][K] @ B[N][K]^T + bias [+ residual])
 void gemm_bf16(torch::Tensor A, torch::Tensor B, torch::Tensor C,
                c10::optional<torch::Tensor> bias,
                c10::optional<torch::Tensor> residual, int64_t K,
-               bool relu) {
+               int64_t act) {
   check_bf16(A, "A");
   check_bf16(B, "B");
-  check_bf16(C, "C");
+  int ldc = check_out2d(C, "C");
   int64_t M = C.size(0), N = C.size(1);
   int64_t Kp = B.size(1);
   TORCH_CHECK(B.size(0) >= N, "B rows < N");
@@ -42,7 +51,7 @@ void gemm_bf16(torch::Tensor A, torch::Tensor B, torch::Tensor C,
     check_bf16(*residual, "residual");
     rptr = residual->data_ptr();
   }
-  int sk = splitk_plan((int)M, (int)N, (int)K);
+  int sk = ldc == (int)N ? splitk_plan((int)M, (int)N, (int)K) : 1;
   torch::Tensor ws;
   float* wptr = nullptr;
   if (sk > 1) {
@@ -52,18 +61,23 @@ void gemm_bf16(torch::Tensor A, torch::Tensor B, torch::Tensor C,
   }
   launch_gemm_bf16(A.data_ptr(), B.data_ptr(), C.data_ptr(), bptr, rptr,
                    (int)M, (int)N, (int)K, (int)Kp,
-                   A.numel() * 2, relu, cur_stream(), wptr, sk);
+                   A.numel() * 2, (int)act, cur_stream(), wptr, sk, ldc);
 }
 
 // X [N,H,W,C] bf16, W2 [Cout][Kp] (pre-transposed HWIO, K-padded)
 void conv2d_bf16(torch::Tensor X, torch::Tensor W2, torch::Tensor Y,
                  c10::optional<torch::Tensor> bias,
                  c10::optional<torch::Tensor> residual, int64_t R, int64_t S,
-                 int64_t sh, int64_t sw, int64_t pt, int64_t pl, bool relu) {
+                 int64_t sh, int64_t sw, int64_t pt, int64_t pl,
+                 int64_t act) {
   check_bf16(X, "X");
   check_bf16(W2, "W");
-  check_bf16(Y, "Y");
   TORCH_CHECK(X.dim() == 4 && Y.dim() == 4);
+  // Y may be a channel-slice of a concat buffer: last dim contiguous,
+  // pixel stride = total concat channels
+  TORCH_CHECK(Y.is_cuda() && Y.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(Y.stride(3) == 1);
+  int ldc = (int)Y.stride(2);
   int Nimg = X.size(0), H = X.size(1), W = X.size(2), Cin = X.size(3);
   int HO = Y.size(1), WO = Y.size(2), Cout = Y.size(3);
   TORCH_CHECK(Cin % 8 == 0, "Cin must be a multiple of 8 (pad first)");
@@ -76,7 +90,7 @@ void conv2d_bf16(torch::Tensor X, torch::Tensor W2, torch::Tensor Y,
   }
   int M = Nimg * HO * WO;
   int K = (int)(R * S) * Cin;
-  int sk = splitk_plan(M, Cout, K);
+  int sk = ldc == Cout ? splitk_plan(M, Cout, K) : 1;
   torch::Tensor ws;
   float* wptr = nullptr;
   if (sk > 1) {
@@ -86,8 +100,8 @@ void conv2d_bf16(torch::Tensor X, torch::Tensor W2, torch::Tensor Y,
   }
   launch_conv2d_bf16(X.data_ptr(), W2.data_ptr(), Y.data_ptr(), bptr, rptr,
                      Nimg, H, W, Cin, (int)R, (int)S, Cout, (int)sh, (int)sw,
-                     (int)pt, (int)pl, HO, WO, (int)W2.size(1), relu,
-                     cur_stream(), wptr, sk);
+                     (int)pt, (int)pl, HO, WO, (int)W2.size(1), (int)act,
+                     cur_stream(), wptr, sk, ldc);
 }
 
 void maxpool_bf16(torch::Tensor X, torch::Tensor Y, int64_t kh, int64_t kw,
@@ -120,6 +134,13 @@ void softmax_bf16(torch::Tensor X, torch::Tensor Y) {
   TORCH_CHECK(Y.scalar_type() == torch::kFloat32);
   launch_softmax_bf16(X.data_ptr(), Y.data_ptr<float>(), X.size(0), X.size(1),
                       cur_stream());
+}
+
+void softmax_bf16_out(torch::Tensor X, torch::Tensor Y) {
+  check_bf16(X, "X");
+  check_bf16(Y, "Y");
+  launch_softmax_bf16_out(X.data_ptr(), Y.data_ptr(), X.size(0), X.size(1),
+                          cur_stream());
 }
 
 void cast_f32_bf16(torch::Tensor X, torch::Tensor Y) {
@@ -245,6 +266,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("avgpool_bf16", &avgpool_bf16);
   m.def("global_mean_bf16", &global_mean_bf16);
   m.def("softmax_bf16", &softmax_bf16);
+  m.def("softmax_bf16_out", &softmax_bf16_out);
   m.def("cast_f32_bf16", &cast_f32_bf16);
   m.def("cast_bf16_f32", &cast_bf16_f32);
   m.def("channel_pad_bf16", &channel_pad_bf16);

@@ -85,7 +85,8 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
                       int relu, ConvParams cp,
                       int64_t sA = 0, int64_t sB = 0, int64_t sC = 0,
                       float* __restrict__ Cpart = nullptr,
-                      int nt_slice = 1 << 28) {
+                      int nt_slice = 1 << 28, int ldc = 0) {
+  if (ldc == 0) ldc = N;  // C row stride (concat-slice / fused-qkv writes)
   constexpr int NWAVES = (TBM / 64) * (TBN / 64);
   constexpr int WC_N = TBN / 64;            // wave columns
   constexpr int CA = TBM / 8 / NWAVES;      // A-staging glds per wave
@@ -342,7 +343,8 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
       const int nvalid = min(8, N - col_g);
       if (HAS_RES) {
         if (nvalid == 8) {
-          bf16x8 rv = *(const bf16x8*)(residual + (int64_t)row_g * N + col_g);
+          bf16x8 rv = *(const bf16x8*)(residual + (int64_t)row_g * ldc +
+                                       col_g);
 #pragma unroll
           for (int e = 0; e < 8; ++e)
             v[e] = (__bf16)(bf2f(v[e]) + bf2f(rv[e]));
@@ -351,20 +353,27 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
           for (int e = 0; e < 8; ++e)
             if (e < nvalid)
               v[e] = (__bf16)(bf2f(v[e]) +
-                              bf2f(residual[(int64_t)row_g * N + col_g + e]));
+                              bf2f(residual[(int64_t)row_g * ldc + col_g +
+                                            e]));
         }
       }
-      if (relu) {
+      if (relu == 1) {
 #pragma unroll
         for (int e = 0; e < 8; ++e)
           v[e] = bf2f(v[e]) > 0.f ? v[e] : (__bf16)0.f;
+      } else if (relu == 5) {  // erf-GELU fused epilogue (ViT MLP)
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          float f = bf2f(v[e]);
+          v[e] = (__bf16)(0.5f * f * (1.f + erff(f * 0.70710678f)));
+        }
       }
       if (nvalid == 8) {
-        *(bf16x8*)(Cout + (int64_t)row_g * N + col_g) = v;
+        *(bf16x8*)(Cout + (int64_t)row_g * ldc + col_g) = v;
       } else {
 #pragma unroll
         for (int e = 0; e < 8; ++e)
-          if (e < nvalid) Cout[(int64_t)row_g * N + col_g + e] = v[e];
+          if (e < nvalid) Cout[(int64_t)row_g * ldc + col_g + e] = v[e];
       }
     }
   }
@@ -394,7 +403,8 @@ void gemm_conv_kernel_256(const __bf16* __restrict__ A,
                           const __bf16* __restrict__ residual,
                           int M, int N, int K, int Kp,
                           int a_bytes, int tiles_m, int tiles_n,
-                          int relu, ConvParams cp) {
+                          int relu, ConvParams cp, int ldc = 0) {
+  if (ldc == 0) ldc = N;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = __builtin_amdgcn_readfirstlane(tid >> 6);
@@ -597,7 +607,8 @@ void gemm_conv_kernel_256(const __bf16* __restrict__ A,
       const int nvalid = min(8, N - col_g);
       if (HAS_RES) {
         if (nvalid == 8) {
-          bf16x8 rv = *(const bf16x8*)(residual + (int64_t)row_g * N + col_g);
+          bf16x8 rv = *(const bf16x8*)(residual + (int64_t)row_g * ldc +
+                                       col_g);
 #pragma unroll
           for (int e = 0; e < 8; ++e)
             v[e] = (__bf16)(bf2f(v[e]) + bf2f(rv[e]));
@@ -606,20 +617,27 @@ void gemm_conv_kernel_256(const __bf16* __restrict__ A,
           for (int e = 0; e < 8; ++e)
             if (e < nvalid)
               v[e] = (__bf16)(bf2f(v[e]) +
-                              bf2f(residual[(int64_t)row_g * N + col_g + e]));
+                              bf2f(residual[(int64_t)row_g * ldc + col_g +
+                                            e]));
         }
       }
-      if (relu) {
+      if (relu == 1) {
 #pragma unroll
         for (int e = 0; e < 8; ++e)
           v[e] = bf2f(v[e]) > 0.f ? v[e] : (__bf16)0.f;
+      } else if (relu == 5) {  // erf-GELU fused epilogue (ViT MLP)
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          float f = bf2f(v[e]);
+          v[e] = (__bf16)(0.5f * f * (1.f + erff(f * 0.70710678f)));
+        }
       }
       if (nvalid == 8) {
-        *(bf16x8*)(Cout + (int64_t)row_g * N + col_g) = v;
+        *(bf16x8*)(Cout + (int64_t)row_g * ldc + col_g) = v;
       } else {
 #pragma unroll
         for (int e = 0; e < 8; ++e)
-          if (e < nvalid) Cout[(int64_t)row_g * N + col_g + e] = v[e];
+          if (e < nvalid) Cout[(int64_t)row_g * ldc + col_g + e] = v[e];
       }
     }
   }
@@ -664,8 +682,12 @@ __global__ void splitk_combine_kernel(const float* __restrict__ Cpart,
     }
 #pragma unroll
     for (int e = 0; e < 8; ++e)
-      if (e < rem)
-        Cout[i + e] = (__bf16)(relu && v[e] < 0.f ? 0.f : v[e]);
+      if (e < rem) {
+        float f = v[e];
+        if (relu == 1 && f < 0.f) f = 0.f;
+        else if (relu == 5) f = 0.5f * f * (1.f + erff(f * 0.70710678f));
+        Cout[i + e] = (__bf16)f;
+      }
   }
 }
 
@@ -683,9 +705,9 @@ static bool use_256(int M, int N, int K) {
 template <int TBM, int TBN, int MODE, int DEPTH = 2>
 static void launch_tile(const void* A, const void* B, void* C,
                         const float* bias, const void* residual, int M,
-                        int N, int K, int Kp, int64_t a_bytes, bool relu,
+                        int N, int K, int Kp, int64_t a_bytes, int act,
                         const ConvParams& cp, hipStream_t stream,
-                        float* ws = nullptr, int splitk = 1) {
+                        float* ws = nullptr, int splitk = 1, int ldc = 0) {
   constexpr int THREADS = (TBM / 64) * (TBN / 64) * 64;
   constexpr int LDS = DEPTH * (TBM + TBN) * BK * 2;
   constexpr int LDS_TOTAL = LDS > TBM * TBN * 2 ? LDS : TBM * TBN * 2;
@@ -700,12 +722,12 @@ static void launch_tile(const void* A, const void* B, void* C,
                        (const __bf16*)B, (__bf16*)C, nullptr,
                        (const __bf16*)nullptr, M, N, K, Kp, (int)a_bytes,
                        tiles_m, tiles_n, 0, cp, (int64_t)0, (int64_t)0,
-                       (int64_t)0, ws, nt_slice);
+                       (int64_t)0, ws, nt_slice, 0);
     int64_t MN = (int64_t)M * N;
     int blocks = (int)std::min<int64_t>((MN / 8 + 255) / 256, 2048);
     hipLaunchKernelGGL(splitk_combine_kernel, dim3(blocks), dim3(256), 0,
                        stream, ws, bias, (const __bf16*)residual,
-                       (__bf16*)C, MN, N, splitk, relu ? 1 : 0);
+                       (__bf16*)C, MN, N, splitk, act);
     return;
   }
   dim3 grid(tiles_m * tiles_n);
@@ -714,15 +736,15 @@ static void launch_tile(const void* A, const void* B, void* C,
                        dim3(THREADS), LDS_TOTAL, stream, (const __bf16*)A,
                        (const __bf16*)B, (__bf16*)C, bias,
                        (const __bf16*)residual, M, N, K, Kp, (int)a_bytes,
-                       tiles_m, tiles_n, relu ? 1 : 0, cp, (int64_t)0,
-                       (int64_t)0, (int64_t)0);
+                       tiles_m, tiles_n, act, cp, (int64_t)0,
+                       (int64_t)0, (int64_t)0, nullptr, 1 << 28, ldc);
   else
     hipLaunchKernelGGL((gemm_conv_kernel<TBM, TBN, MODE, false, DEPTH>), grid,
                        dim3(THREADS), LDS_TOTAL, stream, (const __bf16*)A,
                        (const __bf16*)B, (__bf16*)C, bias,
                        (const __bf16*)nullptr, M, N, K, Kp, (int)a_bytes,
-                       tiles_m, tiles_n, relu ? 1 : 0, cp, (int64_t)0,
-                       (int64_t)0, (int64_t)0);
+                       tiles_m, tiles_n, act, cp, (int64_t)0,
+                       (int64_t)0, (int64_t)0, nullptr, 1 << 28, ldc);
 }
 
 // How many K-slices should this shape use? (1 = no split)
@@ -747,9 +769,9 @@ int splitk_plan(int M, int N, int K) {
 template <int MODE>
 static void dispatch(const void* A, const void* B, void* C, const float* bias,
                      const void* residual, int M, int N, int K, int Kp,
-                     int64_t a_bytes, bool relu, const ConvParams& cp,
+                     int64_t a_bytes, int act, const ConvParams& cp,
                      hipStream_t stream, float* ws = nullptr,
-                     int splitk = 1) {
+                     int splitk = 1, int ldc = 0) {
   if (use_256(M, N, K)) {
     int tiles_m = (M + BM2 - 1) / BM2;
     int tiles_n = (N + BN2 - 1) / BN2;
@@ -759,13 +781,13 @@ static void dispatch(const void* A, const void* B, void* C, const float* bias,
                          dim3(NTHREADS2), LDS2_BYTES, stream,
                          (const __bf16*)A, (const __bf16*)B, (__bf16*)C, bias,
                          (const __bf16*)residual, M, N, K, Kp, (int)a_bytes,
-                         tiles_m, tiles_n, relu ? 1 : 0, cp);
+                         tiles_m, tiles_n, act, cp, ldc);
     else
       hipLaunchKernelGGL((gemm_conv_kernel_256<MODE, false>), grid,
                          dim3(NTHREADS2), LDS2_BYTES, stream,
                          (const __bf16*)A, (const __bf16*)B, (__bf16*)C, bias,
                          (const __bf16*)nullptr, M, N, K, Kp, (int)a_bytes,
-                         tiles_m, tiles_n, relu ? 1 : 0, cp);
+                         tiles_m, tiles_n, act, cp, ldc);
     return;
   }
   // DEPTH=3 measured WORSE here (120 KiB LDS -> 1 block/CU, losing the
@@ -773,10 +795,11 @@ static void dispatch(const void* A, const void* B, void* C, const float* bias,
   // stem 537->351 TF, s1 3x3 500->325. Keep DEPTH=2 + 2 blocks/CU.
   if (N <= 96)
     launch_tile<256, 64, MODE, 2>(A, B, C, bias, residual, M, N, K, Kp,
-                                  a_bytes, relu, cp, stream, ws, splitk);
+                                  a_bytes, act, cp, stream, ws, splitk, ldc);
   else
     launch_tile<128, 128, MODE, 2>(A, B, C, bias, residual, M, N, K, Kp,
-                                   a_bytes, relu, cp, stream, ws, splitk);
+                                   a_bytes, act, cp, stream, ws, splitk,
+                                   ldc);
 }
 
 void launch_gemm_batched_bf16(const void* A, const void* B, void* C,
@@ -797,19 +820,19 @@ void launch_gemm_batched_bf16(const void* A, const void* B, void* C,
 
 void launch_gemm_bf16(const void* A, const void* B, void* C,
                       const float* bias, const void* residual, int M, int N,
-                      int K, int Kp, int64_t a_bytes, bool relu,
-                      hipStream_t stream, float* ws, int splitk) {
+                      int K, int Kp, int64_t a_bytes, int act,
+                      hipStream_t stream, float* ws, int splitk, int ldc) {
   ConvParams cp{};
-  dispatch<0>(A, B, C, bias, residual, M, N, K, Kp, a_bytes, relu, cp,
-              stream, ws, splitk);
+  dispatch<0>(A, B, C, bias, residual, M, N, K, Kp, a_bytes, act, cp,
+              stream, ws, splitk, ldc);
 }
 
 void launch_conv2d_bf16(const void* X, const void* B, void* C,
                         const float* bias, const void* residual, int Nimg,
                         int H, int W, int Cin, int R, int S, int Cout, int sh,
                         int sw, int pt, int pl, int HO, int WO, int Kp,
-                        bool relu, hipStream_t stream, float* ws,
-                        int splitk) {
+                        int act, hipStream_t stream, float* ws,
+                        int splitk, int ldc) {
   int M = Nimg * HO * WO;
   int K = R * S * Cin;
   ConvParams cp;
@@ -820,6 +843,6 @@ void launch_conv2d_bf16(const void* X, const void* B, void* C,
   cp.magic_HW = magic40(HO * WO);
   cp.magic_WO = magic40(WO);
   int64_t a_bytes = (int64_t)Nimg * H * W * Cin * 2;
-  dispatch<1>(X, B, C, bias, residual, M, Cout, K, Kp, a_bytes, relu, cp,
-              stream, ws, splitk);
+  dispatch<1>(X, B, C, bias, residual, M, Cout, K, Kp, a_bytes, act, cp,
+              stream, ws, splitk, ldc);
 }

@@ -5,17 +5,18 @@
 
 // gemm_conv.hip — C = act(A @ B^T + bias [+ residual]), bf16 in/out, f32 acc
 int splitk_plan(int M, int N, int K);
+// act: 0 none, 1 relu, 5 gelu; ldc: C row stride (0 = N)
 void launch_gemm_bf16(const void* A, const void* B, void* C,
                       const float* bias, const void* residual, int M, int N,
-                      int K, int Kp, int64_t a_bytes, bool relu,
+                      int K, int Kp, int64_t a_bytes, int act,
                       hipStream_t stream, float* ws = nullptr,
-                      int splitk = 1);
+                      int splitk = 1, int ldc = 0);
 void launch_conv2d_bf16(const void* X, const void* B, void* C,
                         const float* bias, const void* residual, int Nimg,
                         int H, int W, int Cin, int R, int S, int Cout, int sh,
                         int sw, int pt, int pl, int HO, int WO, int Kp,
-                        bool relu, hipStream_t stream, float* ws = nullptr,
-                        int splitk = 1);
+                        int act, hipStream_t stream, float* ws = nullptr,
+                        int splitk = 1, int ldc = 0);
 
 // pool_reduce.hip
 void launch_maxpool_bf16(const void* X, void* Y, int Nimg, int H, int W,
@@ -28,6 +29,8 @@ void launch_global_mean_bf16(const void* X, void* Y, int Nimg, int HW, int C,
                              hipStream_t stream);
 void launch_softmax_bf16(const void* X, float* Y, int M, int N,
                          hipStream_t stream);
+void launch_softmax_bf16_out(const void* X, void* Y, int M, int N,
+                             hipStream_t stream);
 
 // layernorm.hip
 void launch_layernorm_bf16(const void* X, const float* gamma,

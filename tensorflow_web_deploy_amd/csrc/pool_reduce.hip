@@ -116,8 +116,9 @@ void launch_global_mean_bf16(const void* X, void* Y, int Nimg, int HW, int C,
 
 // --------------------------------------------------------------- softmax
 
+template <typename OUT>
 __global__ void softmax_kernel(const __bf16* __restrict__ X,
-                               float* __restrict__ Y, int M, int N) {
+                               OUT* __restrict__ Y, int M, int N) {
   // one wave per row (classes ~1e3); wave-wide shuffle reduce
   int row = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   int lane = threadIdx.x & 63;
@@ -132,14 +133,23 @@ __global__ void softmax_kernel(const __bf16* __restrict__ X,
 #pragma unroll
   for (int off = 32; off; off >>= 1) s += __shfl_xor(s, off, 64);
   float inv = 1.f / s;
-  float* y = Y + (int64_t)row * N;
-  for (int i = lane; i < N; i += 64) y[i] = __expf(bf2f(x[i]) - m) * inv;
+  OUT* y = Y + (int64_t)row * N;
+  for (int i = lane; i < N; i += 64)
+    y[i] = (OUT)(__expf(bf2f(x[i]) - m) * inv);
 }
 
 void launch_softmax_bf16(const void* X, float* Y, int M, int N,
                          hipStream_t stream) {
   int wpb = 4;  // 4 waves (rows) per block
   int blocks = (M + wpb - 1) / wpb;
-  hipLaunchKernelGGL(softmax_kernel, dim3(blocks), dim3(wpb * 64), 0, stream,
-                     (const __bf16*)X, Y, M, N);
+  hipLaunchKernelGGL((softmax_kernel<float>), dim3(blocks), dim3(wpb * 64),
+                     0, stream, (const __bf16*)X, Y, M, N);
+}
+
+void launch_softmax_bf16_out(const void* X, void* Y, int M, int N,
+                             hipStream_t stream) {
+  int wpb = 4;
+  int blocks = (M + wpb - 1) / wpb;
+  hipLaunchKernelGGL((softmax_kernel<__bf16>), dim3(blocks), dim3(wpb * 64),
+                     0, stream, (const __bf16*)X, (__bf16*)Y, M, N);
 }

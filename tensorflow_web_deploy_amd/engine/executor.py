@@ -123,6 +123,19 @@ def _dense(node, ins, consts):
     return y
 
 
+@cpu_op("DenseND")
+def _dense_nd(node, ins, consts):
+    y = torch.matmul(ins[0], ins[1])
+    if node.attrs.get("bias"):
+        y = y + ins[2]
+    act = node.attrs.get("act")
+    if act == "relu":
+        y = F.relu(y)
+    elif act == "gelu":
+        y = F.gelu(y)
+    return y
+
+
 @cpu_op("Conv2D")
 def _conv_raw(node, ins, consts):
     return _conv2d(ins[0], ins[1], node.attrs["strides"],

@@ -150,6 +150,40 @@ class GpuExecutor:
             shapes[name] = tuple(shape)
             return t
 
+        # ---- concat elimination (Inception): when every member of a
+        # channel ConcatV2 is a ConvBlock consumed only by the concat,
+        # each conv writes its channel-slice of the concat buffer
+        # directly (ldc-strided epilogue) and the cat copy disappears.
+        prods = {n.name: n for n in g.nodes}
+        n_consumers: Dict[str, int] = {}
+        for n in g.nodes:
+            for i in n.inputs:
+                n_consumers[i] = n_consumers.get(i, 0) + 1
+        concat_slice: Dict[str, tuple] = {}   # member -> (cat, c0, w)
+        concat_fused: Dict[str, int] = {}     # cat name -> total C
+        for n in g.nodes:
+            if n.op != "ConcatV2":
+                continue
+            members = n.inputs[:-1]
+            axis = int(self.consts[n.inputs[-1]].reshape(-1)[0])
+            if axis not in (-1, 3):
+                continue
+            ok = all(m in prods and prods[m].op == "ConvBlock"
+                     and n_consumers.get(m, 0) == 1
+                     and not prods[m].attrs.get("residual")
+                     for m in members)
+            if not ok:
+                continue
+            widths = [self.consts[prods[m].inputs[1]].shape[3]
+                      for m in members]
+            if any(w % 8 for w in widths):
+                continue
+            c0 = 0
+            for m, w in zip(members, widths):
+                concat_slice[m] = (n.name, c0, w)
+                c0 += w
+            concat_fused[n.name] = c0
+
         for node in g.nodes:
             op = node.op
             if op in ("ConvBlock",):
@@ -165,7 +199,15 @@ class GpuExecutor:
                 else:
                     HO, WO = _valid_out(H, W, R, S, sh, sw)
                     pt = pl = 0
-                Y = alloc(node.name, (N_, HO, WO, O))
+                if node.name in concat_slice:
+                    cat, c0, wdt = concat_slice[node.name]
+                    if cat not in bufs:
+                        alloc(cat, (N_, HO, WO, concat_fused[cat]))
+                    Y = bufs[cat][..., c0:c0 + wdt]
+                    bufs[node.name] = Y
+                    shapes[node.name] = (N_, HO, WO, O)
+                else:
+                    Y = alloc(node.name, (N_, HO, WO, O))
                 bias = self._bias(node.attrs.get("bias"), O)
                 res = bufs[node.attrs["residual"]] if node.attrs.get("residual") else None
                 relu = bool(node.attrs.get("relu"))
@@ -174,7 +216,8 @@ class GpuExecutor:
                                      device=dev)
                     steps.append(lambda X=X, Xp=Xp: E.channel_pad_bf16(X, Xp))
                     X = Xp
-                if R == 1 and S == 1 and sh == 1 and sw == 1:
+                if (R == 1 and S == 1 and sh == 1 and sw == 1
+                        and Y.is_contiguous()):
                     M = N_ * HO * WO
                     A2 = X.view(M, Cp)
                     Y2 = Y.view(M, O)
@@ -264,14 +307,12 @@ class GpuExecutor:
                     Y = alloc(node.name, (M, N), torch.float32)
                     steps.append(lambda X=X, Y=Y: E.softmax_bf16(X, Y))
                 else:
-                    # attention probs: keep bf16 for the downstream GEMM
+                    # attention probs: bf16 in -> bf16 out directly
                     Y = alloc(node.name, sh_in)
-                    Yf = torch.zeros((M, N), dtype=torch.float32, device=dev)
                     Yv = Y.view(M, N)
                     Xv = X.view(M, N)
-                    steps.append(lambda Xv=Xv, Yf=Yf, Yv=Yv:
-                                 (E.softmax_bf16(Xv, Yf),
-                                  E.cast_f32_bf16(Yf, Yv)))
+                    steps.append(lambda Xv=Xv, Yv=Yv:
+                                 E.softmax_bf16_out(Xv, Yv))
             elif op == "LayerNorm":
                 xn = node.inputs[0]
                 gamma = self._bias(node.inputs[1], 0)
@@ -285,6 +326,22 @@ class GpuExecutor:
                 Y = alloc(node.name, shapes[xn])
                 steps.append(lambda X=bufs[xn], Y=Y:
                              E.bias_act_bf16(X, None, Y, ACT_GELU))
+            elif op == "DenseND":
+                xn = node.inputs[0]
+                sa = shapes[xn]
+                wt, K, N, Kp, Np = self._prep_dense_weight(node.inputs[1],
+                                                           False)
+                assert Np == N, "DenseND weight N must be %8"
+                M = int(np.prod(sa[:-1]))
+                bias = self._bias(node.attrs.get("bias"), N)
+                act = {None: 0, "relu": ACT_RELU, "gelu": ACT_GELU}[
+                    node.attrs.get("act")]
+                Y = alloc(node.name, (*sa[:-1], N))
+                A2 = bufs[xn].view(M, K)
+                Y2 = Y.view(M, N)
+                steps.append(lambda A2=A2, wt=wt, Y2=Y2, bias=bias, K=K,
+                             act=act:
+                             E.gemm_bf16(A2, wt, Y2, bias, None, K, act))
             elif op in ("BatchMatMul", "BatchMatMulV2"):
                 an, bn = node.inputs
                 sa = shapes[an]
@@ -428,6 +485,11 @@ class GpuExecutor:
                 Y = alloc(node.name, shapes[xn])
                 steps.append(lambda X=bufs[xn], Y=Y, sc=sc, sh_=sh_:
                              E.scale_shift_bf16(X, sc, sh_, Y, False))
+            elif op == "ConcatV2" and node.name in concat_fused:
+                # members already wrote their slices; the buffer IS the
+                # concat result (shape was registered at first member)
+                N_, HO, WO, _ = shapes[node.inputs[0]]
+                shapes[node.name] = (N_, HO, WO, concat_fused[node.name])
             elif op == "ConcatV2":
                 ins = node.inputs[:-1]
                 axis = int(self.consts[node.inputs[-1]].reshape(-1)[0])

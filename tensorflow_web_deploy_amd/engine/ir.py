@@ -371,6 +371,61 @@ def fuse_gelu(graph: Graph) -> Graph:
     return Graph(new_nodes, graph.consts, graph.inputs, graph.outputs)
 
 
+def fuse_dense_bmm(graph: Graph) -> Graph:
+    """Fuse BatchMatMulV2(x, W[2D const]) [+ AddV2 bias] [+ Gelu|Relu]
+    -> DenseND (rows-flattened GEMM with fused bias/activation epilogue).
+    This is the ViT projection/MLP pattern (bias and GELU were separate
+    kernel launches: measured 12.5% of the ViT forward)."""
+    consumers: Dict[str, List[Node]] = {}
+    for n in graph.nodes:
+        for i in n.inputs:
+            consumers.setdefault(i, []).append(n)
+    out_set = set(graph.outputs)
+
+    def sole_consumer(name):
+        cs = consumers.get(name, [])
+        return cs[0] if len(cs) == 1 and name not in out_set else None
+
+    drop = set()
+    created = set()
+    new_nodes: List[Node] = []
+    for n in graph.nodes:
+        if n.name in drop:
+            continue
+        if (n.op in ("BatchMatMul", "BatchMatMulV2")
+                and n.inputs[1] in graph.consts
+                and graph.consts[n.inputs[1]].ndim == 2
+                and not n.attrs.get("adj_x") and not n.attrs.get("adj_y")):
+            bias = None
+            act = None
+            tail = n
+            c = sole_consumer(tail.name)
+            if (c is not None and c.op in ("Add", "AddV2")
+                    and c.inputs[0] == tail.name
+                    and c.inputs[1] in graph.consts
+                    and graph.consts[c.inputs[1]].ndim <= 1):
+                bias = c.inputs[1]
+                drop.add(c.name)
+                tail = c
+                c = sole_consumer(tail.name)
+            if c is not None and c.op in ("Relu", "Gelu"):
+                act = c.op.lower()
+                drop.add(c.name)
+                tail = c
+            ins = [n.inputs[0], n.inputs[1]]
+            if bias:
+                ins.append(bias)
+            nd = Node(tail.name, "DenseND", ins,
+                      {"bias": bias, "act": act})
+            created.add(id(nd))
+            new_nodes.append(nd)
+        else:
+            new_nodes.append(n)
+    new_nodes = [x for x in new_nodes
+                 if id(x) in created or x.name not in drop]
+    return Graph(new_nodes, graph.consts, graph.inputs, graph.outputs)
+
+
 def toposort(graph: Graph) -> Graph:
     nmap = graph.node_map()
     order: List[Node] = []
@@ -407,6 +462,7 @@ def optimize(graph: Graph) -> Graph:
     g = fold_batchnorm(g)
     g = fuse_layernorm(g)
     g = fuse_gelu(g)
+    g = fuse_dense_bmm(g)
     g = fuse_conv_blocks(g)
     g = toposort(g)
     return g
