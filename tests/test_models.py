@@ -29,9 +29,10 @@ def test_vit_b16():
     eng, ops = _run("vit_b16", 224, 224, 3, 1000)
     # the primitive-op graph must fully fuse
     assert ops["LayerNorm"] == 25   # 2/layer + final
-    assert ops["Gelu"] == 12
+    assert ops["DenseND"] == 72     # qkv/proj/mlp with fused bias+act
     assert ops["Softmax"] == 13     # 12 attention + classifier
     assert "SquaredDifference" not in ops and "Erf" not in ops
+    assert "Gelu" not in ops        # absorbed into DenseND epilogue
 
 
 def test_vit_layernorm_fusion_equivalence():
