@@ -257,6 +257,24 @@ void resize_norm_batch(torch::Tensor arena, torch::Tensor meta,
                            cur_stream());
 }
 
+// Attention GEMM on 4D views: every tensor supplies its own strides.
+// A: [B1, M, ...] with k contiguous; B: [B1, N-rows ...]; C likewise.
+// Python passes logical dims + strides explicitly (views, zero-copy).
+void gemm_strided_bf16(torch::Tensor A, torch::Tensor B, torch::Tensor C,
+                       int64_t batch1, int64_t batch2, int64_t M, int64_t N,
+                       int64_t K, int64_t lda, int64_t ldb, int64_t ldc,
+                       int64_t sA1, int64_t sA2, int64_t sB1, int64_t sB2,
+                       int64_t sC1, int64_t sC2) {
+  TORCH_CHECK(A.is_cuda() && A.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(B.is_cuda() && B.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(C.is_cuda() && C.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(K % 8 == 0, "strided gemm K must be %8");
+  launch_gemm_strided_bf16(A.data_ptr(), B.data_ptr(), C.data_ptr(),
+                           (int)batch1, (int)batch2, (int)M, (int)N, (int)K,
+                           (int)lda, (int)ldb, (int)ldc, sA1, sA2, sB1, sB2,
+                           sC1, sC2, A.numel(), cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -276,6 +294,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scale_shift_bf16", &scale_shift_bf16);
   m.def("layernorm_bf16", &layernorm_bf16);
   m.def("gemm_batched_bf16", &gemm_batched_bf16);
+  m.def("gemm_strided_bf16", &gemm_strided_bf16);
   m.def("resize_norm", &resize_norm);
   m.def("resize_norm_batch", &resize_norm_batch);
 }

@@ -85,21 +85,31 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
                       int relu, ConvParams cp,
                       int64_t sA = 0, int64_t sB = 0, int64_t sC = 0,
                       float* __restrict__ Cpart = nullptr,
-                      int nt_slice = 1 << 28, int ldc = 0) {
+                      int nt_slice = 1 << 28, int ldc = 0,
+                      int lda = 0, int ldb = 0, int batch2 = 1,
+                      int64_t sA2 = 0, int64_t sB2 = 0, int64_t sC2 = 0) {
   if (ldc == 0) ldc = N;  // C row stride (concat-slice / fused-qkv writes)
+  if (lda == 0) lda = K;  // A row stride (strided-batch attention views)
+  if (ldb == 0) ldb = Kp;
   constexpr int NWAVES = (TBM / 64) * (TBN / 64);
   constexpr int WC_N = TBN / 64;            // wave columns
   constexpr int CA = TBM / 8 / NWAVES;      // A-staging glds per wave
   constexpr int CB = TBN / 8 / NWAVES;      // B-staging glds per wave
   constexpr int ATILE = TBM * BK * 2;       // bytes
   constexpr int BTILE = TBN * BK * 2;
-  // strided-batch support (ViT attention: blockIdx.y = batch index)
+  // strided-batch support (attention: blockIdx.y = flattened (b1,b2))
   {
-    const int64_t bz = blockIdx.y;
-    A += bz * sA;
-    B += bz * sB;
-    Cout += bz * sC;
-    if (HAS_RES) residual += bz * sC;
+    const int bz = blockIdx.y;
+    const int b1 = bz / batch2;
+    const int b2 = bz - b1 * batch2;
+    const int64_t offA = b1 * sA + b2 * sA2;
+    const int64_t offB = b1 * sB + b2 * sB2;
+    const int64_t offC = b1 * sC + b2 * sC2;
+    A += offA;
+    B += offB;
+    Cout += offC;
+    if (HAS_RES) residual += offC;
+    a_bytes -= (int)(offA * 2);
   }
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -122,7 +132,7 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
   const auto rsrcA = __builtin_amdgcn_make_buffer_rsrc(
       (void*)A, /*stride*/ 0, a_bytes, 0x00020000);
   const auto rsrcB = __builtin_amdgcn_make_buffer_rsrc(
-      (void*)B, 0, (int)((int64_t)N * Kp * 2), 0x00020000);
+      (void*)B, 0, (int)(((int64_t)(N - 1) * ldb + Kp) * 2), 0x00020000);
 
   // ---- per-lane staging state ----
   // Staging instruction j covers tile rows j*8 + lane/8, lane slot =
@@ -140,7 +150,7 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
     int row_g = row0 + row_l;
     if (MODE == 0) {
       a_valid[i] = row_g < M;
-      a_base[i] = row_g * K;
+      a_base[i] = row_g * lda;
     } else {
       a_valid[i] = row_g < M;
       uint32_t rg = (uint32_t)min(row_g, max(M - 1, 0));
@@ -164,7 +174,7 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
     int row_l = j * 8 + srow_base;
     int col_g = col0 + row_l;
     b_valid[i] = col_g < N;
-    b_voff[i] = col_g * Kp;
+    b_voff[i] = col_g * ldb;
   }
 
   const int swz = srow_base & 7;  // row&7 of this lane's staged row
@@ -802,6 +812,25 @@ static void dispatch(const void* A, const void* B, void* C, const float* bias,
                                    ldc);
 }
 
+void launch_gemm_strided_bf16(const void* A, const void* B, void* C,
+                              int batch1, int batch2, int M, int N, int K,
+                              int lda, int ldb, int ldc, int64_t sA1,
+                              int64_t sA2, int64_t sB1, int64_t sB2,
+                              int64_t sC1, int64_t sC2, int64_t a_total,
+                              hipStream_t stream) {
+  ConvParams cp{};
+  int tiles_m = (M + BM - 1) / BM;
+  int tiles_n = (N + BN - 1) / BN;
+  dim3 grid(tiles_m * tiles_n, batch1 * batch2);
+  hipLaunchKernelGGL((gemm_conv_kernel<128, 128, 0, false, 2>), grid,
+                     dim3(NTHREADS), LDS_BYTES, stream, (const __bf16*)A,
+                     (const __bf16*)B, (__bf16*)C, nullptr,
+                     (const __bf16*)nullptr, M, N, K, K,
+                     (int)(a_total * 2), tiles_m, tiles_n, 0, cp,
+                     sA1, sB1, sC1, nullptr, 1 << 28, ldc, lda, ldb,
+                     batch2, sA2, sB2, sC2);
+}
+
 void launch_gemm_batched_bf16(const void* A, const void* B, void* C,
                               int batch, int M, int N, int K, int Kp,
                               int64_t sA, int64_t sB, int64_t sC,
@@ -814,8 +843,8 @@ void launch_gemm_batched_bf16(const void* A, const void* B, void* C,
                      dim3(NTHREADS), LDS_BYTES, stream, (const __bf16*)A,
                      (const __bf16*)B, (__bf16*)C, nullptr,
                      (const __bf16*)nullptr, M, N, K, Kp,
-                     (int)(M * (int64_t)K * 2), tiles_m, tiles_n, 0, cp,
-                     sA, sB, sC);
+                     (int)(batch * M * (int64_t)K * 2), tiles_m, tiles_n, 0,
+                     cp, sA, sB, sC);
 }
 
 void launch_gemm_bf16(const void* A, const void* B, void* C,

@@ -37,6 +37,16 @@ void launch_layernorm_bf16(const void* X, const float* gamma,
                            const float* beta, void* Y, int64_t rows, int C,
                            float eps, hipStream_t stream);
 
+// fully-strided batched GEMM: C[b1,b2] = A[b1,b2] @ B[b1,b2]^T with
+// independent row strides (lda/ldb/ldc) and 2D batch strides — lets
+// attention run directly on [N, tokens, heads, dim] views (no transposes)
+void launch_gemm_strided_bf16(const void* A, const void* B, void* C,
+                              int batch1, int batch2, int M, int N, int K,
+                              int lda, int ldb, int ldc, int64_t sA1,
+                              int64_t sA2, int64_t sB1, int64_t sB2,
+                              int64_t sC1, int64_t sC2, int64_t a_total,
+                              hipStream_t stream);
+
 // gemm_conv.hip (strided-batch, MODE 0, no epilogue fusion)
 void launch_gemm_batched_bf16(const void* A, const void* B, void* C,
                               int batch, int M, int N, int K, int Kp,

@@ -159,6 +159,10 @@ class GpuExecutor:
         for n in g.nodes:
             for i in n.inputs:
                 n_consumers[i] = n_consumers.get(i, 0) + 1
+        consumer_nodes: Dict[str, list] = {}
+        for n in g.nodes:
+            for i in n.inputs:
+                consumer_nodes.setdefault(i, []).append(n)
         concat_slice: Dict[str, tuple] = {}   # member -> (cat, c0, w)
         concat_fused: Dict[str, int] = {}     # cat name -> total C
         for n in g.nodes:
@@ -358,50 +362,99 @@ class GpuExecutor:
                                              False))
                 else:
                     sb = shapes[bn]
-                    batch = int(np.prod(sa[:-2]))
                     M, K = sa[-2], sa[-1]
                     adj_y = bool(node.attrs.get("adj_y"))
-                    A3 = bufs[an].view(batch, M, K)
-                    if adj_y:
-                        # scores = Q @ Kmat^T: B rows are [N][K] already
-                        Nn = sb[-2]
-                        B3 = bufs[bn].view(batch, Nn, K)
-                        Kp8 = K
-                        if K % 8:
-                            raise NotImplementedError("adj_y K%8")
-                        Y = alloc(node.name, (*sa[:-2], M, Nn))
-                        Y3 = Y.view(batch, M, Nn)
-                        steps.append(lambda A3=A3, B3=B3, Y3=Y3:
-                                     E.gemm_batched_bf16(A3, B3, Y3))
+                    At = bufs[an]
+                    Bt_in = bufs[bn]
+                    # normalize to 4D [b1, b2, M, K] (b2=1 for 3D)
+                    if At.dim() == 3:
+                        At = At[:, None]
+                        Bt_in = Bt_in[:, None]
+                        sa4 = (sa[0], 1, *sa[1:])
+                        sb4 = (sb[0], 1, *sb[1:])
                     else:
-                        # P @ V: transpose V to [N][K] layout and zero-pad
-                        # K to a multiple of 8 (exact: pads contribute 0)
-                        Nn = sb[-1]
+                        sa4, sb4 = sa, sb
+                    b1, b2 = sa4[0], sa4[1]
+                    if adj_y:
+                        # scores = Q @ Kmat^T on the strided head views —
+                        # no transpose materialization
+                        Nn = sb4[-2]
+                        assert K % 8 == 0
+                        Y = alloc(node.name, (*sa[:-2], M, Nn))
+                        Y4 = Y.view(b1, b2, M, Nn)
+                        steps.append(
+                            lambda At=At, Bt=Bt_in, Y4=Y4, b1=b1, b2=b2,
+                            M=M, Nn=Nn, K=K:
+                            E.gemm_strided_bf16(
+                                At, Bt, Y4, b1, b2, M, Nn, K,
+                                At.stride(2), Bt.stride(2), Y4.stride(2),
+                                At.stride(0), At.stride(1),
+                                Bt.stride(0), Bt.stride(1),
+                                Y4.stride(0), Y4.stride(1)))
+                    else:
+                        # P @ V: V still needs the [N][K] layout (one
+                        # transpose+pad copy); the OUTPUT is written
+                        # directly in the post-transpose layout when the
+                        # sole consumer is Transpose([0,2,1,3])
+                        Nn = sb4[-1]
                         K8 = _pad8(K)
+                        batch = b1 * b2
+                        A3 = At.reshape(batch, M, K) if At.is_contiguous()                             else At.contiguous().view(batch, M, K)
                         Bt = torch.zeros((batch, Nn, K8),
                                          dtype=torch.bfloat16, device=dev)
                         Ap = (A3 if K8 == K else
                               torch.zeros((batch, M, K8),
                                           dtype=torch.bfloat16, device=dev))
-                        Bsrc = bufs[bn].view(batch, K, Nn)
-                        Y = alloc(node.name, (*sa[:-2], M, Nn))
-                        Y3 = Y.view(batch, M, Nn)
+                        # Bsrc may be a strided head view; the per-step
+                        # copy below materializes it (transpose+pad)
+                        Bsrc = Bt_in
+                        cons = consumer_nodes.get(node.name, [])
+                        pre_t = (len(cons) == 1 and cons[0].op == "Transpose"
+                                 and [int(p) for p in self.consts[
+                                     cons[0].inputs[1]].reshape(-1)]
+                                 == [0, 2, 1, 3] and len(sa4) == 4)
+                        if pre_t:
+                            # buffer laid out [b1, M, b2, Nn]; logical
+                            # output is its [0,2,1,3] permute
+                            base = torch.zeros((b1, M, b2, Nn),
+                                               dtype=torch.bfloat16,
+                                               device=dev)
+                            Y4 = base.permute(0, 2, 1, 3)
+                            bufs[node.name] = Y4
+                            shapes[node.name] = (b1, b2, M, Nn)
+                        else:
+                            Y = alloc(node.name, (*sa[:-2], M, Nn))
+                            Y4 = Y.view(b1, b2, M, Nn)
 
-                        def _pv(A3=A3, Ap=Ap, Bt=Bt, Bsrc=Bsrc, Y3=Y3, K=K,
-                                K8=K8):
-                            Bt[:, :, :K].copy_(Bsrc.transpose(1, 2))
+                        def _pv(A3=A3, Ap=Ap, Bt=Bt, Bsrc=Bsrc, Y4=Y4, K=K,
+                                K8=K8, b1=b1, b2=b2, M=M, Nn=Nn):
+                            Bt.view(b1, b2, Nn, K8)[:, :, :, :K].copy_(
+                                Bsrc.transpose(-1, -2))
                             if K8 != K:
                                 Ap[:, :, :K].copy_(A3)
-                            E.gemm_batched_bf16(Ap, Bt, Y3)
+                            Ap4 = Ap.view(b1, b2, M, K8)
+                            Bt4 = Bt.view(b1, b2, Nn, K8)
+                            E.gemm_strided_bf16(
+                                Ap4, Bt4, Y4, b1, b2, M, Nn, K8,
+                                Ap4.stride(2), Bt4.stride(2), Y4.stride(2),
+                                Ap4.stride(0), Ap4.stride(1),
+                                Bt4.stride(0), Bt4.stride(1),
+                                Y4.stride(0), Y4.stride(1))
                         steps.append(_pv)
             elif op == "Transpose":
                 xn = node.inputs[0]
                 perm = [int(p) for p in
                         self.consts[node.inputs[1]].reshape(-1)]
                 out_shape = tuple(shapes[xn][p] for p in perm)
-                Y = alloc(node.name, out_shape)
-                steps.append(lambda X=bufs[xn], Y=Y, perm=perm:
-                             Y.copy_(X.permute(perm)))
+                if perm[-1] == len(perm) - 1:
+                    # inner dim unchanged: serve a zero-copy strided view
+                    # (the strided GEMM consumes it directly)
+                    bufs[node.name] = bufs[xn].permute(perm)
+                    shapes[node.name] = out_shape
+                else:
+                    Y = alloc(node.name, out_shape)
+                    steps.append(lambda X=bufs[xn], Y=Y, perm=perm:
+                                 Y.copy_(X.permute(perm)))
             elif op == "Mul":
                 an, bn = node.inputs
                 cb = self.consts.get(bn)
@@ -422,7 +475,13 @@ class GpuExecutor:
                 if -1 in tgt:
                     known = int(np.prod([t for t in tgt if t != -1]))
                     tgt = [numel // known if t == -1 else t for t in tgt]
-                bufs[node.name] = bufs[xn].view(tgt)
+                src = bufs[xn]
+                if src.is_contiguous():
+                    bufs[node.name] = src.view(tgt)
+                else:
+                    Y = alloc(node.name, tgt)
+                    steps.append(lambda src=src, Y=Y, tgt=tgt:
+                                 Y.copy_(src.reshape(tgt)))
                 shapes[node.name] = tuple(tgt)
             elif op == "Squeeze":
                 xn = node.inputs[0]

@@ -426,6 +426,56 @@ def fuse_dense_bmm(graph: Graph) -> Graph:
     return Graph(new_nodes, graph.consts, graph.inputs, graph.outputs)
 
 
+def fold_scalar_mul(graph: Graph) -> Graph:
+    """Fold Mul(x, scalar_const) into an upstream DenseND/Dense weight
+    (tracing through Transpose/Reshape views). The ViT attention scale
+    (q * 1/sqrt(d)) becomes part of the q-projection weights: the Mul
+    node (a full-tensor pass) disappears."""
+    nmap = graph.node_map()
+    consumers: Dict[str, int] = {}
+    for n in graph.nodes:
+        for i in n.inputs:
+            consumers[i] = consumers.get(i, 0) + 1
+    drop = set()
+    alias: Dict[str, str] = {}
+    for n in graph.nodes:
+        if n.op != "Mul" or len(n.inputs) != 2:
+            continue
+        c = graph.consts.get(n.inputs[1])
+        if c is None or c.size != 1:
+            continue
+        scale = float(c.reshape(-1)[0])
+        # walk to the producing dense through view ops (each sole-consumer)
+        cur = nmap.get(n.inputs[0])
+        chain_ok = True
+        while cur is not None and cur.op in ("Reshape", "Transpose"):
+            if consumers.get(cur.name, 0) != 1:
+                chain_ok = False
+                break
+            cur = nmap.get(cur.inputs[0])
+        if (not chain_ok or cur is None
+                or cur.op not in ("DenseND", "Dense")
+                or consumers.get(cur.name, 0) != 1
+                or cur.inputs[1] not in graph.consts):
+            continue
+        graph.consts[cur.inputs[1]] = (
+            graph.consts[cur.inputs[1]] * scale).astype(
+                graph.consts[cur.inputs[1]].dtype)
+        bname = cur.attrs.get("bias")
+        if bname and bname in graph.consts:
+            graph.consts[bname] = (graph.consts[bname] * scale).astype(
+                graph.consts[bname].dtype)
+        alias[n.name] = n.inputs[0]
+        drop.add(n.name)
+    if not drop:
+        return graph
+    new_nodes = [n for n in graph.nodes if n.name not in drop]
+    for n in new_nodes:
+        n.inputs = [alias.get(i, i) for i in n.inputs]
+    outputs = [alias.get(o, o) for o in graph.outputs]
+    return Graph(new_nodes, graph.consts, graph.inputs, outputs)
+
+
 def toposort(graph: Graph) -> Graph:
     nmap = graph.node_map()
     order: List[Node] = []
@@ -463,6 +513,7 @@ def optimize(graph: Graph) -> Graph:
     g = fuse_layernorm(g)
     g = fuse_gelu(g)
     g = fuse_dense_bmm(g)
+    g = fold_scalar_mul(g)
     g = fuse_conv_blocks(g)
     g = toposort(g)
     return g
