@@ -110,3 +110,34 @@ def test_trace_recorded():
         assert {"batch", "close", "done", "infer_ms"} <= set(t)
     finally:
         b.stop()
+
+
+def test_concurrent_stress():
+    """Race fuzz: many threads hammering submit while the
+    batcher drains; every future must resolve exactly once, correctly."""
+    import threading
+    eng = RecordingEngine(delay=0.001)
+    b = DynamicBatcher(eng, max_batch=32, max_wait_ms=1.0)
+    b.start()
+    results = {}
+    lock = threading.Lock()
+
+    def client(cid):
+        for j in range(20):
+            v = float(cid * 100 + j)
+            r = b.infer_sync(np.full((2, 2, 1), v, np.float32), timeout=30)
+            with lock:
+                results[(cid, j)] = (r[0], 4.0 * v)
+
+    threads = [threading.Thread(target=client, args=(i,)) for i in range(16)]
+    try:
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join(timeout=60)
+        assert len(results) == 16 * 20
+        for (cid, j), (got, want) in results.items():
+            assert got == pytest.approx(want), (cid, j)
+        assert sum(eng.batches) == 16 * 20
+    finally:
+        b.stop()

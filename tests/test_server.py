@@ -90,3 +90,25 @@ def test_e2e_lenet_from_pb(lenet_pb):
     j = r.get_json()
     s = sum(p["prob"] for p in j["predictions"])
     assert 0 < s <= 1.0 + 1e-5
+
+
+def test_debug_trace_endpoint():
+    from tensorflow_web_deploy_amd.sched.batcher import (BatchedEngine,
+                                                         DynamicBatcher)
+    cfg = Config(model_name="lenet", device="cpu")
+    batcher = DynamicBatcher(FakeEngine(10), max_batch=8, max_wait_ms=1.0)
+    batcher.start()
+    try:
+        engine = BatchedEngine(batcher)
+        app = create_app(cfg, engine, batcher=batcher)
+        app.config["TESTING"] = True
+        c = app.test_client()
+        c.post("/predict", data=_png_bytes(),
+               content_type="application/octet-stream")
+        r = c.get("/debug/trace")
+        assert r.status_code == 200
+        j = r.get_json()
+        assert "batches" in j and len(j["batches"]) >= 1
+        assert j["config"]["max_batch"] == 8
+    finally:
+        batcher.stop()
