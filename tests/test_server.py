@@ -95,7 +95,7 @@ def test_e2e_lenet_from_pb(lenet_pb):
 def test_debug_trace_endpoint():
     from tensorflow_web_deploy_amd.sched.batcher import (BatchedEngine,
                                                          DynamicBatcher)
-    cfg = Config(model_name="lenet", device="cpu")
+    cfg = Config(model_name="lenet", device="cpu", max_batch=8)
     batcher = DynamicBatcher(FakeEngine(10), max_batch=8, max_wait_ms=1.0)
     batcher.start()
     try:
