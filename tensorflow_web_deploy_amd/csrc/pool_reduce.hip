@@ -116,10 +116,46 @@ void launch_global_mean_bf16(const void* X, void* Y, int Nimg, int HW, int C,
 
 // --------------------------------------------------------------- softmax
 
-template <typename OUT>
+template <typename OUT, int NV>
 __global__ void softmax_kernel(const __bf16* __restrict__ X,
                                OUT* __restrict__ Y, int M, int N) {
-  // one wave per row (classes ~1e3); wave-wide shuffle reduce
+  // one wave per row; row held in registers (NV values per lane) so the
+  // row is read from HBM exactly once (attention rows: N=196, NV=4)
+  int row = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  int lane = threadIdx.x & 63;
+  if (row >= M) return;
+  const __bf16* x = X + (int64_t)row * N;
+  float r[NV];
+  float m = -3.4e38f;
+#pragma unroll
+  for (int j = 0; j < NV; ++j) {
+    int i = j * 64 + lane;
+    r[j] = i < N ? bf2f(x[i]) : -3.4e38f;
+    m = fmaxf(m, r[j]);
+  }
+#pragma unroll
+  for (int off = 32; off; off >>= 1) m = fmaxf(m, __shfl_xor(m, off, 64));
+  float s = 0.f;
+#pragma unroll
+  for (int j = 0; j < NV; ++j) {
+    r[j] = __expf(r[j] - m);
+    s += r[j];
+  }
+#pragma unroll
+  for (int off = 32; off; off >>= 1) s += __shfl_xor(s, off, 64);
+  float inv = 1.f / s;
+  OUT* y = Y + (int64_t)row * N;
+#pragma unroll
+  for (int j = 0; j < NV; ++j) {
+    int i = j * 64 + lane;
+    if (i < N) y[i] = (OUT)(r[j] * inv);
+  }
+}
+
+// large-N fallback (3 passes; classifier heads are tiny anyway)
+template <typename OUT>
+__global__ void softmax_big_kernel(const __bf16* __restrict__ X,
+                                   OUT* __restrict__ Y, int M, int N) {
   int row = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   int lane = threadIdx.x & 63;
   if (row >= M) return;
@@ -138,18 +174,31 @@ __global__ void softmax_kernel(const __bf16* __restrict__ X,
     y[i] = (OUT)(__expf(bf2f(x[i]) - m) * inv);
 }
 
+template <typename OUT>
+static void softmax_dispatch(const __bf16* X, OUT* Y, int M, int N,
+                             hipStream_t stream) {
+  int wpb = 4;
+  dim3 grid((M + wpb - 1) / wpb), blk(wpb * 64);
+  if (N <= 256)
+    hipLaunchKernelGGL((softmax_kernel<OUT, 4>), grid, blk, 0, stream, X, Y,
+                       M, N);
+  else if (N <= 512)
+    hipLaunchKernelGGL((softmax_kernel<OUT, 8>), grid, blk, 0, stream, X, Y,
+                       M, N);
+  else if (N <= 1024)
+    hipLaunchKernelGGL((softmax_kernel<OUT, 16>), grid, blk, 0, stream, X, Y,
+                       M, N);
+  else
+    hipLaunchKernelGGL((softmax_big_kernel<OUT>), grid, blk, 0, stream, X, Y,
+                       M, N);
+}
+
 void launch_softmax_bf16(const void* X, float* Y, int M, int N,
                          hipStream_t stream) {
-  int wpb = 4;  // 4 waves (rows) per block
-  int blocks = (M + wpb - 1) / wpb;
-  hipLaunchKernelGGL((softmax_kernel<float>), dim3(blocks), dim3(wpb * 64),
-                     0, stream, (const __bf16*)X, Y, M, N);
+  softmax_dispatch<float>((const __bf16*)X, Y, M, N, stream);
 }
 
 void launch_softmax_bf16_out(const void* X, void* Y, int M, int N,
                              hipStream_t stream) {
-  int wpb = 4;
-  int blocks = (M + wpb - 1) / wpb;
-  hipLaunchKernelGGL((softmax_kernel<__bf16>), dim3(blocks), dim3(wpb * 64),
-                     0, stream, (const __bf16*)X, (__bf16*)Y, M, N);
+  softmax_dispatch<__bf16>((const __bf16*)X, (__bf16*)Y, M, N, stream);
 }
