@@ -65,6 +65,12 @@ def main():
     from tensorflow_web_deploy_amd.ops.native import ext
     E = ext()
 
+    print("== ViT-B/16 b64 GEMM shapes ==")
+    for (m, n, k) in [(12544, 768, 768), (12544, 3072, 768),
+                      (12544, 768, 3072), (3136, 512, 2048)]:
+        tf, us = bench_gemm(E, m, n, k, args.iters)
+        print(f"  {m}x{n}x{k}: {tf:8.1f} TF/s  ({us:9.1f} us)")
+
     print("== square GEMMs (A[M,K] @ B[N,K]^T bf16) ==")
     for n in (2048, 4096, 8192):
         tf, us = bench_gemm(E, n, n, n, args.iters)

@@ -24,6 +24,7 @@
 #include <hip/hip_bf16.h>
 #include <algorithm>
 #include <cstdint>
+#include <cstdlib>
 
 #include "kernels.h"
 
@@ -704,12 +705,20 @@ __global__ void splitk_combine_kernel(const float* __restrict__ Cpart,
 // Tile dispatch: 256^2 pipelined kernel when the grid still fills the
 // 256-CU chip and the tile isn't mostly padding; a narrow (256,64) tile
 // for N <= 96 layers; else the general (128,128) tile.
+static int b256_min() {
+  static int v = [] {
+    const char* e = getenv("TWD_B256_MIN");
+    return e ? atoi(e) : 256;
+  }();
+  return v;
+}
+
 static bool use_256(int M, int N, int K) {
   long b256 = ((M + BM2 - 1) / BM2) * (long)((N + BN2 - 1) / BN2);
   // At K = 64 (one K-step) the 256^2 tile is epilogue-dominated
   // (128 KiB LDS restage per block); the 128^2 tile wins there
   // (measured: 1x1 C64->256 189 -> 209 TF; K=128 already favors 256^2).
-  return M >= 192 && N >= 192 && K >= 128 && b256 >= 256;
+  return M >= 192 && N >= 192 && K >= 128 && b256 >= b256_min();
 }
 
 template <int TBM, int TBN, int MODE, int DEPTH = 2>
