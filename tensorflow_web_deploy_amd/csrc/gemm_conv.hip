@@ -708,7 +708,7 @@ __global__ void splitk_combine_kernel(const float* __restrict__ Cpart,
 static int b256_min() {
   static int v = [] {
     const char* e = getenv("TWD_B256_MIN");
-    return e ? atoi(e) : 256;
+    return e ? atoi(e) : 128;  // measured: ViT proj/fc +11-16%
   }();
   return v;
 }
