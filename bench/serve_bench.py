@@ -90,6 +90,8 @@ def main():
     p.add_argument("--port", type=int, default=5123)
     p.add_argument("--max-batch", type=int, default=64)
     p.add_argument("--max-wait-ms", type=float, default=2.0)
+    p.add_argument("--http-workers", type=int, default=0,
+                   help=">0: multi-process SO_REUSEPORT serving stack")
     args = p.parse_args()
 
     import torch
@@ -98,6 +100,7 @@ def main():
     from tensorflow_web_deploy_amd.sched.batcher import (BatchedEngine,
                                                          DynamicBatcher)
     from tensorflow_web_deploy_amd.server.app import create_app
+    from tensorflow_web_deploy_amd.parallel.service import MultiprocServer
 
     size = 299 if args.model == "inception_v3" else 224
     gray = args.model == "lenet"
@@ -109,7 +112,11 @@ def main():
                  num_classes=10 if gray else 1000,
                  max_batch=args.max_batch, max_wait_ms=args.max_wait_ms,
                  gpus=args.gpus, port=args.port)
-    if args.gpus > 1:
+    mserver = None
+    batcher = None
+    if args.http_workers > 0:
+        mserver = MultiprocServer(cfg, http_workers=args.http_workers)
+    elif args.gpus > 1:
         from tensorflow_web_deploy_amd.parallel.replicas import \
             ReplicatedEngine
         backend = ReplicatedEngine(cfg)
@@ -118,26 +125,28 @@ def main():
         backend = GpuServingEngine(cfg)
     else:
         backend = LocalEngine(cfg)
-    # warm the engine (compile + HIP-graph capture for every bucket)
-    b = 1
-    while True:
-        backend.infer(np.zeros((b, h, w, cfg.input_channels), np.float32))
-        if b >= args.max_batch:
-            break
-        b = min(b * 2, args.max_batch)
-    batcher = DynamicBatcher(backend, max_batch=cfg.max_batch,
-                             max_wait_ms=cfg.max_wait_ms)
-    batcher.start()
-    engine = BatchedEngine(batcher, raw=hasattr(backend, "infer_blobs"))
-    if hasattr(backend, "pre"):
-        backend.pre.warm_pool()
-    app = create_app(cfg, engine)
+    if mserver is None:
+        # warm the engine (compile + HIP-graph capture for every bucket)
+        b = 1
+        while True:
+            backend.infer(np.zeros((b, h, w, cfg.input_channels),
+                                   np.float32))
+            if b >= args.max_batch:
+                break
+            b = min(b * 2, args.max_batch)
+        batcher = DynamicBatcher(backend, max_batch=cfg.max_batch,
+                                 max_wait_ms=cfg.max_wait_ms)
+        batcher.start()
+        engine = BatchedEngine(batcher, raw=hasattr(backend, "infer_blobs"))
+        if hasattr(backend, "pre"):
+            backend.pre.warm_pool()
+        app = create_app(cfg, engine)
 
-    th = threading.Thread(
-        target=lambda: app.run(host="127.0.0.1", port=args.port,
-                               threaded=True, use_reloader=False),
-        daemon=True)
-    th.start()
+        th = threading.Thread(
+            target=lambda: app.run(host="127.0.0.1", port=args.port,
+                                   threaded=True, use_reloader=False),
+            daemon=True)
+        th.start()
     time.sleep(1.0)
 
     # Load generation runs in SEPARATE PROCESSES so the measurement is not
@@ -147,12 +156,20 @@ def main():
     img = _png_bytes(h, w, gray)
     url = f"http://127.0.0.1:{args.port}/predict"
 
-    # warmup through the full stack
+    # warmup through the full stack (retry while HTTP workers boot)
     import urllib.request
-    for _ in range(4):
-        req = urllib.request.Request(url, data=img, headers={
-            "Content-Type": "application/octet-stream"})
-        urllib.request.urlopen(req, timeout=60).read()
+    deadline = time.time() + 120
+    done = 0
+    while done < 4:
+        try:
+            req = urllib.request.Request(url, data=img, headers={
+                "Content-Type": "application/octet-stream"})
+            urllib.request.urlopen(req, timeout=60).read()
+            done += 1
+        except Exception:
+            if time.time() > deadline:
+                raise
+            time.sleep(0.5)
 
     nproc = min(args.clients, 16)
     per_proc_threads = max(1, args.clients // nproc)
@@ -174,7 +191,8 @@ def main():
         pr.join(timeout=10)
     ls = sorted(lat)
     n = len(ls)
-    batcher.stop()
+    if batcher is not None:
+        batcher.stop()
     out = {
         "metric": "serve_images_per_sec" if args.qps <= 0 else "serve_p50_ms",
         "mode": "closed" if args.qps <= 0 else f"open@{args.qps}qps",
@@ -184,9 +202,12 @@ def main():
         "p50_ms": round(1000 * ls[len(ls) // 2], 2) if ls else None,
         "p99_ms": round(1000 * ls[int(len(ls) * 0.99)], 2) if ls else None,
         "requests": n,
-        "batch_trace_tail": batcher.trace[-3:],
+        "http_workers": args.http_workers,
+        "batch_trace_tail": batcher.trace[-3:] if batcher else None,
     }
     print(json.dumps(out))
+    if mserver is not None:
+        mserver.shutdown()
 
 
 if __name__ == "__main__":

@@ -79,6 +79,25 @@ class LocalEngine:
         out = self.ex.run({self.cfg.input_name: x})[0]
         return out.float().cpu().numpy()[:n]
 
+    def infer_arrays(self, arrs) -> np.ndarray:
+        """Decoded uint8 HWC arrays at original sizes -> probs (CPU
+        resize path; GpuServingEngine overrides with the HIP kernel)."""
+        from PIL import Image
+        batch = []
+        for a in arrs:
+            img = Image.fromarray(a.squeeze(-1) if a.shape[-1] == 1 else a)
+            img = img.resize((self.cfg.input_width, self.cfg.input_height),
+                             Image.BILINEAR)
+            f = np.asarray(img, dtype=np.float32)
+            if f.ndim == 2:
+                f = f[:, :, None]
+            if self.cfg.normalize == "unit":
+                f = f / 255.0
+            elif self.cfg.normalize == "inception":
+                f = f / 127.5 - 1.0
+            batch.append(f)
+        return self.infer(np.stack(batch, 0))
+
 
 class GpuServingEngine(LocalEngine):
     """LocalEngine + GPU ingest: raw image bytes -> side-stream HIP

@@ -1,0 +1,215 @@
+"""Shared inference service + multi-process HTTP serving.
+
+The single-process Flask stack tops out near 600 req/s (werkzeug +
+decode + GIL). Production shape (SURVEY §3b scaled out):
+
+  service process: DynamicBatcher + engine (GPU; ReplicatedEngine for
+      DP>1). Receives decoded uint8 images over an mp queue, batches,
+      GPU-preprocesses (resize/normalize on the side stream), replies to
+      per-worker response queues.
+  N HTTP worker processes: each binds the SAME port with SO_REUSEPORT
+      (kernel load-balances accepts), runs the Flask app threaded,
+      decodes uploads in its own interpreter (parallel decode, no shared
+      GIL), and round-trips the service.
+"""
+from __future__ import annotations
+
+import itertools
+import os
+import queue
+import socket
+import threading
+import time
+import traceback
+from typing import Optional
+
+import numpy as np
+import torch.multiprocessing as mp
+
+from ..conf.config import Config
+
+
+def _service_main(cfg: Config, in_q, resp_qs, err_q, ready_ev):
+    try:
+        import torch
+        from concurrent.futures import ThreadPoolExecutor
+
+        from ..engine.service import GpuServingEngine, LocalEngine
+        from ..sched.batcher import DynamicBatcher
+        if cfg.gpus > 1:
+            from .replicas import ReplicatedEngine
+            engine = ReplicatedEngine(cfg)
+        elif torch.cuda.is_available():
+            engine = GpuServingEngine(cfg)
+        else:
+            engine = LocalEngine(cfg)
+        batcher = DynamicBatcher(engine, max_batch=cfg.max_batch,
+                                 max_wait_ms=cfg.max_wait_ms)
+        batcher.start()
+        # warm common batch buckets so serving never hits a compile
+        h, w, c = cfg.input_height, cfg.input_width, cfg.input_channels
+        b = 1
+        while True:
+            engine.infer(np.zeros((b, h, w, c), np.float32))
+            if b >= cfg.max_batch:
+                break
+            b = min(b * 2, cfg.max_batch)
+        ready_ev.set()
+        pool = ThreadPoolExecutor(max_workers=128)
+
+        def respond(wid, rid, fut):
+            try:
+                resp_qs[wid].put((rid, fut.result(timeout=60.0), None))
+            except Exception as e:  # noqa: BLE001
+                resp_qs[wid].put((rid, None, repr(e)))
+
+        while True:
+            item = in_q.get()
+            if item is None:
+                break
+            wid, rid, arr = item
+            fut = batcher.submit(arr)
+            pool.submit(respond, wid, rid, fut)
+        batcher.stop()
+    except Exception:
+        err_q.put(traceback.format_exc())
+        raise
+
+
+class ServiceClient:
+    """Engine-interface adapter used inside an HTTP worker process."""
+
+    def __init__(self, wid: int, in_q, resp_q, cfg: Config):
+        self.wid = wid
+        self.in_q = in_q
+        self.resp_q = resp_q
+        self.cfg = cfg
+        self._rid = itertools.count()
+        self._lock = threading.Lock()
+        self._waiting: dict = {}
+        self._reader = threading.Thread(target=self._drain, daemon=True)
+        self._reader.start()
+
+    def _drain(self):
+        while True:
+            rid, row, err = self.resp_q.get()
+            with self._lock:
+                slot = self._waiting.pop(rid, None)
+            if slot is not None:
+                slot["row"] = row
+                slot["err"] = err
+                slot["ev"].set()
+
+    def _roundtrip(self, arr: np.ndarray):
+        rid = next(self._rid)
+        slot = {"ev": threading.Event()}
+        with self._lock:
+            self._waiting[rid] = slot
+        self.in_q.put((self.wid, rid, arr))
+        if not slot["ev"].wait(timeout=60.0):
+            raise TimeoutError("inference service timeout")
+        if slot["err"]:
+            raise RuntimeError(slot["err"])
+        return slot["row"]
+
+    # engine interface used by server/app.py
+    def infer_arrays(self, arrs):
+        return np.stack([self._roundtrip(a) for a in arrs], axis=0)
+
+    def infer(self, batch: np.ndarray) -> np.ndarray:
+        return np.stack([self._roundtrip(batch[i])
+                         for i in range(batch.shape[0])], axis=0)
+
+    def decode(self, blob: bytes) -> np.ndarray:
+        """Entropy decode in THIS worker process (parallel across
+        workers); resize/normalize happen on the GPU in the service."""
+        import io
+
+        from PIL import Image
+        img = Image.open(io.BytesIO(blob))
+        img = img.convert("L" if self.cfg.input_channels == 1 else "RGB")
+        arr = np.asarray(img, dtype=np.uint8)
+        if arr.ndim == 2:
+            arr = arr[:, :, None]
+        return arr
+
+
+def _http_worker_main(wid: int, cfg: Config, in_q, resp_q):
+    from ..server.app import create_app
+    client = ServiceClient(wid, in_q, resp_q, cfg)
+    app = create_app(cfg, client)
+    sock = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+    sock.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+    sock.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEPORT, 1)
+    sock.bind((cfg.host, cfg.port))
+    sock.listen(512)
+    from werkzeug.serving import make_server
+    srv = make_server(cfg.host, cfg.port, app, threaded=True,
+                      fd=sock.fileno())
+    srv.serve_forever()
+
+
+class MultiprocServer:
+    """N SO_REUSEPORT HTTP workers + 1 inference-service process."""
+
+    def __init__(self, cfg: Config, http_workers: int = 8,
+                 start_timeout: float = 900.0):
+        self.cfg = cfg
+        ctx = mp.get_context("spawn")
+        self.in_q = ctx.Queue()
+        self.err_q = ctx.Queue()
+        self.resp_qs = [ctx.Queue() for _ in range(http_workers)]
+        ready = ctx.Event()
+        self.service = ctx.Process(
+            target=_service_main,
+            args=(cfg, self.in_q, self.resp_qs, self.err_q, ready),
+            daemon=True)
+        self.service.start()
+        deadline = time.time() + start_timeout
+        while not ready.wait(timeout=1.0):
+            if not self.service.is_alive() or time.time() > deadline:
+                try:
+                    raise RuntimeError(self.err_q.get_nowait())
+                except queue.Empty:
+                    raise RuntimeError("inference service failed to start")
+        self.workers = []
+        for wid in range(http_workers):
+            p = ctx.Process(target=_http_worker_main,
+                            args=(wid, cfg, self.in_q, self.resp_qs[wid]),
+                            daemon=True)
+            p.start()
+            self.workers.append(p)
+
+    def shutdown(self):
+        for p in self.workers:
+            p.terminate()
+        try:
+            self.in_q.put(None)
+        except Exception:
+            pass
+        self.service.join(timeout=10.0)
+        if self.service.is_alive():
+            self.service.terminate()
+
+
+def main(argv=None):
+    import argparse
+
+    from ..conf.config import add_config_args, config_from_args
+    p = argparse.ArgumentParser("multi-process serving")
+    add_config_args(p)
+    p.add_argument("--http-workers", type=int, default=8)
+    args = p.parse_args(argv)
+    cfg = config_from_args(args)
+    srv = MultiprocServer(cfg, http_workers=args.http_workers)
+    print(f"serving on {cfg.host}:{cfg.port} with {args.http_workers} "
+          f"HTTP workers; Ctrl-C to stop")
+    try:
+        while True:
+            time.sleep(3600)
+    except KeyboardInterrupt:
+        srv.shutdown()
+
+
+if __name__ == "__main__":
+    main()
