@@ -72,10 +72,30 @@ def _service_main(cfg: Config, in_q, resp_qs, err_q, ready_ev):
                 resp_qs[wid].put((rid, None, repr(e)))
 
         shms: dict = {}
+        stats = {"items": 0, "t0": time.time()}
+
+        def _stat_loop():
+            import sys
+            while True:
+                time.sleep(5.0)
+                tr = batcher.trace[-200:]
+                if tr:
+                    bs = [t["batch"] for t in tr]
+                    ms = [t["infer_ms"] for t in tr]
+                    print(f"[service] items={stats['items']} "
+                          f"rate={stats['items']/(time.time()-stats['t0']):.0f}/s "
+                          f"batches={len(batcher.trace)} "
+                          f"avg_batch={sum(bs)/len(bs):.1f} "
+                          f"avg_infer={sum(ms)/len(ms):.1f}ms "
+                          f"max_infer={max(ms):.1f}ms",
+                          file=sys.stderr, flush=True)
+        if os.environ.get("TWD_SERVICE_STATS"):
+            threading.Thread(target=_stat_loop, daemon=True).start()
         while True:
             item = in_q.get()
             if item is None:
                 break
+            stats["items"] += 1
             wid, rid, payload = item
             if isinstance(payload, tuple):  # shm descriptor
                 shm_name, slot, shape = payload
