@@ -63,11 +63,14 @@ def _service_main(cfg: Config, in_q, resp_qs, err_q, ready_ev):
                 break
             b = min(b * 2, cfg.max_batch)
         ready_ev.set()
-        pool = ThreadPoolExecutor(max_workers=128)
 
+        # responses ride the future's done-callback (runs in the batcher
+        # thread): a per-request responder-thread pool caused a GIL
+        # stampede after every batch (measured 25-45 ms per batch for
+        # ~3 ms of GPU work)
         def respond(wid, rid, fut):
             try:
-                resp_qs[wid].put((rid, fut.result(timeout=60.0), None))
+                resp_qs[wid].put((rid, fut.result(timeout=0), None))
             except Exception as e:  # noqa: BLE001
                 resp_qs[wid].put((rid, None, repr(e)))
 
@@ -111,7 +114,8 @@ def _service_main(cfg: Config, in_q, resp_qs, err_q, ready_ev):
             else:
                 arr = payload
             fut = batcher.submit(arr)
-            pool.submit(respond, wid, rid, fut)
+            fut.add_done_callback(
+                lambda f, wid=wid, rid=rid: respond(wid, rid, f))
         batcher.stop()
     except Exception:
         err_q.put(traceback.format_exc())
@@ -132,6 +136,7 @@ class ServiceClient:
         self._shm = shared_memory.SharedMemory(
             create=True, size=SLOT_BYTES * SLOTS_PER_WORKER)
         self._free = list(range(SLOTS_PER_WORKER))
+        self._stats = [] if os.environ.get("TWD_SERVICE_STATS") else None
         self._reader = threading.Thread(target=self._drain, daemon=True)
         self._reader.start()
 
@@ -149,6 +154,7 @@ class ServiceClient:
                 slot["ev"].set()
 
     def _roundtrip(self, arr: np.ndarray):
+        t0 = time.perf_counter()
         rid = next(self._rid)
         slot = {"ev": threading.Event(), "shm_slot": None}
         payload = arr
@@ -168,6 +174,14 @@ class ServiceClient:
             raise TimeoutError("inference service timeout")
         if slot["err"]:
             raise RuntimeError(slot["err"])
+        if self._stats is not None:
+            self._stats.append(time.perf_counter() - t0)
+            if len(self._stats) % 200 == 0:
+                import sys
+                ls = sorted(self._stats[-200:])
+                print(f"[client {self.wid}] roundtrip p50="
+                      f"{1000*ls[100]:.1f}ms p90={1000*ls[180]:.1f}ms "
+                      f"n={len(self._stats)}", file=sys.stderr, flush=True)
         return slot["row"]
 
     # engine interface used by server/app.py
