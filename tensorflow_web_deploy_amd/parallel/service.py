@@ -85,6 +85,11 @@ def _service_main(cfg: Config, in_q, resp_qs, err_q, ready_ev):
                 if tr:
                     bs = [t["batch"] for t in tr]
                     ms = [t["infer_ms"] for t in tr]
+                    sg = stats.get("seg", {})
+                    segs = " ".join(f"{k}={1000*v/max(stats['items'],1):.2f}ms"
+                                    for k, v in sg.items())
+                    print(f"[service-seg per-item] {segs}",
+                          file=sys.stderr, flush=True)
                     print(f"[service] items={stats['items']} "
                           f"rate={stats['items']/(time.time()-stats['t0']):.0f}/s "
                           f"batches={len(batcher.trace)} "
@@ -94,8 +99,12 @@ def _service_main(cfg: Config, in_q, resp_qs, err_q, ready_ev):
                           file=sys.stderr, flush=True)
         if os.environ.get("TWD_SERVICE_STATS"):
             threading.Thread(target=_stat_loop, daemon=True).start()
+        seg = {"get": 0.0, "copy": 0.0, "submit": 0.0}
+        stats["seg"] = seg
         while True:
+            tg0 = time.perf_counter()
             item = in_q.get()
+            seg["get"] += time.perf_counter() - tg0
             if item is None:
                 break
             stats["items"] += 1
@@ -110,12 +119,16 @@ def _service_main(cfg: Config, in_q, resp_qs, err_q, ready_ev):
                 view = np.ndarray(shape, np.uint8,
                                   buffer=shm.buf,
                                   offset=slot * SLOT_BYTES)
+                tc0 = time.perf_counter()
                 arr = view.copy()  # slot is freed by the worker on reply
+                seg["copy"] += time.perf_counter() - tc0
             else:
                 arr = payload
+            ts0 = time.perf_counter()
             fut = batcher.submit(arr)
             fut.add_done_callback(
                 lambda f, wid=wid, rid=rid: respond(wid, rid, f))
+            seg["submit"] += time.perf_counter() - ts0
         batcher.stop()
     except Exception:
         err_q.put(traceback.format_exc())
