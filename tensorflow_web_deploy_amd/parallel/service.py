@@ -220,6 +220,10 @@ class ServiceClient:
 
 
 def _http_worker_main(wid: int, cfg: Config, in_q, resp_q):
+    import sys
+    # response-drain thread must not starve behind werkzeug request
+    # threads: switch the GIL faster in HTTP workers
+    sys.setswitchinterval(0.0005)
     from ..server.app import create_app
     client = ServiceClient(wid, in_q, resp_q, cfg)
     app = create_app(cfg, client)
