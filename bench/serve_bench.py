@@ -134,6 +134,12 @@ def main():
             if b >= args.max_batch:
                 break
             b = min(b * 2, args.max_batch)
+        # GC hygiene: per-request ndarray churn triggers periodic gen-2
+        # collections that showed up as 90-280 ms latency spikes
+        import gc
+        gc.collect()
+        gc.freeze()
+        gc.set_threshold(200000, 100, 100)
         batcher = DynamicBatcher(backend, max_batch=cfg.max_batch,
                                  max_wait_ms=cfg.max_wait_ms)
         batcher.start()

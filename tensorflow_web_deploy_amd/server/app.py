@@ -154,6 +154,10 @@ def main(argv=None):
         batcher.start()
         engine = BatchedEngine(batcher,
                                raw=hasattr(backend_engine, "infer_blobs"))
+    import gc
+    gc.collect()
+    gc.freeze()
+    gc.set_threshold(200000, 100, 100)
     app = create_app(cfg, engine)
     app.run(host=cfg.host, port=cfg.port, threaded=True)
 
