@@ -23,6 +23,13 @@ import numpy as np
 import torch
 
 
+def _pool_init():
+    import gc
+    gc.collect()
+    gc.freeze()
+    gc.set_threshold(200000, 100, 100)
+
+
 def _decode_worker(args):
     """Module-level for process-pool pickling: bytes -> uint8 HWC."""
     data, channels = args
@@ -54,7 +61,8 @@ class GpuPreprocessor:
         # batch stalls); processes sidestep the GIL entirely.
         import multiprocessing as mp
         self._pool = ProcessPoolExecutor(
-            max_workers=8, mp_context=mp.get_context("spawn"))
+            max_workers=8, mp_context=mp.get_context("spawn"),
+            initializer=_pool_init)
         self._pool_warm = False
 
     def _decode_cpu(self, data: bytes) -> np.ndarray:
@@ -74,9 +82,13 @@ class GpuPreprocessor:
         Upload + resize/normalize run on the side stream; the returned
         tensor is safe to consume on `compute_stream` (event-synced).
         """
-        arrs = list(self._pool.map(_decode_worker,
-                                   [(b, self.c) for b in blobs],
-                                   chunksize=4))
+        if len(blobs) <= 4:
+            # small batches: pool round-trip costs more than decode
+            arrs = [_decode_worker((b, self.c)) for b in blobs]
+        else:
+            arrs = list(self._pool.map(_decode_worker,
+                                       [(b, self.c) for b in blobs],
+                                       chunksize=4))
         return self.process_arrays(arrs, compute_stream)
 
     def warm_pool(self):
