@@ -129,8 +129,13 @@ class GpuServingEngine(LocalEngine):
 
     @torch.inference_mode()
     def infer_blobs(self, blobs) -> np.ndarray:
+        import os
+        import time as _t
+        dbg = os.environ.get("TWD_INFER_STATS")
+        t0 = _t.perf_counter()
         n = len(blobs)
         batch = self.pre.process(blobs)
+        t1 = _t.perf_counter()
         nb = _bucket(n)
         if nb != n:
             # the preprocessor's output buffer has headroom; zero-fill
@@ -139,7 +144,15 @@ class GpuServingEngine(LocalEngine):
             full[n:].zero_()
             batch = full
         out = self.ex.run({self.cfg.input_name: batch})[0]
-        return out.float().cpu().numpy()[:n]
+        t2 = _t.perf_counter()
+        res = out.float().cpu().numpy()[:n]
+        t3 = _t.perf_counter()
+        if dbg and t3 - t0 > 0.03:
+            import sys
+            print(f"[infer n={n}->{nb}] pre={1000*(t1-t0):.1f} "
+                  f"run={1000*(t2-t1):.1f} sync={1000*(t3-t2):.1f}ms",
+                  file=sys.stderr, flush=True)
+        return res
 
 
 class FakeEngine:
