@@ -23,6 +23,7 @@ ext = cpp_extension.CUDAExtension(
         os.path.join(CSRC, "elementwise.hip"),
         os.path.join(CSRC, "layernorm.hip"),
         os.path.join(CSRC, "preprocess.hip"),
+        os.path.join(CSRC, "attn_fused.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

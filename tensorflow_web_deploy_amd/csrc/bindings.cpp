@@ -275,6 +275,23 @@ void gemm_strided_bf16(torch::Tensor A, torch::Tensor B, torch::Tensor C,
                            sC1, sC2, A.numel(), cur_stream());
 }
 
+// Fused attention: Q/K/V/O are 4D views [B1, S, B2, D] or [B1, B2, S, D]
+// etc — every tensor supplies explicit strides; D must be 64, S <= 256.
+void attn_fused_bf16(torch::Tensor Q, torch::Tensor K, torch::Tensor V,
+                     torch::Tensor O, int64_t batch1, int64_t batch2,
+                     int64_t S, int64_t sQ1, int64_t sQ2, int64_t lqd,
+                     int64_t sK1, int64_t sK2, int64_t lkd, int64_t sV1,
+                     int64_t sV2, int64_t lvd, int64_t sO1, int64_t sO2,
+                     int64_t lod, double scale) {
+  TORCH_CHECK(Q.is_cuda() && Q.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(S <= 256, "fused attention supports S <= 256");
+  launch_attn_fused_bf16(Q.data_ptr(), K.data_ptr(), V.data_ptr(),
+                         O.data_ptr(), (int)batch1, (int)batch2, (int)S,
+                         sQ1, sQ2, (int)lqd, sK1, sK2, (int)lkd, sV1, sV2,
+                         (int)lvd, sO1, sO2, (int)lod, (float)scale,
+                         cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -295,6 +312,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_bf16", &layernorm_bf16);
   m.def("gemm_batched_bf16", &gemm_batched_bf16);
   m.def("gemm_strided_bf16", &gemm_strided_bf16);
+  m.def("attn_fused_bf16", &attn_fused_bf16);
   m.def("resize_norm", &resize_norm);
   m.def("resize_norm_batch", &resize_norm_batch);
 }

@@ -77,3 +77,11 @@ void launch_resize_norm(const void* src, int sh, int sw, int sc, float* dst,
 void launch_resize_norm_batch(const void* arena, const int* meta,
                               float* dst, int n, int dh, int dw, int dc,
                               int norm, hipStream_t stream);
+
+// attn_fused.hip — fused softmax attention (S <= 256, D = 64)
+void launch_attn_fused_bf16(const void* Q, const void* K, const void* V,
+                            void* O, int batch1, int batch2, int S,
+                            int64_t sQ1, int64_t sQ2, int lqd, int64_t sK1,
+                            int64_t sK2, int lkd, int64_t sV1, int64_t sV2,
+                            int lvd, int64_t sO1, int64_t sO2, int lod,
+                            float scale, hipStream_t stream);

@@ -232,3 +232,44 @@ def test_gemm_batched(E):
     E.gemm_batched_bf16(A, B, C)
     ref = torch.matmul(A.float(), B.float().transpose(1, 2))
     torch.testing.assert_close(C.float(), ref, rtol=2e-2, atol=0.3)
+
+
+def test_attn_fused(E):
+    """Fused attention vs torch fp32 reference on strided head views."""
+    torch.manual_seed(11)
+    B, H, S, D = 3, 4, 196, 64
+    # natural qkv layout [B, S, H, D] (no transposes, like the executor)
+    q = _bf(torch.randn(B, S, H, D))
+    k = _bf(torch.randn(B, S, H, D))
+    v = _bf(torch.randn(B, S, H, D))
+    out = torch.zeros(B, S, H, D, dtype=torch.bfloat16, device="cuda")
+    scale = 1.0 / D ** 0.5
+    E.attn_fused_bf16(q, k, v, out, B, H, S,
+                      q.stride(0), q.stride(2), q.stride(1),
+                      k.stride(0), k.stride(2), k.stride(1),
+                      v.stride(0), v.stride(2), v.stride(1),
+                      out.stride(0), out.stride(2), out.stride(1),
+                      scale)
+    qf = q.float().permute(0, 2, 1, 3)
+    kf = k.float().permute(0, 2, 1, 3)
+    vf = v.float().permute(0, 2, 1, 3)
+    ref = torch.softmax(qf @ kf.transpose(-1, -2) * scale, -1) @ vf
+    got = out.float().permute(0, 2, 1, 3)
+    torch.testing.assert_close(got, ref, rtol=2e-2, atol=2e-2)
+    # spiked input forces large rescales (guide §5.4 rule 26)
+    q2 = q.clone()
+    q2[:, 5] *= 30
+    k2 = k.clone()
+    k2[:, 100] *= 30
+    out2 = torch.zeros_like(out)
+    E.attn_fused_bf16(q2, k2, v, out2, B, H, S,
+                      q2.stride(0), q2.stride(2), q2.stride(1),
+                      k2.stride(0), k2.stride(2), k2.stride(1),
+                      v.stride(0), v.stride(2), v.stride(1),
+                      out2.stride(0), out2.stride(2), out2.stride(1),
+                      scale)
+    ref2 = torch.softmax(q2.float().permute(0, 2, 1, 3) @
+                         k2.float().permute(0, 2, 1, 3).transpose(-1, -2)
+                         * scale, -1) @ vf
+    torch.testing.assert_close(out2.float().permute(0, 2, 1, 3), ref2,
+                               rtol=3e-2, atol=3e-2)
