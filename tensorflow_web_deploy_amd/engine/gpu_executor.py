@@ -188,8 +188,12 @@ class GpuExecutor:
                 c0 += w
             concat_fused[n.name] = c0
 
+        fused_skip: set = set()
         for node in g.nodes:
             op = node.op
+            if node.name in fused_skip:
+                # consumed by the fused-attention step
+                continue
             if op in ("ConvBlock",):
                 xn = node.inputs[0]
                 X = bufs[xn]
@@ -375,6 +379,59 @@ class GpuExecutor:
                     else:
                         sa4, sb4 = sa, sb
                     b1, b2 = sa4[0], sa4[1]
+                    # ---- fused attention: BMM(adj_y) -> Softmax -> BMM
+                    # collapses into one flash-style kernel (no S x S
+                    # score tensor in HBM) when shapes fit
+                    if adj_y and K == 64 and M <= 256 and len(sa4) == 4:
+                        c1 = consumer_nodes.get(node.name, [])
+                        sm = c1[0] if len(c1) == 1 else None
+                        pv = None
+                        if sm is not None and sm.op == "Softmax":
+                            c2 = consumer_nodes.get(sm.name, [])
+                            if (len(c2) == 1
+                                    and c2[0].op in ("BatchMatMul",
+                                                     "BatchMatMulV2")
+                                    and not c2[0].attrs.get("adj_y")
+                                    and c2[0].inputs[0] == sm.name):
+                                pv = c2[0]
+                        if pv is not None:
+                            vt = bufs[pv.inputs[1]]
+                            if vt.dim() == 4 and vt.stride(3) == 1:
+                                b1v, b2v = sa4[0], sa4[1]
+                                Sv = M
+                                cons = consumer_nodes.get(pv.name, [])
+                                pre_t = (len(cons) == 1
+                                         and cons[0].op == "Transpose"
+                                         and [int(p) for p in self.consts[
+                                             cons[0].inputs[1]].reshape(-1)]
+                                         == [0, 2, 1, 3])
+                                if pre_t:
+                                    base = torch.zeros((b1v, Sv, b2v, 64),
+                                                       dtype=torch.bfloat16,
+                                                       device=dev)
+                                    Y4 = base.permute(0, 2, 1, 3)
+                                else:
+                                    Y = alloc(pv.name, (b1v, b2v, Sv, 64))
+                                    Y4 = Y
+                                bufs[pv.name] = Y4
+                                shapes[pv.name] = (b1v, b2v, Sv, 64)
+                                shapes[node.name] = (b1v, b2v, M, M)
+                                shapes[sm.name] = (b1v, b2v, M, M)
+                                fused_skip.update({sm.name, pv.name})
+                                steps.append(
+                                    lambda Aq=At, Bk=Bt_in, Vv=vt, Y4=Y4,
+                                    b1v=b1v, b2v=b2v, Sv=Sv:
+                                    E.attn_fused_bf16(
+                                        Aq, Bk, Vv, Y4, b1v, b2v, Sv,
+                                        Aq.stride(0), Aq.stride(1),
+                                        Aq.stride(2),
+                                        Bk.stride(0), Bk.stride(1),
+                                        Bk.stride(2),
+                                        Vv.stride(0), Vv.stride(1),
+                                        Vv.stride(2),
+                                        Y4.stride(0), Y4.stride(1),
+                                        Y4.stride(2), 1.0))
+                                continue
                     if adj_y:
                         # scores = Q @ Kmat^T on the strided head views —
                         # no transpose materialization
