@@ -351,6 +351,33 @@ class GpuExecutor:
                 steps.append(lambda A2=A2, wt=wt, Y2=Y2, bias=bias, K=K,
                              act=act:
                              E.gemm_bf16(A2, wt, Y2, bias, None, K, act))
+            elif op in ("BatchMatMul", "BatchMatMulV2") \
+                    and node.name in pending_attn:
+                At, Bk, b1v, b2v, Sv = pending_attn.pop(node.name)
+                Vv = bufs[node.inputs[1]]
+                assert Vv.dim() == 4 and Vv.stride(3) == 1, \
+                    "fused attention: V view must be 4D k-contiguous"
+                cons = consumer_nodes.get(node.name, [])
+                pre_t = (len(cons) == 1 and cons[0].op == "Transpose"
+                         and [int(p) for p in self.consts[
+                             cons[0].inputs[1]].reshape(-1)] == [0, 2, 1, 3])
+                if pre_t:
+                    base = torch.zeros((b1v, Sv, b2v, 64),
+                                       dtype=torch.bfloat16, device=dev)
+                    Y4 = base.permute(0, 2, 1, 3)
+                else:
+                    Y4 = alloc(node.name, (b1v, b2v, Sv, 64))
+                bufs[node.name] = Y4
+                shapes[node.name] = (b1v, b2v, Sv, 64)
+                steps.append(
+                    lambda Aq=At, Bk=Bk, Vv=Vv, Y4=Y4, b1v=b1v, b2v=b2v,
+                    Sv=Sv:
+                    E.attn_fused_bf16(
+                        Aq, Bk, Vv, Y4, b1v, b2v, Sv,
+                        Aq.stride(0), Aq.stride(1), Aq.stride(2),
+                        Bk.stride(0), Bk.stride(1), Bk.stride(2),
+                        Vv.stride(0), Vv.stride(1), Vv.stride(2),
+                        Y4.stride(0), Y4.stride(1), Y4.stride(2), 1.0))
             elif op in ("BatchMatMul", "BatchMatMulV2"):
                 an, bn = node.inputs
                 sa = shapes[an]
