@@ -20,6 +20,25 @@ if args.shape == "square":
     C = torch.zeros(n, n, dtype=torch.bfloat16, device="cuda")
     for _ in range(args.iters):
         E.gemm_bf16(A, B, C, None, None, n, False)
+elif args.shape == "ln":
+    X = torch.randn(12544, 768, device="cuda").bfloat16()
+    g = torch.rand(768, device="cuda")
+    b = torch.randn(768, device="cuda")
+    Y = torch.zeros_like(X)
+    for _ in range(args.iters):
+        E.layernorm_bf16(X, g, b, Y, 1e-6)
+elif args.shape == "attn":
+    B, H, S, D = 64, 12, 196, 64
+    q = torch.randn(B, S, H, D, device="cuda").bfloat16()
+    k = torch.randn(B, S, H, D, device="cuda").bfloat16()
+    v = torch.randn(B, S, H, D, device="cuda").bfloat16()
+    o = torch.zeros_like(q)
+    for _ in range(args.iters):
+        E.attn_fused_bf16(q, k, v, o, B, H, S,
+                          q.stride(0), q.stride(2), q.stride(1),
+                          k.stride(0), k.stride(2), k.stride(1),
+                          v.stride(0), v.stride(2), v.stride(1),
+                          o.stride(0), o.stride(2), o.stride(1), 0.125)
 elif args.shape == "s13x3":
     X = torch.randn(64, 56, 56, 64, device="cuda").bfloat16()
     W2 = torch.randn(64, 576, device="cuda").bfloat16()

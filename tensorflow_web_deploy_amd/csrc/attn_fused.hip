@@ -85,9 +85,14 @@ void attn_fused_kernel(const __bf16* __restrict__ Q,
 #pragma unroll
       for (int e = 0; e < 8; ++e) vv[e] = (__bf16)0.f;
     }
+    // lane-rotated write order: without it the 8 lanes of a group all
+    // write d-rows 8 apart (stride = a multiple of 32 banks) at the
+    // same key column -> 8-way conflict on every ds_write_b16
 #pragma unroll
-    for (int e = 0; e < 8; ++e)
-      vt_lds[(slot * 8 + e) * vt_ld + row] = vv[e];
+    for (int e = 0; e < 8; ++e) {
+      int dd = (e + lane) & 7;
+      vt_lds[(slot * 8 + dd) * vt_ld + row] = vv[dd];
+    }
   }
 
   // ---- load this wave's Q fragments (A-operand layout) ----
