@@ -27,7 +27,8 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 __device__ __forceinline__ float bf2f(__bf16 v) { return (float)v; }
 
 #define AT_D 64          // head dim (fixed)
-#define AT_QW 64         // query rows per wave
+#define AT_QW 32         // query rows per wave (8 thin waves)
+#define AT_QM (AT_QW / 16)  // m-fragments per wave
 #define AT_KT 32         // keys per inner tile
 #define AT_PLD 40        // P-bounce row stride (odd 16B slots)
 
@@ -36,7 +37,7 @@ extern __shared__ __attribute__((aligned(16))) char smem[];
 // LDS layout (dynamic): K [Sp][64] bf16 (row-swizzled like the GEMM
 // tiles), then Vt [64][Sp] bf16 (row padded +8 to break conflicts),
 // then P-bounce [4 waves][64][AT_KT] bf16.
-__global__ __launch_bounds__(256)
+__global__ __launch_bounds__(512)
 void attn_fused_kernel(const __bf16* __restrict__ Q,
                        const __bf16* __restrict__ K,
                        const __bf16* __restrict__ V,
@@ -61,12 +62,12 @@ void attn_fused_kernel(const __bf16* __restrict__ Q,
   const int vt_ld = Sp + 8;                  // Vt row stride (elements)
   __bf16* k_lds = (__bf16*)smem;             // [Sp][64], swizzled
   __bf16* vt_lds = k_lds + Sp * AT_D;        // [64][vt_ld]
-  __bf16* p_lds = vt_lds + AT_D * vt_ld;     // [4][64][AT_PLD]
+  __bf16* p_lds = vt_lds + AT_D * vt_ld;     // [8][AT_QW][AT_PLD]
 
   // ---- stage K (swizzled rows, like the GEMM B tile) and Vt ----
   // Each thread loads 16B chunks of K/V rows; K goes in with the
   // slot^=(row&7) involution so ds_read_b128 fragments are spread.
-  for (int idx = tid; idx < Sp * 8; idx += 256) {
+  for (int idx = tid; idx < Sp * 8; idx += 512) {
     int row = idx >> 3;          // key index
     int slot = idx & 7;          // 16B slot within the 128B row
     bf16x8 kv;
@@ -98,9 +99,9 @@ void attn_fused_kernel(const __bf16* __restrict__ Q,
   // ---- load this wave's Q fragments (A-operand layout) ----
   // A frag (16x16x32): lane holds A[row = l&15][k = (l>>4)*8 .. +7].
   const int q0 = wave * AT_QW;
-  bf16x8 qf[4][2];  // [m-frag][kk]
+  bf16x8 qf[AT_QM][2];  // [m-frag][kk]
 #pragma unroll
-  for (int m = 0; m < 4; ++m)
+  for (int m = 0; m < AT_QM; ++m)
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       int row = q0 + m * 16 + (lane & 15);
@@ -120,17 +121,17 @@ void attn_fused_kernel(const __bf16* __restrict__ Q,
 
   // ---- online-softmax state ----
   // D-frag layout: for (m, reg r): q-row = m*16 + (lane>>4)*4 + r.
-  float mrow[4][4], lrow[4][4];
+  float mrow[AT_QM][4], lrow[AT_QM][4];
 #pragma unroll
-  for (int m = 0; m < 4; ++m)
+  for (int m = 0; m < AT_QM; ++m)
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       mrow[m][r] = -3.4e38f;
       lrow[m][r] = 0.f;
     }
-  f32x4 oacc[4][4];  // [m][d-frag], D layout: col d = n*16 + (l&15)
+  f32x4 oacc[AT_QM][4];  // [m][d-frag]: col d = n*16 + (l&15)
 #pragma unroll
-  for (int m = 0; m < 4; ++m)
+  for (int m = 0; m < AT_QM; ++m)
 #pragma unroll
     for (int n = 0; n < 4; ++n) oacc[m][n] = {0.f, 0.f, 0.f, 0.f};
 
@@ -140,9 +141,9 @@ void attn_fused_kernel(const __bf16* __restrict__ Q,
   for (int t = 0; t < ntile; ++t) {
     const int kbase = t * AT_KT;
     // ---- S-block = Q @ K_t^T : acc [4m][2n] ----
-    f32x4 sacc[4][2];
+    f32x4 sacc[AT_QM][2];
 #pragma unroll
-    for (int m = 0; m < 4; ++m)
+    for (int m = 0; m < AT_QM; ++m)
 #pragma unroll
       for (int n = 0; n < 2; ++n) sacc[m][n] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
@@ -156,7 +157,7 @@ void attn_fused_kernel(const __bf16* __restrict__ Q,
         bf16x8 kf = *(bf16x8*)(k_lds + krow * AT_D +
                                ((slot ^ (krow & 7)) << 3));
 #pragma unroll
-        for (int m = 0; m < 4; ++m)
+        for (int m = 0; m < AT_QM; ++m)
           sacc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               qf[m][kk], kf, sacc[m][n], 0, 0, 0);
       }
@@ -167,11 +168,11 @@ void attn_fused_kernel(const __bf16* __restrict__ Q,
       int key = kbase + n * 16 + (lane & 15);
       if (key >= S)
 #pragma unroll
-        for (int m = 0; m < 4; ++m)
+        for (int m = 0; m < AT_QM; ++m)
           sacc[m][n] = {-3.4e38f, -3.4e38f, -3.4e38f, -3.4e38f};
     }
 #pragma unroll
-    for (int m = 0; m < 4; ++m)
+    for (int m = 0; m < AT_QM; ++m)
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         float mx = fmaxf(sacc[m][0][r], sacc[m][1][r]);
@@ -199,7 +200,7 @@ void attn_fused_kernel(const __bf16* __restrict__ Q,
     // write: q-row = m*16+(l>>4)*4+r, key-col = n*16+(l&15)
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 #pragma unroll
-    for (int m = 0; m < 4; ++m)
+    for (int m = 0; m < AT_QM; ++m)
 #pragma unroll
       for (int n = 0; n < 2; ++n)
 #pragma unroll
@@ -217,7 +218,7 @@ void attn_fused_kernel(const __bf16* __restrict__ Q,
       const __bf16* vp = vt_lds + drow * vt_ld + kbase + (lane >> 4) * 8;
       bf16x8 vf = *(const bf16x8*)(vp);
 #pragma unroll
-      for (int m = 0; m < 4; ++m) {
+      for (int m = 0; m < AT_QM; ++m) {
         // A frag: lane holds P[q = m*16 + (l&15)][k = (l>>4)*8 .. +7]
         bf16x8 pf = *(const bf16x8*)(pb + (m * 16 + (lane & 15)) * AT_PLD +
                                      (lane >> 4) * 8);
@@ -229,7 +230,7 @@ void attn_fused_kernel(const __bf16* __restrict__ Q,
 
   // ---- epilogue: O / l, store (strided output view) ----
 #pragma unroll
-  for (int m = 0; m < 4; ++m) {
+  for (int m = 0; m < AT_QM; ++m) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       int qr = q0 + m * 16 + (lane >> 4) * 4 + r;
@@ -253,9 +254,9 @@ void launch_attn_fused_bf16(const void* Q, const void* K, const void* V,
   int Sp = (S + AT_KT - 1) / AT_KT * AT_KT;
   int vt_ld = Sp + 8;
   size_t lds = (size_t)Sp * AT_D * 2 + (size_t)AT_D * vt_ld * 2 +
-               4 * AT_QW * AT_PLD * 2;
+               8 * AT_QW * AT_PLD * 2;
   dim3 grid(batch1 * batch2);
-  hipLaunchKernelGGL(attn_fused_kernel, grid, dim3(256), lds, stream,
+  hipLaunchKernelGGL(attn_fused_kernel, grid, dim3(512), lds, stream,
                      (const __bf16*)Q, (const __bf16*)K, (const __bf16*)V,
                      (__bf16*)O, batch2, S, Sp, sQ1, sQ2, lqd, sK1, sK2,
                      lkd, sV1, sV2, lvd, sO1, sO2, lod, scale);
