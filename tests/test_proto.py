@@ -80,3 +80,19 @@ def test_frozen_pb_file_roundtrip(tmp_path, lenet_pb):
 def test_truncated_raises():
     with pytest.raises(ValueError):
         list(wire.iter_fields(b"\x0a\xff"))
+
+
+def test_bf16_frozen_graph_roundtrip(tmp_path):
+    """bf16-stored weights in a frozen graph parse to f32 numpy."""
+    import torch
+
+    from tensorflow_web_deploy_amd.ckpt import wire
+    w = np.random.randn(4, 4).astype(np.float32)
+    wbf = torch.from_numpy(w).bfloat16()
+    raw = wbf.view(torch.uint16).numpy().tobytes()
+    enc = (wire.field_varint(1, gd.DT_BFLOAT16)
+           + wire.field_bytes(2, gd._encode_shape([4, 4]))
+           + wire.field_bytes(4, raw))
+    out = gd._parse_tensor(enc)
+    assert out.dtype == np.float32 and out.shape == (4, 4)
+    np.testing.assert_allclose(out, wbf.float().numpy(), rtol=0, atol=0)
