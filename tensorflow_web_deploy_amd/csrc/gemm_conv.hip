@@ -44,11 +44,6 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 extern __shared__ __attribute__((aligned(16))) char smem[];
 
-struct MagicDiv {
-  uint32_t m_lo, m_hi;  // 64-bit magic split (host packs m = ceil(2^40/d))
-  uint32_t pad0, pad1;
-};
-
 // n / d for n < 2^24, d < 2^16 with m = ceil(2^40/d)
 __device__ __forceinline__ uint32_t mdiv(uint32_t n, uint64_t m) {
   return (uint32_t)(((uint64_t)n * m) >> 40);

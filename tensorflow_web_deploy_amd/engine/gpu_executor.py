@@ -15,7 +15,6 @@ only cold glue (concat, reshape bookkeeping) touches torch ops.
 """
 from __future__ import annotations
 
-import math
 from typing import Callable, Dict, List, Optional
 
 import numpy as np

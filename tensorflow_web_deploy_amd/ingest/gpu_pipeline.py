@@ -16,7 +16,7 @@ tests/test_ingest_gpu.py.
 from __future__ import annotations
 
 import io
-from concurrent.futures import ProcessPoolExecutor, ThreadPoolExecutor
+from concurrent.futures import ProcessPoolExecutor
 from typing import List, Optional, Sequence
 
 import numpy as np

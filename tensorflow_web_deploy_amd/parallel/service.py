@@ -40,7 +40,6 @@ SLOTS_PER_WORKER = 128
 def _service_main(cfg: Config, in_q, resp_qs, err_q, ready_ev):
     try:
         import torch
-        from concurrent.futures import ThreadPoolExecutor
 
         from ..engine.service import GpuServingEngine, LocalEngine
         from ..sched.batcher import DynamicBatcher
