@@ -550,30 +550,52 @@ void gemm_conv_kernel_256(const __bf16* __restrict__ A,
     const char* Bhalf = smem + HALF_OFF(buf, 2 + (wc >> 1));
     const int brow0 = (wc & 1) * 64;  // 64-col strip within the B half
 
-#pragma unroll
-    for (int kk = 0; kk < 2; ++kk) {
-      bf16x8 bfr[4];
+    // register-pipelined fragment reads: group g+1's LDS reads issue
+    // before group g's MFMAs so the first MFMA of each group does not
+    // stall on lgkm (PMC: 37% of wave cycles parked)
+    {
+      bf16x8 bfr[2][4], afr[2][4];
 #pragma unroll
       for (int ni = 0; ni < 4; ++ni) {
         int row = brow0 + ni * 16 + (lane & 15);
-        bfr[ni] = *(bf16x8*)(Bhalf + lds_byte2(row, kk * 4 + (lane >> 4)));
+        bfr[0][ni] = *(bf16x8*)(Bhalf + lds_byte2(row, (lane >> 4)));
       }
 #pragma unroll
-      for (int mh = 0; mh < 2; ++mh) {
-        bf16x8 afr[4];
+      for (int q = 0; q < 4; ++q) {
+        int row = q * 16 + (lane & 15);
+        afr[0][q] = *(bf16x8*)(Ahalf + lds_byte2(row, (lane >> 4)));
+      }
 #pragma unroll
-        for (int q = 0; q < 4; ++q) {
-          int mi = mh * 4 + q;
-          int row = mi * 16 + (lane & 15);
-          afr[q] = *(bf16x8*)(Ahalf + lds_byte2(row, kk * 4 + (lane >> 4)));
+      for (int g = 0; g < 4; ++g) {         // (kk,mh) = g>>1, g&1
+        const int cur = g & 1;
+        if (g < 3) {
+          const int kk1 = (g + 1) >> 1;
+          const int mh1 = (g + 1) & 1;
+#pragma unroll
+          for (int q = 0; q < 4; ++q) {
+            int row = (mh1 * 4 + q) * 16 + (lane & 15);
+            afr[cur ^ 1][q] = *(bf16x8*)(Ahalf +
+                lds_byte2(row, kk1 * 4 + (lane >> 4)));
+          }
+          if (mh1 == 0) {
+#pragma unroll
+            for (int ni = 0; ni < 4; ++ni) {
+              int row = brow0 + ni * 16 + (lane & 15);
+              bfr[1][ni] = *(bf16x8*)(Bhalf +
+                  lds_byte2(row, kk1 * 4 + (lane >> 4)));
+            }
+          }
         }
+        const int kk = g >> 1;
+        const int mh = g & 1;
+        (void)kk;
         __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int q = 0; q < 4; ++q)
 #pragma unroll
           for (int ni = 0; ni < 4; ++ni)
             acc[mh * 4 + q][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                afr[q], bfr[ni], acc[mh * 4 + q][ni], 0, 0, 0);
+                afr[cur][q], bfr[g >> 1][ni], acc[mh * 4 + q][ni], 0, 0, 0);
         __builtin_amdgcn_s_setprio(0);
       }
     }
