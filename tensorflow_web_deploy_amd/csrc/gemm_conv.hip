@@ -242,20 +242,35 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
   };
 
   auto compute = [&](int buf) {
+    // register-pipelined: kk=1 fragments load before kk=0's MFMAs issue
+    bf16x8 af[2][4], bfr[2][4];
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+      int row = wr * 64 + mi * 16 + (lane & 15);
+      af[0][mi] = *(bf16x8*)(smem + a_off(buf) +
+                             lds_byte(row, (lane >> 4)));
+    }
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      int row = wc * 64 + ni * 16 + (lane & 15);
+      bfr[0][ni] = *(bf16x8*)(smem + b_off(buf) +
+                              lds_byte(row, (lane >> 4)));
+    }
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      bf16x8 af[4], bfr[4];
+      if (kk == 0) {
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi) {
-        int row = wr * 64 + mi * 16 + (lane & 15);
-        int col8 = kk * 4 + (lane >> 4);
-        af[mi] = *(bf16x8*)(smem + a_off(buf) + lds_byte(row, col8));
-      }
+        for (int mi = 0; mi < 4; ++mi) {
+          int row = wr * 64 + mi * 16 + (lane & 15);
+          af[1][mi] = *(bf16x8*)(smem + a_off(buf) +
+                                 lds_byte(row, 4 + (lane >> 4)));
+        }
 #pragma unroll
-      for (int ni = 0; ni < 4; ++ni) {
-        int row = wc * 64 + ni * 16 + (lane & 15);
-        int col8 = kk * 4 + (lane >> 4);
-        bfr[ni] = *(bf16x8*)(smem + b_off(buf) + lds_byte(row, col8));
+        for (int ni = 0; ni < 4; ++ni) {
+          int row = wc * 64 + ni * 16 + (lane & 15);
+          bfr[1][ni] = *(bf16x8*)(smem + b_off(buf) +
+                                  lds_byte(row, 4 + (lane >> 4)));
+        }
       }
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -263,7 +278,7 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
 #pragma unroll
         for (int ni = 0; ni < 4; ++ni)
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
+              af[kk][mi], bfr[kk][ni], acc[mi][ni], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
     }
   };
