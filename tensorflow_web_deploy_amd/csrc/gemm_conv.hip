@@ -844,8 +844,17 @@ static void dispatch(const void* A, const void* B, void* C, const float* bias,
   // DEPTH=3 measured WORSE here (120 KiB LDS -> 1 block/CU, losing the
   // cross-block overlap that was hiding half the staging latency):
   // stem 537->351 TF, s1 3x3 500->325. Keep DEPTH=2 + 2 blocks/CU.
-  if (N <= 96)
+  if (N <= 96) {
     launch_tile<256, 64, MODE, 2>(A, B, C, bias, residual, M, N, K, Kp,
+                                  a_bytes, act, cp, stream, ws, splitk, ldc);
+    return;
+  }
+  // mid-size M underfills the chip at 128^2 (e.g. Inception 17x17
+  // factorized convs: 290 blocks = ~1 block/CU); halve the M-tile to
+  // double the grid when it buys >= ~1.5 blocks/CU
+  long b128 = ((M + BM - 1) / BM) * (long)((N + BN - 1) / BN);
+  if (splitk == 1 && b128 >= 128 && b128 < 320 && getenv("TWD_NO_6412") == nullptr)
+    launch_tile<64, 128, MODE, 2>(A, B, C, bias, residual, M, N, K, Kp,
                                   a_bytes, act, cp, stream, ws, splitk, ldc);
   else
     launch_tile<128, 128, MODE, 2>(A, B, C, bias, residual, M, N, K, Kp,
