@@ -823,6 +823,18 @@ static void dispatch(const void* A, const void* B, void* C, const float* bias,
                      int64_t a_bytes, int act, const ConvParams& cp,
                      hipStream_t stream, float* ws = nullptr,
                      int splitk = 1, int ldc = 0) {
+  // 256^2-class shapes that underfill at 256^2 tiles (e.g. ViT
+  // projections: M=12544, N=768 -> 147 blocks on 256 CUs) get a
+  // (128,256) tile instead: same 8-wave pipeline, twice the blocks
+  {
+    long b256 = ((M + BM2 - 1) / BM2) * (long)((N + BN2 - 1) / BN2);
+    if (M >= 192 && N >= 192 && K >= 128 && b256 < b256_min()
+        && 2 * b256 >= 256 && getenv("TWD_NO_128256") == nullptr) {
+      launch_tile<128, 256, MODE, 2>(A, B, C, bias, residual, M, N, K, Kp,
+                                     a_bytes, act, cp, stream, ws, 1, ldc);
+      return;
+    }
+  }
   if (use_256(M, N, K)) {
     int tiles_m = (M + BM2 - 1) / BM2;
     int tiles_n = (N + BN2 - 1) / BN2;
