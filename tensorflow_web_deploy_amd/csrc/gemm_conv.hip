@@ -686,6 +686,258 @@ void gemm_conv_kernel_256(const __bf16* __restrict__ A,
   }
 }
 
+// ================================================================
+// 256x256 PHASED variant (env TWD_256P=1): per K-tile, 4 phases of
+// {quadrant ds_reads + one half-tile stage + barrier + lgkm + 16 MFMA +
+// barrier} — the cdna_hip_programming.md §5 8-phase structure's
+// role-split (T5's prerequisite), with B fragments held in registers
+// across the tile and vmcnt counted once per K-tile.
+// ================================================================
+template <int MODE, bool HAS_RES>
+__global__ __launch_bounds__(NTHREADS2, 2)
+void gemm_conv_kernel_256p(const __bf16* __restrict__ A,
+                           const __bf16* __restrict__ B,
+                           __bf16* __restrict__ Cout,
+                           const float* __restrict__ bias,
+                           const __bf16* __restrict__ residual,
+                           int M, int N, int K, int Kp,
+                           int a_bytes, int tiles_m, int tiles_n,
+                           int relu, ConvParams cp, int ldc = 0) {
+  if (ldc == 0) ldc = N;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = __builtin_amdgcn_readfirstlane(tid >> 6);
+
+  int nwg = tiles_m * tiles_n;
+  int bid = blockIdx.x;
+  {
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = bid & 7, off = bid >> 3;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
+  }
+  const int tn = bid / tiles_m;
+  const int tm = bid - tn * tiles_m;
+  const int row0 = tm * BM2;
+  const int col0 = tn * BN2;
+
+  const auto rsrcA = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)A, 0, a_bytes, 0x00020000);
+  const auto rsrcB = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)B, 0, (int)((int64_t)N * Kp * 2), 0x00020000);
+
+  const int srow_base = lane >> 3;
+  const int sslot = lane & 7;
+  const int swz = srow_base & 7;
+  int32_t a_base[4], a_hb[4], a_wb[4];
+  bool a_valid[4];
+  int32_t b_voff[4];
+  bool b_valid[4];
+#pragma unroll
+  for (int c2 = 0; c2 < 4; ++c2) {
+    int h = c2 >> 1, i = c2 & 1;
+    int row_l = h * 128 + (wave * 2 + i) * 8 + srow_base;
+    int row_g = row0 + row_l;
+    if (MODE == 0) {
+      a_valid[c2] = row_g < M;
+      a_base[c2] = row_g * K;
+    } else {
+      a_valid[c2] = row_g < M;
+      uint32_t rg = (uint32_t)min(row_g, max(M - 1, 0));
+      uint32_t n_img = mdiv(rg, cp.magic_HW);
+      uint32_t rem = rg - n_img * (uint32_t)(cp.HO * cp.WO);
+      uint32_t ho = mdiv(rem, cp.magic_WO);
+      uint32_t wo = rem - ho * (uint32_t)cp.WO;
+      a_hb[c2] = (int)ho * cp.sh - cp.pt;
+      a_wb[c2] = (int)wo * cp.sw - cp.pl;
+      a_base[c2] = (((int)n_img * cp.H + a_hb[c2]) * cp.W + a_wb[c2]) * cp.C;
+    }
+    int col_g = col0 + row_l;
+    b_valid[c2] = col_g < N;
+    b_voff[c2] = col_g * Kp;
+  }
+
+  const int wr = wave >> 2, wc = wave & 3;
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      float b0 = 0.f;
+      if (bias) {
+        int col = col0 + wc * 64 + ni * 16 + (lane & 15);
+        if (col < N) b0 = bias[col];
+      }
+      acc[mi][ni] = {b0, b0, b0, b0};
+    }
+
+  const int nt = (K + BK - 1) / BK;
+
+  auto a_voffset = [&](int c2, int k) -> int32_t {
+    if (MODE == 0)
+      return (a_valid[c2] && k < K) ? (a_base[c2] + k) * 2 : 0x7ffffff0;
+    uint32_t rs = mdiv((uint32_t)k, cp.magic_C);
+    uint32_t c = (uint32_t)k - rs * (uint32_t)cp.C;
+    uint32_t r = mdiv(rs, cp.magic_S);
+    uint32_t sx = rs - r * (uint32_t)cp.S;
+    int hi = a_hb[c2] + (int)r;
+    int wi = a_wb[c2] + (int)sx;
+    bool ok = a_valid[c2] && k < K && (uint32_t)hi < (uint32_t)cp.H &&
+              (uint32_t)wi < (uint32_t)cp.W;
+    return ok ? (a_base[c2] + ((int)r * cp.W + (int)sx) * cp.C + (int)c) * 2
+              : 0x7ffffff0;
+  };
+
+  auto stage_half = [&](int h, int t) {
+    const int kb = t * BK;
+    const int k = kb + ((sslot ^ swz) << 3);
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int32_t voff;
+      if (h < 2) {
+        voff = a_voffset(h * 2 + i, k);
+      } else {
+        int c2 = (h - 2) * 2 + i;
+        voff = b_valid[c2] ? (b_voff[c2] + k) * 2 : 0x7ffffff0;
+      }
+      __builtin_amdgcn_raw_ptr_buffer_load_lds(
+          h < 2 ? rsrcA : rsrcB,
+          (__attribute__((address_space(3))) void*)(
+              smem + HALF_OFF(t & 1, h) + (wave * 2 + i) * 1024),
+          16, voff, 0, 0, 0);
+    }
+  };
+
+  auto lds_byte2 = [](int row, int col8) {
+    return row * (BK * 2) + ((col8 ^ (row & 7)) << 4);
+  };
+
+  stage_half(0, 0); stage_half(1, 0); stage_half(2, 0); stage_half(3, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  for (int t = 0; t < nt; ++t) {
+    const int buf = t & 1;
+    const bool pf = (t + 1) < nt;
+    const char* Ahalf = smem + HALF_OFF(buf, wr);
+    const char* Bhalf = smem + HALF_OFF(buf, 2 + (wc >> 1));
+    const int brow0 = (wc & 1) * 64;
+
+    bf16x8 bfr[2][4];  // [kk][ni], held across all 4 phases
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      // quadrant p: m-frags {2p, 2p+1}, both kk
+      bf16x8 afr[2][2];  // [mi-sub][kk]
+      if (p == 0) {
+        // verify tile t resident BEFORE any LDS read of this buffer
+        if (pf) {
+          stage_half(0, t + 1);
+          asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+        } else {
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
+        __builtin_amdgcn_s_barrier();
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+          for (int ni = 0; ni < 4; ++ni) {
+            int row = brow0 + ni * 16 + (lane & 15);
+            bfr[kk][ni] = *(bf16x8*)(Bhalf +
+                lds_byte2(row, kk * 4 + (lane >> 4)));
+          }
+      }
+#pragma unroll
+      for (int q = 0; q < 2; ++q)
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk) {
+          int row = (p * 2 + q) * 16 + (lane & 15);
+          afr[q][kk] = *(bf16x8*)(Ahalf +
+              lds_byte2(row, kk * 4 + (lane >> 4)));
+        }
+      if (pf && p > 0) stage_half(p, t + 1);
+      if (p > 0) __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int q = 0; q < 2; ++q)
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+          for (int ni = 0; ni < 4; ++ni)
+            acc[p * 2 + q][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afr[q][kk], bfr[kk][ni], acc[p * 2 + q][ni], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  __bf16* cl = (__bf16*)smem;
+#pragma unroll
+  for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = wr * 128 + mi * 16 + (lane >> 4) * 4 + r;
+        int col = wc * 64 + ni * 16 + (lane & 15);
+        cl[row * BN2 + col] = (__bf16)acc[mi][ni][r];
+      }
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  const int wrow = tid >> 5;
+  const int wcol8 = (tid & 31) * 8;
+#pragma unroll
+  for (int pp = 0; pp < 16; ++pp) {
+    int row_l = pp * 16 + wrow;
+    int row_g = row0 + row_l;
+    int col_g = col0 + wcol8;
+    if (row_g < M && col_g < N) {
+      bf16x8 v = *(bf16x8*)(cl + row_l * BN2 + wcol8);
+      const int nvalid = min(8, N - col_g);
+      if (HAS_RES) {
+        if (nvalid == 8) {
+          bf16x8 rv = *(const bf16x8*)(residual + (int64_t)row_g * ldc +
+                                       col_g);
+#pragma unroll
+          for (int e = 0; e < 8; ++e)
+            v[e] = (__bf16)(bf2f(v[e]) + bf2f(rv[e]));
+        } else {
+#pragma unroll
+          for (int e = 0; e < 8; ++e)
+            if (e < nvalid)
+              v[e] = (__bf16)(bf2f(v[e]) +
+                              bf2f(residual[(int64_t)row_g * ldc + col_g +
+                                            e]));
+        }
+      }
+      if (relu == 1) {
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          v[e] = bf2f(v[e]) > 0.f ? v[e] : (__bf16)0.f;
+      } else if (relu == 5) {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          float f = bf2f(v[e]);
+          v[e] = (__bf16)(0.5f * f * (1.f + erff(f * 0.70710678f)));
+        }
+      }
+      if (nvalid == 8) {
+        *(bf16x8*)(Cout + (int64_t)row_g * ldc + col_g) = v;
+      } else {
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          if (e < nvalid) Cout[(int64_t)row_g * ldc + col_g + e] = v[e];
+      }
+    }
+  }
+}
+
+static bool use_256p() {
+  static bool v = getenv("TWD_256P") != nullptr;
+  return v;
+}
+
 // ---------------------------------------------------------------- host
 
 static uint64_t magic40(uint32_t d) {
@@ -840,6 +1092,21 @@ static void dispatch(const void* A, const void* B, void* C, const float* bias,
     int tiles_m = (M + BM2 - 1) / BM2;
     int tiles_n = (N + BN2 - 1) / BN2;
     dim3 grid(tiles_m * tiles_n);
+    if (use_256p()) {
+      if (residual)
+        hipLaunchKernelGGL((gemm_conv_kernel_256p<MODE, true>), grid,
+                           dim3(NTHREADS2), LDS2_BYTES, stream,
+                           (const __bf16*)A, (const __bf16*)B, (__bf16*)C,
+                           bias, (const __bf16*)residual, M, N, K, Kp,
+                           (int)a_bytes, tiles_m, tiles_n, act, cp, ldc);
+      else
+        hipLaunchKernelGGL((gemm_conv_kernel_256p<MODE, false>), grid,
+                           dim3(NTHREADS2), LDS2_BYTES, stream,
+                           (const __bf16*)A, (const __bf16*)B, (__bf16*)C,
+                           bias, (const __bf16*)nullptr, M, N, K, Kp,
+                           (int)a_bytes, tiles_m, tiles_n, act, cp, ldc);
+      return;
+    }
     if (residual)
       hipLaunchKernelGGL((gemm_conv_kernel_256<MODE, true>), grid,
                          dim3(NTHREADS2), LDS2_BYTES, stream,
