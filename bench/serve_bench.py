@@ -92,6 +92,8 @@ def main():
     p.add_argument("--max-wait-ms", type=float, default=2.0)
     p.add_argument("--http-workers", type=int, default=0,
                    help=">0: multi-process SO_REUSEPORT serving stack")
+    p.add_argument("--fast", action="store_true",
+                   help="C++ epoll frontend + batch loop (server/fast.py)")
     args = p.parse_args()
 
     import torch
@@ -114,7 +116,28 @@ def main():
                  gpus=args.gpus, port=args.port)
     mserver = None
     batcher = None
-    if args.http_workers > 0:
+    fast = None
+    if args.fast:
+        import dataclasses
+        from tensorflow_web_deploy_amd.server.fast import FastServer
+        cfg = dataclasses.replace(cfg, host="127.0.0.1")
+        fast = FastServer(cfg, workers=3)
+        backend = fast.engine
+        b = 1
+        while True:
+            backend.infer(np.zeros((b, h, w, cfg.input_channels),
+                                   np.float32))
+            if b >= args.max_batch:
+                break
+            b = min(b * 2, args.max_batch)
+        if hasattr(backend, "pre"):
+            backend.pre.warm_pool()
+        import gc
+        gc.collect()
+        gc.freeze()
+        gc.set_threshold(200000, 100, 100)
+        fast.start_background()
+    elif args.http_workers > 0:
         mserver = MultiprocServer(cfg, http_workers=args.http_workers)
     elif args.gpus > 1:
         from tensorflow_web_deploy_amd.parallel.replicas import \
@@ -125,7 +148,7 @@ def main():
         backend = GpuServingEngine(cfg)
     else:
         backend = LocalEngine(cfg)
-    if mserver is None:
+    if mserver is None and fast is None:
         # warm the engine (compile + HIP-graph capture for every bucket)
         b = 1
         while True:
@@ -209,11 +232,14 @@ def main():
         "p99_ms": round(1000 * ls[int(len(ls) * 0.99)], 2) if ls else None,
         "requests": n,
         "http_workers": args.http_workers,
+        "fast": bool(args.fast),
         "batch_trace_tail": batcher.trace[-3:] if batcher else None,
     }
     print(json.dumps(out))
     if mserver is not None:
         mserver.shutdown()
+    if fast is not None:
+        fast.stop()
 
 
 if __name__ == "__main__":

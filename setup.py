@@ -31,10 +31,17 @@ ext = cpp_extension.CUDAExtension(
     },
 )
 
+# Pure-C++ epoll HTTP frontend (no torch/HIP): serving fast path
+httpd = cpp_extension.CppExtension(
+    name="tensorflow_web_deploy_amd._httpd",
+    sources=[os.path.join(CSRC, "httpd.cpp")],
+    extra_compile_args={"cxx": ["-O2", "-std=c++17"]},
+)
+
 setup(
     name="tensorflow_web_deploy_amd",
     version="0.1.0",
     packages=["tensorflow_web_deploy_amd"],
-    ext_modules=[ext],
+    ext_modules=[ext, httpd],
     cmdclass={"build_ext": cpp_extension.BuildExtension},
 )

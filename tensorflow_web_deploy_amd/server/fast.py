@@ -1,0 +1,173 @@
+"""Batch-loop server over the C++ epoll frontend (`csrc/httpd.cpp`).
+
+The Flask app (server/app.py) keeps the full reference-parity surface;
+this module serves the same `POST /predict` contract through a C++
+frontend that accepts, parses, batches and responds WITHOUT the GIL.
+The Python process runs one loop: wait for a whole batch (GIL released
+inside C++), decode+infer it, hand back responses. No per-request
+Python ever runs, which removes the werkzeug/batcher GIL contention
+measured in BASELINE.md (batch work 0.3-3 ms alone vs 73-91 ms in situ
+under Flask).
+
+  python -m tensorflow_web_deploy_amd.server.fast --model resnet50 ...
+"""
+from __future__ import annotations
+
+import json
+import threading
+from typing import Optional
+
+import numpy as np
+
+from ..conf.config import Config
+from .labels import load_labels
+
+
+class FastServer:
+    def __init__(self, cfg: Config, engine=None, workers: int = 3,
+                 port: Optional[int] = None):
+        from .._httpd import Frontend
+        self.cfg = cfg
+        self.labels = load_labels(cfg.labels_path, cfg.num_classes)
+        if engine is None:
+            engine = _default_engine(cfg)
+        self.engine = engine
+        self.fe = Frontend(cfg.host or "0.0.0.0",
+                           cfg.port if port is None else port,
+                           workers, 8192)
+        self.port = self.fe.port()
+        self._stop = threading.Event()
+        self.model_name = cfg.model_name or cfg.model_path
+
+    # -- one batch ----------------------------------------------------
+    def _infer(self, bodies):
+        """bodies -> list of (status, payload-json-str); isolates bad
+        images so one corrupt upload cannot fail the whole batch."""
+        eng = self.engine
+        try:
+            if hasattr(eng, "infer_blobs"):
+                probs = eng.infer_blobs(bodies)
+            else:
+                arrs = [_decode(b, self.cfg) for b in bodies]
+                probs = eng.infer(np.stack(arrs, 0))
+            return [(200, self._payload(p)) for p in probs]
+        except Exception:
+            pass
+        # slow path: find the undecodable items, serve the rest
+        out = [None] * len(bodies)
+        good, idx = [], []
+        for i, b in enumerate(bodies):
+            try:
+                good.append(_decode(b, self.cfg, raw=hasattr(
+                    eng, "infer_blobs")))
+                idx.append(i)
+            except Exception as e:
+                out[i] = (400, json.dumps(
+                    {"error": f"cannot decode image: {e}"}))
+        if good:
+            try:
+                if hasattr(eng, "infer_arrays"):
+                    probs = eng.infer_arrays(good)
+                else:
+                    probs = eng.infer(np.stack(good, 0))
+                for i, p in zip(idx, probs):
+                    out[i] = (200, self._payload(p))
+            except Exception as e:  # engine failure: honest 500s
+                for i in idx:
+                    out[i] = (500, json.dumps({"error": str(e)}))
+        return out
+
+    def _payload(self, probs) -> str:
+        k = min(self.cfg.top_k, len(probs))
+        if k < len(probs):
+            part = np.argpartition(probs, -k)[-k:]
+            top = part[np.argsort(probs[part])[::-1]]
+        else:
+            top = np.argsort(probs)[::-1]
+        return json.dumps({
+            "predictions": [
+                {"label": self.labels[int(i)], "class_id": int(i),
+                 "prob": float(probs[int(i)])} for i in top
+            ],
+            "model": self.model_name,
+        })
+
+    # -- loop ---------------------------------------------------------
+    def step(self, timeout_ms: Optional[float] = None) -> int:
+        """One wait_batch -> infer -> respond cycle; returns batch size."""
+        batch = self.fe.wait_batch(self.cfg.max_batch,
+                                   self.cfg.max_wait_ms)
+        if not batch:
+            return 0
+        results = self._infer([b for (_, b) in batch])
+        self.fe.respond_many([
+            (rid, st, body.encode())
+            for (rid, _), (st, body) in zip(batch, results)
+        ])
+        return len(batch)
+
+    def serve_forever(self):
+        try:
+            while not self._stop.is_set():
+                self.step()
+        finally:
+            self.fe.stop()
+
+    def start_background(self) -> threading.Thread:
+        t = threading.Thread(target=self.serve_forever, daemon=True)
+        t.start()
+        return t
+
+    def stop(self):
+        self._stop.set()
+        self.fe.stop()
+
+
+def _decode(data: bytes, cfg: Config, raw: bool = False):
+    """bytes -> uint8 HWC (raw=True: original size for infer_arrays)
+    or f32 model-input array."""
+    if raw:
+        import io
+
+        from PIL import Image
+        img = Image.open(io.BytesIO(data))
+        img = img.convert("L" if cfg.input_channels == 1 else "RGB")
+        a = np.asarray(img, dtype=np.uint8)
+        return a[:, :, None] if a.ndim == 2 else a
+    from ..ingest.preprocess import decode_image
+    return decode_image(data, (cfg.input_width, cfg.input_height),
+                        cfg.input_channels, cfg.normalize)
+
+
+def _default_engine(cfg: Config):
+    import torch
+    if cfg.gpus > 1:
+        from ..parallel.replicas import ReplicatedEngine
+        return ReplicatedEngine(cfg)
+    if torch.cuda.is_available():
+        from ..engine.service import GpuServingEngine
+        return GpuServingEngine(cfg)
+    from ..engine.service import LocalEngine
+    return LocalEngine(cfg)
+
+
+def main(argv=None):
+    import argparse
+    import gc
+
+    from ..conf.config import add_config_args, config_from_args
+    p = argparse.ArgumentParser("tensorflow_web_deploy_amd fast server")
+    add_config_args(p)
+    p.add_argument("--http-workers", type=int, default=3)
+    args = p.parse_args(argv)
+    cfg = config_from_args(args)
+    srv = FastServer(cfg, workers=args.http_workers)
+    gc.collect()
+    gc.freeze()
+    gc.set_threshold(200000, 100, 100)
+    print(f"[fast] serving {srv.model_name} on :{srv.port}", flush=True)
+    srv.serve_forever()
+
+
+if __name__ == "__main__":
+    main()

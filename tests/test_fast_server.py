@@ -1,0 +1,132 @@
+"""C++ epoll frontend (_httpd) + FastServer loop, CPU-only.
+
+Exercises the real TCP path: raw-body POST /predict, keep-alive reuse,
+inline /healthz + 404, bad-image isolation within a batch, and a
+concurrent multi-client burst (the GIL-contention scenario the frontend
+exists to remove — BASELINE.md serving analysis)."""
+import http.client
+import io
+import json
+import threading
+import urllib.error
+import urllib.request
+
+import numpy as np
+import pytest
+from PIL import Image
+
+from tensorflow_web_deploy_amd.conf.config import Config
+from tensorflow_web_deploy_amd.engine.service import FakeEngine
+
+
+def _png_bytes(w=28, h=28):
+    arr = (np.random.rand(h, w) * 255).astype("uint8")
+    img = Image.fromarray(arr, "L")
+    buf = io.BytesIO()
+    img.save(buf, "PNG")
+    return buf.getvalue()
+
+
+@pytest.fixture
+def server():
+    from tensorflow_web_deploy_amd.server.fast import FastServer
+    cfg = Config(model_name="lenet", device="cpu", host="127.0.0.1",
+                 max_batch=8, max_wait_ms=5)
+    srv = FastServer(cfg, engine=FakeEngine(10), workers=2, port=0)
+    srv.start_background()
+    yield srv
+    srv.stop()
+
+
+def _post(port, body, path="/predict", timeout=10):
+    req = urllib.request.Request(
+        f"http://127.0.0.1:{port}{path}", data=body,
+        headers={"Content-Type": "application/octet-stream"})
+    with urllib.request.urlopen(req, timeout=timeout) as r:
+        return r.status, json.loads(r.read())
+
+
+def test_predict_roundtrip(server):
+    status, j = _post(server.port, _png_bytes())
+    assert status == 200
+    assert len(j["predictions"]) == 5
+    p0 = j["predictions"][0]
+    assert set(p0) == {"label", "class_id", "prob"}
+    probs = [p["prob"] for p in j["predictions"]]
+    assert probs == sorted(probs, reverse=True)
+
+
+def test_healthz_inline_and_404(server):
+    with urllib.request.urlopen(
+            f"http://127.0.0.1:{server.port}/healthz", timeout=10) as r:
+        assert json.loads(r.read())["status"] == "ok"
+    with pytest.raises(urllib.error.HTTPError) as ei:
+        urllib.request.urlopen(
+            f"http://127.0.0.1:{server.port}/nope", timeout=10)
+    assert ei.value.code == 404
+
+
+def test_keep_alive_reuse(server):
+    conn = http.client.HTTPConnection("127.0.0.1", server.port, timeout=10)
+    try:
+        for _ in range(3):
+            conn.request("POST", "/predict", body=_png_bytes())
+            r = conn.getresponse()
+            assert r.status == 200
+            json.loads(r.read())  # must fully drain to reuse
+    finally:
+        conn.close()
+
+
+def test_bad_image_isolated(server):
+    """A corrupt upload 400s; a good one in the same window still 200s."""
+    results = {}
+
+    def go(name, body):
+        try:
+            results[name] = _post(server.port, body)[0]
+        except urllib.error.HTTPError as e:
+            results[name] = e.code
+
+    ts = [threading.Thread(target=go, args=("bad", b"not an image")),
+          threading.Thread(target=go, args=("good", _png_bytes()))]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join(timeout=15)
+    assert results["bad"] == 400
+    assert results["good"] == 200
+
+
+def test_concurrent_burst(server):
+    ok = []
+    lock = threading.Lock()
+
+    def client(n):
+        for _ in range(n):
+            status, j = _post(server.port, _png_bytes())
+            assert status == 200 and len(j["predictions"]) == 5
+            with lock:
+                ok.append(1)
+
+    ts = [threading.Thread(target=client, args=(5,)) for _ in range(8)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join(timeout=30)
+    assert len(ok) == 40
+    st = server.fe.stats()
+    assert st["responded"] >= 40
+
+
+def test_stats_and_stop():
+    from tensorflow_web_deploy_amd.server.fast import FastServer
+    cfg = Config(model_name="lenet", device="cpu", host="127.0.0.1",
+                 max_batch=4, max_wait_ms=2)
+    srv = FastServer(cfg, engine=FakeEngine(10), workers=1, port=0)
+    t = srv.start_background()
+    _post(srv.port, _png_bytes())
+    assert srv.fe.stats()["requests"] >= 1
+    srv.stop()
+    t.join(timeout=10)
+    assert not t.is_alive()
