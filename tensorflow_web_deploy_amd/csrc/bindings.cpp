@@ -315,4 +315,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fused_bf16", &attn_fused_bf16);
   m.def("resize_norm", &resize_norm);
   m.def("resize_norm_batch", &resize_norm_batch);
+  // Pin an externally-allocated host buffer (e.g. a /dev/shm mapping the
+  // decode workers write into) so H2D from it is a true async DMA.
+  m.def("host_register", [](int64_t ptr, int64_t bytes) {
+    hipError_t e = hipHostRegister(reinterpret_cast<void*>(ptr),
+                                   (size_t)bytes, hipHostRegisterDefault);
+    return e == hipSuccess;
+  });
+  m.def("host_unregister", [](int64_t ptr) {
+    (void)hipHostUnregister(reinterpret_cast<void*>(ptr));
+  });
 }

@@ -42,6 +42,205 @@ def _decode_worker(args):
     return arr
 
 
+# --- shared-memory decode pool ---------------------------------------
+# Measured (round 1): pool.map that returns decoded arrays costs ~56 ms
+# of pure IPC per 64-image batch (72 ms round-trip vs 52 ms SERIAL
+# decode), and ProcessPoolExecutor dispatch ALONE costs 20-40 ms per 64
+# no-op tasks. So this pool uses neither: persistent worker processes
+# share one mmap'd /dev/shm file (spawn children reattach by path — no
+# resource_tracker involvement). Per batch the parent writes a task
+# table into the mapping and wakes each worker with ONE pipe byte;
+# worker j decodes tasks j, j+K, ... straight into fixed-size pixel
+# slots and answers with one byte. Only 2 syscalls per worker per
+# batch; blobs out and pixels back never touch a pipe. The parent's
+# view of the slot area is hipHostRegister'ed once so the H2D of the
+# slot prefix is a true pinned-memory DMA.
+#
+# Mapping layout:
+#   [task table: slots x 2 int64 (in_off, in_len)]
+#   [meta table: slots x 4 int32 (h, w, c, status)]
+#   [input area: packed request blobs]
+#   [slot area:  slots x slot_bytes decoded pixels]
+
+_SHM = {"mm": None, "path": None}
+
+
+def _shm_attach(path):
+    import mmap
+    f = open(path, "r+b")
+    mm = mmap.mmap(f.fileno(), 0)
+    _SHM["mm"] = np.frombuffer(mm, dtype=np.uint8)
+    _SHM["_keep"] = (f, mm)
+    return _SHM["mm"]
+
+
+def _ring_worker(path, widx, nworkers, channels, slot_bytes, slots,
+                 in_bytes, wake_rx, done_tx):
+    """Persistent decode worker: wake byte -> decode my stripe -> done
+    byte. Runs in its own process; PIL decode releases the GIL there."""
+    import gc
+    gc.collect()
+    gc.freeze()
+    from PIL import Image
+    buf = _shm_attach(path)
+    tasks = np.frombuffer(buf[:slots * 16], dtype=np.int64)
+    meta = np.frombuffer(buf[slots * 16:slots * 16 + slots * 16],
+                         dtype=np.int32)
+    base_in = slots * 32
+    base_out = base_in + in_bytes
+    while True:
+        try:
+            msg = wake_rx.recv_bytes()
+        except EOFError:
+            return
+        if msg == b"q":
+            return
+        n = int(tasks[0])  # slot 0 of the task table carries the count
+        for i in range(widx, n, nworkers):
+            in_off = int(tasks[2 * (i + 1)])
+            in_len = int(tasks[2 * (i + 1) + 1])
+            try:
+                img = Image.open(io.BytesIO(
+                    bytes(buf[base_in + in_off:base_in + in_off + in_len])))
+                img = img.convert("L" if channels == 1 else "RGB")
+                arr = np.asarray(img, dtype=np.uint8)
+                if arr.ndim == 2:
+                    arr = arr[:, :, None]
+                if arr.nbytes > slot_bytes:
+                    meta[4 * i:4 * i + 4] = (0, 0, 0, 2)  # too big
+                    continue
+                o = base_out + i * slot_bytes
+                buf[o:o + arr.nbytes] = arr.reshape(-1)
+                meta[4 * i:4 * i + 4] = (arr.shape[0], arr.shape[1],
+                                         arr.shape[2], 0)
+            except Exception:
+                meta[4 * i:4 * i + 4] = (0, 0, 0, 1)  # bad image
+        try:
+            done_tx.send_bytes(b"d")
+        except (BrokenPipeError, OSError):
+            return
+
+
+class ShmDecodePool:
+    """Decode JPEG/PNG blobs into fixed-size shared-memory slots.
+
+    decode_batch() returns int32 meta rows (slot_byte_offset, h, w, c)
+    relative to the slot area, directly consumable by
+    resize_norm_batch; returns None when the batch doesn't fit the
+    layout or a worker died (caller uses the legacy pipe path);
+    raises ValueError naming the first undecodable image.
+    """
+
+    def __init__(self, channels: int, workers: Optional[int] = None,
+                 slot_bytes: int = 1 << 19, slots: int = 256,
+                 in_bytes: int = 64 << 20):
+        import multiprocessing as mp
+        import os
+        import tempfile
+        self.c = channels
+        self.slot_bytes = slot_bytes
+        self.slots = slots
+        self.in_bytes = in_bytes
+        self.hdr = slots * 32
+        total = self.hdr + in_bytes + slots * slot_bytes
+        fd, self.path = tempfile.mkstemp(prefix="twd_dec_",
+                                         dir="/dev/shm")
+        os.ftruncate(fd, total)
+        os.close(fd)
+        self.buf = _shm_attach(self.path)
+        self.buf[:: 4096] = 0  # pre-touch: fault every page once, now
+        self._tasks = np.frombuffer(self.buf[:slots * 16], dtype=np.int64)
+        self._meta = np.frombuffer(
+            self.buf[slots * 16:slots * 32], dtype=np.int32)
+        if workers is None:
+            workers = min(32, max(4, (os.cpu_count() or 8) - 2))
+        self.workers = workers
+        ctx = mp.get_context("spawn")
+        self._wake, self._done, self._procs = [], [], []
+        for j in range(workers):
+            wr, ww = ctx.Pipe(duplex=False)   # parent -> worker
+            dr, dw = ctx.Pipe(duplex=False)   # worker -> parent
+            pr = ctx.Process(
+                target=_ring_worker,
+                args=(self.path, j, workers, channels, slot_bytes,
+                      slots, in_bytes, wr, dw),
+                daemon=True)
+            pr.start()
+            wr.close()
+            dw.close()
+            self._wake.append(ww)
+            self._done.append(dr)
+            self._procs.append(pr)
+        self._dead = False
+        self._warm = False
+
+    def warm(self, timeout: float = 60.0):
+        if self._warm:
+            return
+        from PIL import Image
+        b = io.BytesIO()
+        Image.new("RGB" if self.c == 3 else "L", (8, 8)).save(b, "PNG")
+        self.decode_batch([b.getvalue()] * min(self.slots,
+                                               2 * self.workers),
+                          timeout=timeout)
+        self._warm = True
+
+    def decode_batch(self, blobs, timeout: float = 30.0):
+        if self._dead:
+            return None
+        n = len(blobs)
+        if (n == 0 or n + 1 > self.slots or
+                sum(len(b) for b in blobs) > self.in_bytes):
+            return None
+        self._tasks[0] = n
+        off = 0
+        base_in = self.hdr
+        for i, b in enumerate(blobs):
+            lb = len(b)
+            self.buf[base_in + off:base_in + off + lb] = \
+                np.frombuffer(b, np.uint8)
+            self._tasks[2 * (i + 1)] = off
+            self._tasks[2 * (i + 1) + 1] = lb
+            off += lb
+        nw = min(self.workers, n)
+        for j in range(nw):
+            self._wake[j].send_bytes(b"g")
+        for j in range(nw):
+            if not self._done[j].poll(timeout):
+                self._dead = True  # worker gone: never trust the ring
+                return None
+            self._done[j].recv_bytes()
+        meta = np.empty((n, 4), np.int32)
+        m = self._meta
+        for i in range(n):
+            st = m[4 * i + 3]
+            if st == 1:
+                raise ValueError(f"cannot decode image {i}")
+            if st == 2:
+                return None  # exceeds a slot: legacy path handles it
+            # offsets RELATIVE to the slot area: the H2D copies the
+            # slot prefix only, so the kernel sees that base
+            meta[i] = (i * self.slot_bytes, m[4 * i], m[4 * i + 1],
+                       m[4 * i + 2])
+        return meta
+
+    def close(self):
+        import os
+        for w in self._wake:
+            try:
+                w.send_bytes(b"q")
+            except (BrokenPipeError, OSError):
+                pass
+        for pr in self._procs:
+            pr.join(timeout=1.0)
+            if pr.is_alive():
+                pr.terminate()
+        try:
+            os.unlink(self.path)
+        except OSError:
+            pass
+
+
 class GpuPreprocessor:
     def __init__(self, height: int, width: int, channels: int,
                  normalize: str = "unit", device: str = "cuda",
@@ -64,6 +263,9 @@ class GpuPreprocessor:
             max_workers=8, mp_context=mp.get_context("spawn"),
             initializer=_pool_init)
         self._pool_warm = False
+        self._shm_pool: Optional[ShmDecodePool] = None
+        self._shm_t = None
+        self._shm_pinned = False
 
     def _decode_cpu(self, data: bytes) -> np.ndarray:
         from PIL import Image
@@ -82,6 +284,10 @@ class GpuPreprocessor:
         Upload + resize/normalize run on the side stream; the returned
         tensor is safe to consume on `compute_stream` (event-synced).
         """
+        if len(blobs) > 4:
+            out = self.process_shm(blobs, compute_stream)
+            if out is not None:
+                return out
         if len(blobs) <= 4:
             # small batches: pool round-trip costs more than decode
             arrs = [_decode_worker((b, self.c)) for b in blobs]
@@ -91,7 +297,49 @@ class GpuPreprocessor:
                                        chunksize=4))
         return self.process_arrays(arrs, compute_stream)
 
+    def _ensure_shm(self):
+        if self._shm_pool is None:
+            import os
+            nw = os.environ.get("TWD_DECODE_WORKERS")
+            self._shm_pool = ShmDecodePool(
+                self.c, workers=int(nw) if nw else None,
+                slot_bytes=1 << 19, slots=max(256, self.max_batch))
+            self._shm_t = torch.from_numpy(self._shm_pool.buf)
+            self._shm_pinned = bool(self.E.host_register(
+                self._shm_t.data_ptr(), self._shm_t.numel()))
+        return self._shm_pool
+
+    def process_shm(self, blobs,
+                    compute_stream: Optional[torch.cuda.Stream] = None
+                    ) -> Optional[torch.Tensor]:
+        """Shm-slot decode path: workers write pixels straight into a
+        hipHostRegister'ed mapping; one DMA of the slot prefix + one
+        resize kernel. Returns None when a blob doesn't fit (caller
+        uses the pipe path)."""
+        pool = self._ensure_shm()
+        meta = pool.decode_batch(blobs)
+        if meta is None:
+            return None
+        n = len(blobs)
+        out = self._out[:n]
+        lo = pool.hdr + pool.in_bytes
+        src = self._shm_t[lo:lo + n * pool.slot_bytes]
+        with torch.cuda.stream(self.stream):
+            arena_gpu = src.to(self.device, non_blocking=True)
+            meta_gpu = torch.from_numpy(meta).to(self.device,
+                                                 non_blocking=True)
+            self.E.resize_norm_batch(arena_gpu, meta_gpu, out, self.norm)
+        ev = torch.cuda.Event()
+        ev.record(self.stream)
+        target = compute_stream or torch.cuda.current_stream()
+        target.wait_event(ev)
+        return out
+
     def warm_pool(self):
+        try:
+            self._ensure_shm().warm()
+        except Exception:
+            pass  # /dev/shm unavailable: pipe path still works
         if not self._pool_warm:
             from PIL import Image
             buf = io.BytesIO()
