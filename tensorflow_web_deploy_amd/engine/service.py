@@ -128,6 +128,20 @@ class GpuServingEngine(LocalEngine):
         return out.float().cpu().numpy()[:n]
 
     @torch.inference_mode()
+    def infer_decoded(self, decoded, n: int) -> np.ndarray:
+        """GPU stage of the pipelined fast server: slot window ->
+        H2D + resize (side stream) -> model -> probs. The CPU decode
+        stage (pre.decode_shm) ran in another thread."""
+        batch = self.pre.upload_shm(decoded, n)
+        nb = _bucket(n)
+        if nb != n:
+            full = self.pre._out[:nb]
+            full[n:].zero_()
+            batch = full
+        out = self.ex.run({self.cfg.input_name: batch})[0]
+        return out.float().cpu().numpy()[:n]
+
+    @torch.inference_mode()
     def infer_blobs(self, blobs) -> np.ndarray:
         import os
         import time as _t

@@ -95,7 +95,8 @@ def _ring_worker(path, widx, nworkers, channels, slot_bytes, slots,
             return
         if msg == b"q":
             return
-        n = int(tasks[0])  # slot 0 of the task table carries the count
+        n = int(tasks[0])   # entry 0: task count
+        sbase = int(tasks[1])  # entry 1: slot-index base (phase half)
         for i in range(widx, n, nworkers):
             in_off = int(tasks[2 * (i + 1)])
             in_len = int(tasks[2 * (i + 1) + 1])
@@ -109,7 +110,7 @@ def _ring_worker(path, widx, nworkers, channels, slot_bytes, slots,
                 if arr.nbytes > slot_bytes:
                     meta[4 * i:4 * i + 4] = (0, 0, 0, 2)  # too big
                     continue
-                o = base_out + i * slot_bytes
+                o = base_out + (sbase + i) * slot_bytes
                 buf[o:o + arr.nbytes] = arr.reshape(-1)
                 meta[4 * i:4 * i + 4] = (arr.shape[0], arr.shape[1],
                                          arr.shape[2], 0)
@@ -143,6 +144,12 @@ class ShmDecodePool:
         self.in_bytes = in_bytes
         self.hdr = slots * 32
         total = self.hdr + in_bytes + slots * slot_bytes
+        # slot area is double-buffered: a pipelined server decodes
+        # batch N+1 into one half while batch N's half is still being
+        # DMA'd / consumed by the GPU stage
+        self.half = (slots - 2) // 2
+        self._phase = 0
+        self.last_window_off = 0  # byte offset of the active half
         fd, self.path = tempfile.mkstemp(prefix="twd_dec_",
                                          dir="/dev/shm")
         os.ftruncate(fd, total)
@@ -185,14 +192,23 @@ class ShmDecodePool:
                           timeout=timeout)
         self._warm = True
 
-    def decode_batch(self, blobs, timeout: float = 30.0):
+    def decode_batch(self, blobs, timeout: float = 30.0,
+                     phase: Optional[int] = None):
+        """phase: which slot half to decode into (a pipelined caller
+        owns the handoff; None = internal alternation). NOT
+        thread-safe — one decoding thread at a time."""
         if self._dead:
             return None
         n = len(blobs)
-        if (n == 0 or n + 1 > self.slots or
+        if (n == 0 or n > self.half or
                 sum(len(b) for b in blobs) > self.in_bytes):
             return None
+        if phase is None:
+            phase = self._phase
+            self._phase ^= 1
+        self.last_window_off = phase * self.half * self.slot_bytes
         self._tasks[0] = n
+        self._tasks[1] = phase * self.half  # slot index base for workers
         off = 0
         base_in = self.hdr
         for i, b in enumerate(blobs):
@@ -309,20 +325,25 @@ class GpuPreprocessor:
                 self._shm_t.data_ptr(), self._shm_t.numel()))
         return self._shm_pool
 
-    def process_shm(self, blobs,
-                    compute_stream: Optional[torch.cuda.Stream] = None
-                    ) -> Optional[torch.Tensor]:
-        """Shm-slot decode path: workers write pixels straight into a
-        hipHostRegister'ed mapping; one DMA of the slot prefix + one
-        resize kernel. Returns None when a blob doesn't fit (caller
-        uses the pipe path)."""
+    def decode_shm(self, blobs, phase=None):
+        """CPU stage: ring-decode into the active slot half. Returns
+        (meta, window_byte_off) or None (caller uses the pipe path);
+        raises ValueError for an undecodable image."""
         pool = self._ensure_shm()
-        meta = pool.decode_batch(blobs)
+        meta = pool.decode_batch(blobs, phase=phase)
         if meta is None:
             return None
-        n = len(blobs)
+        return meta, pool.last_window_off
+
+    def upload_shm(self, decoded, n: int,
+                   compute_stream: Optional[torch.cuda.Stream] = None
+                   ) -> torch.Tensor:
+        """GPU stage: one pinned DMA of the slot window + one resize
+        kernel on the side stream."""
+        meta, win = decoded
+        pool = self._shm_pool
         out = self._out[:n]
-        lo = pool.hdr + pool.in_bytes
+        lo = pool.hdr + pool.in_bytes + win
         src = self._shm_t[lo:lo + n * pool.slot_bytes]
         with torch.cuda.stream(self.stream):
             arena_gpu = src.to(self.device, non_blocking=True)
@@ -334,6 +355,15 @@ class GpuPreprocessor:
         target = compute_stream or torch.cuda.current_stream()
         target.wait_event(ev)
         return out
+
+    def process_shm(self, blobs,
+                    compute_stream: Optional[torch.cuda.Stream] = None
+                    ) -> Optional[torch.Tensor]:
+        """Single-thread form: decode_shm + upload_shm."""
+        decoded = self.decode_shm(blobs)
+        if decoded is None:
+            return None
+        return self.upload_shm(decoded, len(blobs), compute_stream)
 
     def warm_pool(self):
         try:

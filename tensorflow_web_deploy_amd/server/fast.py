@@ -53,7 +53,12 @@ class FastServer:
             return [(200, self._payload(p)) for p in probs]
         except Exception:
             pass
-        # slow path: find the undecodable items, serve the rest
+        return self._infer_isolating(bodies)
+
+    def _infer_isolating(self, bodies):
+        """Per-item decode (NO shm ring — safe from any thread), bad
+        images 400, rest served via the legacy pinned-arena path."""
+        eng = self.engine
         out = [None] * len(bodies)
         good, idx = [], []
         for i, b in enumerate(bodies):
@@ -107,9 +112,71 @@ class FastServer:
         return len(batch)
 
     def serve_forever(self):
+        if hasattr(self.engine, "infer_decoded"):
+            return self._serve_pipelined()
         try:
             while not self._stop.is_set():
                 self.step()
+        finally:
+            self.fe.stop()
+
+    # -- two-stage pipeline: decode(N+1) overlaps GPU+respond(N) ------
+    def _serve_pipelined(self):
+        """Ingest thread: wait_batch + ring-decode into a slot half it
+        OWNS (token from free_q; the GPU thread returns the token only
+        after the H2D of that half has been issued and synced — without
+        the token handoff the decode of batch N+2 could overwrite half
+        A while batch N's DMA from it is still in flight). Main thread:
+        upload + model + respond. Fallback paths never touch the ring
+        (it is single-decoder by design)."""
+        import queue
+        q = queue.Queue()
+        free_q = queue.Queue()
+        free_q.put(0)
+        free_q.put(1)
+        eng = self.engine
+
+        def ingest():
+            while not self._stop.is_set():
+                batch = self.fe.wait_batch(self.cfg.max_batch,
+                                           self.cfg.max_wait_ms)
+                if not batch:
+                    continue
+                bodies = [b for (_, b) in batch]
+                ph = free_q.get()
+                try:
+                    decoded = eng.pre.decode_shm(bodies, phase=ph)
+                except Exception:
+                    decoded = None  # bad image: isolate downstream
+                if decoded is None:
+                    free_q.put(ph)  # half unused
+                    ph = None
+                q.put((batch, bodies, decoded, ph))
+            q.put(None)
+
+        t = threading.Thread(target=ingest, daemon=True)
+        t.start()
+        try:
+            while not self._stop.is_set():
+                item = q.get()
+                if item is None:
+                    break
+                batch, bodies, decoded, ph = item
+                if decoded is not None:
+                    try:
+                        probs = eng.infer_decoded(decoded, len(bodies))
+                        results = [(200, self._payload(p))
+                                   for p in probs]
+                    except Exception:
+                        results = self._infer_isolating(bodies)
+                    finally:
+                        free_q.put(ph)  # H2D synced inside infer_decoded
+                else:
+                    results = self._infer_isolating(bodies)
+                self.fe.respond_many([
+                    (rid, st, body.encode())
+                    for (rid, _), (st, body) in zip(batch, results)
+                ])
         finally:
             self.fe.stop()
 
