@@ -32,7 +32,8 @@ def test_decode_matches_reference(pool):
     blobs = [_jpeg(96, 128), _jpeg(224, 224), _jpeg(31, 57)]
     meta = pool.decode_batch(blobs)
     assert meta is not None
-    base = pool.hdr + pool.in_bytes
+    # meta offsets are relative to the active (double-buffered) half
+    base = pool.hdr + pool.in_bytes + pool.last_window_off
     for i, b in enumerate(blobs):
         ref = _decode_worker((b, 3))
         off, h, w, c = meta[i]
