@@ -317,9 +317,12 @@ class GpuPreprocessor:
         if self._shm_pool is None:
             import os
             nw = os.environ.get("TWD_DECODE_WORKERS")
+            # half-size (slots-2)//2 must cover max_batch: each
+            # pipeline phase decodes a full batch into one half
             self._shm_pool = ShmDecodePool(
                 self.c, workers=int(nw) if nw else None,
-                slot_bytes=1 << 19, slots=max(256, self.max_batch))
+                slot_bytes=1 << 19,
+                slots=max(258, 2 * self.max_batch + 2))
             self._shm_t = torch.from_numpy(self._shm_pool.buf)
             self._shm_pinned = bool(self.E.host_register(
                 self._shm_t.data_ptr(), self._shm_t.numel()))
