@@ -24,6 +24,7 @@ ext = cpp_extension.CUDAExtension(
         os.path.join(CSRC, "layernorm.hip"),
         os.path.join(CSRC, "preprocess.hip"),
         os.path.join(CSRC, "attn_fused.hip"),
+        os.path.join(CSRC, "depthwise.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

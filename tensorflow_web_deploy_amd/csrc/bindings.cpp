@@ -313,6 +313,34 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_batched_bf16", &gemm_batched_bf16);
   m.def("gemm_strided_bf16", &gemm_strided_bf16);
   m.def("attn_fused_bf16", &attn_fused_bf16);
+  m.def("dwconv_bf16", [](torch::Tensor X, torch::Tensor W2,
+                          c10::optional<torch::Tensor> bias,
+                          c10::optional<torch::Tensor> residual,
+                          torch::Tensor Y, int64_t R, int64_t S,
+                          int64_t sh, int64_t sw, int64_t pt, int64_t pl,
+                          int64_t act) {
+    TORCH_CHECK(X.scalar_type() == torch::kBFloat16 && X.is_contiguous());
+    TORCH_CHECK(Y.scalar_type() == torch::kBFloat16 && Y.is_contiguous());
+    TORCH_CHECK(W2.scalar_type() == torch::kBFloat16 && W2.is_contiguous());
+    const float* bptr = nullptr;
+    if (bias) {
+      TORCH_CHECK(bias->scalar_type() == torch::kFloat32);
+      bptr = bias->data_ptr<float>();
+    }
+    const void* rptr = nullptr;
+    if (residual) {
+      TORCH_CHECK(residual->scalar_type() == torch::kBFloat16 &&
+                  residual->is_contiguous());
+      rptr = residual->data_ptr();
+    }
+    int N = X.size(0), H = X.size(1), Wd = X.size(2), C = X.size(3);
+    int HO = Y.size(1), WO = Y.size(2);
+    TORCH_CHECK(Y.size(3) == C && Y.size(0) == N);
+    launch_dwconv_bf16(X.data_ptr(), W2.data_ptr(), bptr, rptr,
+                       Y.data_ptr(), N, H, Wd, C, HO, WO, (int)R, (int)S,
+                       (int)sh, (int)sw, (int)pt, (int)pl, (int)act,
+                       c10::hip::getCurrentHIPStream());
+  });
   m.def("resize_norm", &resize_norm);
   m.def("resize_norm_batch", &resize_norm_batch);
   // Pin an externally-allocated host buffer (e.g. a /dev/shm mapping the

@@ -382,6 +382,12 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
 #pragma unroll
         for (int e = 0; e < 8; ++e)
           v[e] = bf2f(v[e]) > 0.f ? v[e] : (__bf16)0.f;
+      } else if (relu == 2) {  // relu6 (MobileNet blocks)
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          float f = bf2f(v[e]);
+          v[e] = (__bf16)(f < 0.f ? 0.f : (f > 6.f ? 6.f : f));
+        }
       } else if (relu == 5) {  // erf-GELU fused epilogue (ViT MLP)
 #pragma unroll
         for (int e = 0; e < 8; ++e) {
@@ -668,6 +674,12 @@ void gemm_conv_kernel_256(const __bf16* __restrict__ A,
 #pragma unroll
         for (int e = 0; e < 8; ++e)
           v[e] = bf2f(v[e]) > 0.f ? v[e] : (__bf16)0.f;
+      } else if (relu == 2) {  // relu6 (MobileNet blocks)
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          float f = bf2f(v[e]);
+          v[e] = (__bf16)(f < 0.f ? 0.f : (f > 6.f ? 6.f : f));
+        }
       } else if (relu == 5) {  // erf-GELU fused epilogue (ViT MLP)
 #pragma unroll
         for (int e = 0; e < 8; ++e) {
@@ -915,6 +927,12 @@ void gemm_conv_kernel_256p(const __bf16* __restrict__ A,
 #pragma unroll
         for (int e = 0; e < 8; ++e)
           v[e] = bf2f(v[e]) > 0.f ? v[e] : (__bf16)0.f;
+      } else if (relu == 2) {  // relu6 (MobileNet blocks)
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          float f = bf2f(v[e]);
+          v[e] = (__bf16)(f < 0.f ? 0.f : (f > 6.f ? 6.f : f));
+        }
       } else if (relu == 5) {
 #pragma unroll
         for (int e = 0; e < 8; ++e) {
@@ -980,6 +998,7 @@ __global__ void splitk_combine_kernel(const float* __restrict__ Cpart,
       if (e < rem) {
         float f = v[e];
         if (relu == 1 && f < 0.f) f = 0.f;
+        else if (relu == 2) f = f < 0.f ? 0.f : (f > 6.f ? 6.f : f);
         else if (relu == 5) f = 0.5f * f * (1.f + erff(f * 0.70710678f));
         Cout[i + e] = (__bf16)f;
       }

@@ -18,6 +18,12 @@ void launch_conv2d_bf16(const void* X, const void* B, void* C,
                         int act, hipStream_t stream, float* ws = nullptr,
                         int splitk = 1, int ldc = 0);
 
+void launch_dwconv_bf16(const void* x, const void* w, const float* bias,
+                        const void* residual, void* y, int N, int H,
+                        int W, int C, int HO, int WO, int R, int S,
+                        int sh, int sw, int pt, int pl, int act,
+                        hipStream_t stream);
+
 // pool_reduce.hip
 void launch_maxpool_bf16(const void* X, void* Y, int Nimg, int H, int W,
                          int C, int kh, int kw, int sh, int sw, int pt,

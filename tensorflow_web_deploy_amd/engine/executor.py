@@ -79,7 +79,8 @@ def _convblock(node, ins, consts):
     if residual is not None:
         y = y + residual
     if node.attrs.get("relu"):
-        y = F.relu(y)
+        y = (torch.clamp(y, 0.0, 6.0) if node.attrs.get("relu6")
+             else F.relu(y))
     return y
 
 
@@ -104,7 +105,8 @@ def _dwconvblock(node, ins, consts):
     if residual is not None:
         y = y + residual
     if node.attrs.get("relu"):
-        y = F.relu(y)
+        y = (torch.clamp(y, 0.0, 6.0) if node.attrs.get("relu6")
+             else F.relu(y))
     return y
 
 

@@ -81,6 +81,20 @@ class GpuExecutor:
         self._weights[wname] = (t, R, S, C, Cp, O, Kp)
         return self._weights[wname]
 
+    def _prep_dw_weight(self, wname: str):
+        """TF depthwise filter [R,S,C,mult] -> flat bf16 [R*S*C]
+        (mult=1; the kernel indexes w[(r*S+s)*C + c])."""
+        key = wname + "/dw"
+        if key in self._weights:
+            return self._weights[key]
+        w = self.consts[wname].astype(np.float32)
+        R, S, C, mult = w.shape
+        assert mult == 1, f"depth_multiplier {mult} unsupported on GPU"
+        t = torch.from_numpy(np.ascontiguousarray(w.reshape(-1))).to(
+            self.device, torch.bfloat16).contiguous()
+        self._weights[key] = (t, R, S)
+        return self._weights[key]
+
     def _prep_dense_weight(self, wname: str, transpose_b: bool):
         key = wname + ("/T" if transpose_b else "")
         if key in self._weights:
@@ -218,7 +232,8 @@ class GpuExecutor:
                     Y = alloc(node.name, (N_, HO, WO, O))
                 bias = self._bias(node.attrs.get("bias"), O)
                 res = bufs[node.attrs["residual"]] if node.attrs.get("residual") else None
-                relu = bool(node.attrs.get("relu"))
+                relu = (2 if node.attrs.get("relu6") else
+                        1 if node.attrs.get("relu") else 0)
                 if Cp != Cin and X.shape[-1] != Cp:
                     Xp = torch.zeros((N_, H, W, Cp), dtype=torch.bfloat16,
                                      device=dev)
@@ -239,6 +254,29 @@ class GpuExecutor:
                                  relu=relu:
                                  E.conv2d_bf16(X, wt, Y, bias, res, R, S,
                                                sh, sw, pt, pl, relu))
+            elif op == "DepthwiseConvBlock":
+                xn = node.inputs[0]
+                X = bufs[xn]
+                N_, H, W, Cin = shapes[xn]
+                wt, R, S = self._prep_dw_weight(node.inputs[1])
+                strides = node.attrs["strides"]
+                sh, sw = int(strides[1]), int(strides[2])
+                if node.attrs.get("padding", "VALID") == "SAME":
+                    HO, WO, pt, pl = _same_pads(H, W, R, S, sh, sw)
+                else:
+                    HO, WO = _valid_out(H, W, R, S, sh, sw)
+                    pt = pl = 0
+                Y = alloc(node.name, (N_, HO, WO, Cin))
+                bias = self._bias(node.attrs.get("bias"), Cin)
+                res = (bufs[node.attrs["residual"]]
+                       if node.attrs.get("residual") else None)
+                act = (2 if node.attrs.get("relu6") else
+                       1 if node.attrs.get("relu") else 0)
+                steps.append(lambda X=X, wt=wt, Y=Y, bias=bias, res=res,
+                             R=R, S=S, sh=sh, sw=sw, pt=pt, pl=pl,
+                             act=act:
+                             E.dwconv_bf16(X, wt, bias, res, Y, R, S,
+                                           sh, sw, pt, pl, act))
             elif op == "Dense":
                 xn = node.inputs[0]
                 X = bufs[xn]
@@ -250,7 +288,8 @@ class GpuExecutor:
                 if bias is not None and Np != N:
                     bias = torch.cat([bias, torch.full((Np - N,), -1e30,
                                                        device=dev)])
-                relu = bool(node.attrs.get("relu"))
+                relu = (2 if node.attrs.get("relu6") else
+                        1 if node.attrs.get("relu") else 0)
                 Yp = alloc(node.name + "/padded", (M, Np))
                 A2 = X.view(M, K)
                 steps.append(lambda A2=A2, wt=wt, Yp=Yp, bias=bias, K=K,
@@ -588,6 +627,11 @@ class GpuExecutor:
                 Y = alloc(node.name, shapes[xn])
                 steps.append(lambda X=bufs[xn], Y=Y:
                              E.bias_act_bf16(X, None, Y, ACT_RELU))
+            elif op == "Relu6":
+                xn = node.inputs[0]
+                Y = alloc(node.name, shapes[xn])
+                steps.append(lambda X=bufs[xn], Y=Y:
+                             E.bias_act_bf16(X, None, Y, ACT_RELU6))
             elif op == "BiasAdd":
                 xn = node.inputs[0]
                 C = shapes[xn][-1]

@@ -145,13 +145,17 @@ def fold_batchnorm(graph: Graph) -> Graph:
             scale = gamma / np.sqrt(var + eps)
             shift = beta - mean * scale
             prod = nmap.get(x)
-            if (prod is not None and prod.op == "Conv2D"
+            if (prod is not None
+                    and prod.op in ("Conv2D", "DepthwiseConv2dNative")
                     and consumers.get(x, 0) == 1
                     and prod.inputs[1] in graph.consts):
-                w = graph.consts[prod.inputs[1]]  # HWIO
-                graph.consts[prod.inputs[1]] = (
-                    w.astype(np.float64) * scale[None, None, None, :]
-                ).astype(w.dtype)
+                w = graph.consts[prod.inputs[1]]
+                if prod.op == "Conv2D":  # HWIO: scale over O
+                    wf = w.astype(np.float64) * scale[None, None, None, :]
+                else:  # depthwise [kh,kw,C,mult]: out channel = (c,mult)
+                    wf = (w.astype(np.float64) *
+                          scale.reshape(w.shape[2], w.shape[3]))
+                graph.consts[prod.inputs[1]] = wf.astype(w.dtype)
                 bias_name = n.name + "/folded_bias"
                 graph.consts[bias_name] = shift.astype(np.float32)
                 bn_bias = Node(n.name, "BiasAdd", [x, bias_name],
@@ -174,7 +178,8 @@ def fuse_conv_blocks(graph: Graph) -> Graph:
     and MatMul [+BiasAdd] [+Relu|Softmax kept separate] -> Dense.
 
     ConvBlock attrs: strides, padding, plus fusion fields:
-      bias: const name or None; residual: tensor name or None; relu: bool.
+      bias: const name or None; residual: tensor name or None; relu: bool
+      (relu6=True refines relu to the 6-clamped form, MobileNet).
     """
     consumers: Dict[str, List[Node]] = {}
     for n in graph.nodes:
@@ -217,13 +222,16 @@ def fuse_conv_blocks(graph: Graph) -> Graph:
                     drop.add(c.name)
                     tail = c
                     c = sole_consumer(tail.name)
-            if c is not None and c.op == "Relu":
+            relu6 = False
+            if c is not None and c.op in ("Relu", "Relu6"):
                 relu = True
+                relu6 = c.op == "Relu6"
                 drop.add(c.name)
                 tail = c
             op = "ConvBlock" if n.op == "Conv2D" else "DepthwiseConvBlock"
             attrs = dict(n.attrs)
-            attrs.update({"bias": bias, "residual": residual, "relu": relu})
+            attrs.update({"bias": bias, "residual": residual, "relu": relu,
+                          "relu6": relu6})
             ins = [n.inputs[0], n.inputs[1]]
             if bias:
                 ins.append(bias)

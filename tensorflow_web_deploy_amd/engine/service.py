@@ -38,6 +38,12 @@ def build_graph(cfg: Config) -> Graph:
         builders["vit_b16"] = lambda: vit.build_vit_b16()
     except ImportError:
         pass
+    try:
+        from ..models import mobilenet_v2
+        builders["mobilenet_v2"] = \
+            lambda: mobilenet_v2.build_mobilenet_v2()
+    except ImportError:
+        pass
     if cfg.model_name not in builders:
         raise ValueError(f"unknown built-in model {cfg.model_name}")
     g = builders[cfg.model_name]()

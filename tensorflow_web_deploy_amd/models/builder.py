@@ -124,6 +124,18 @@ class GraphBuilder:
     def relu(self, x: str, name: str) -> str:
         return self.node("Relu", name, [x])
 
+    def relu6(self, x: str, name: str) -> str:
+        return self.node("Relu6", name, [x])
+
+    def depthwise_conv2d(self, x: str, name: str, k: int, cin: int,
+                         stride: int = 1, padding: str = "SAME",
+                         mult: int = 1) -> str:
+        """TF DepthwiseConv2dNative: filter [kh, kw, C, mult]."""
+        w = self.weight(f"{name}/depthwise_weights", (k, k, cin, mult))
+        return self.node("DepthwiseConv2dNative", name, [x, w],
+                         strides=[1, stride, stride, 1], padding=padding,
+                         data_format="NHWC", T=gd.AttrValue(type=gd.DT_FLOAT))
+
     def maxpool(self, x: str, name: str, k: int, stride: int,
                 padding: str = "SAME") -> str:
         return self.node("MaxPool", name, [x], ksize=[1, k, k, 1],

@@ -273,3 +273,42 @@ def test_attn_fused(E):
                          * scale, -1) @ vf
     torch.testing.assert_close(out2.float().permute(0, 2, 1, 3), ref2,
                                rtol=3e-2, atol=3e-2)
+
+
+@pytest.mark.parametrize("C,stride,act,res", [
+    (32, 1, 2, False),    # MobileNet stem-depthwise shape class
+    (144, 2, 2, False),   # stride-2 downsample
+    (960, 1, 1, False),   # widest depthwise
+    (12, 1, 0, True),     # channel tail (C%8!=0) + residual
+])
+def test_dwconv_bf16(E, C, stride, act, res):
+    """Depthwise conv vs torch fp32 (TF SAME semantics, groups=C)."""
+    torch.manual_seed(C + stride)
+    N, H, W, R = 2, 17, 23, 3
+    x = torch.randn(N, H, W, C)
+    w = torch.randn(R, R, C, 1) * 0.5
+    bias = torch.randn(C)
+    # TF SAME pads
+    HO = (H + stride - 1) // stride
+    WO = (W + stride - 1) // stride
+    ph = max((HO - 1) * stride + R - H, 0)
+    pw = max((WO - 1) * stride + R - W, 0)
+    pt, pl = ph // 2, pw // 2
+    xc = x.permute(0, 3, 1, 2)
+    xp = torch.nn.functional.pad(xc, (pl, pw - pl, pt, ph - pt))
+    wt = w.permute(2, 3, 0, 1).reshape(C, 1, R, R)
+    ref = torch.nn.functional.conv2d(xp, wt, bias=bias, stride=stride,
+                                     groups=C).permute(0, 2, 3, 1)
+    rr = torch.randn(N, HO, WO, C) if res else None
+    if res:
+        ref = ref + rr
+    if act == 1:
+        ref = torch.relu(ref)
+    elif act == 2:
+        ref = torch.clamp(ref, 0.0, 6.0)
+
+    y = torch.empty(N, HO, WO, C, dtype=torch.bfloat16, device="cuda")
+    E.dwconv_bf16(_bf(x), _bf(w.reshape(-1)), bias.float().cuda(),
+                  _bf(rr) if res else None, y, R, R, stride, stride,
+                  pt, pl, act)
+    torch.testing.assert_close(y.float().cpu(), ref, atol=0.1, rtol=0.05)

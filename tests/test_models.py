@@ -25,6 +25,21 @@ def test_inception_v3():
     assert ops["AvgPool"] == 9 and ops["MaxPool"] == 4
 
 
+def test_mobilenet_v2():
+    eng, ops = _run("mobilenet_v2", 224, 224, 3, 1000)
+    # 17 inverted residuals; all BN/Relu6 absorbed into the blocks
+    assert ops["DepthwiseConvBlock"] == 17
+    assert ops["ConvBlock"] == 35
+    assert "Relu6" not in ops and "FusedBatchNormV3" not in ops
+    # depthwise blocks carry the relu6 epilogue
+    dw = [n for n in eng.graph.nodes if n.op == "DepthwiseConvBlock"]
+    assert all(n.attrs.get("relu6") and n.attrs.get("relu") for n in dw)
+    # project convs are linear (no act), expand convs are relu6
+    cb = [n for n in eng.graph.nodes if n.op == "ConvBlock"]
+    assert any(not n.attrs.get("relu") for n in cb)
+    assert any(n.attrs.get("relu6") for n in cb)
+
+
 def test_vit_b16():
     eng, ops = _run("vit_b16", 224, 224, 3, 1000)
     # the primitive-op graph must fully fuse

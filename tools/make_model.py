@@ -14,7 +14,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 def main():
     p = argparse.ArgumentParser()
-    p.add_argument("model", choices=["lenet", "resnet50", "inception_v3",
+    p.add_argument("model", choices=["lenet", "resnet50", "inception_v3", "mobilenet_v2",
                                      "vit_b16"])
     p.add_argument("out")
     p.add_argument("--seed", type=int, default=0)
@@ -33,6 +33,9 @@ def main():
         "vit_b16": __import__(
             "tensorflow_web_deploy_amd.models.vit",
             fromlist=["x"]).build_vit_b16,
+        "mobilenet_v2": __import__(
+            "tensorflow_web_deploy_amd.models.mobilenet_v2",
+            fromlist=["x"]).build_mobilenet_v2,
     }
     g = builders[args.model](seed=args.seed)
     if args.savedmodel:
