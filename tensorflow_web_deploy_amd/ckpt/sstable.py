@@ -9,7 +9,8 @@ our own writer).
 
 Per-block trailer: 1 compression byte (0 = none, 1 = snappy) + 4-byte
 masked crc32c. A small pure-Python snappy decoder covers compressed
-blocks; crc verification is skipped (documented).
+blocks; block crc32c (Castagnoli, LevelDB-masked) is computed on write
+and verified on read (a stored 0 — older fixtures — is tolerated).
 """
 from __future__ import annotations
 
@@ -105,9 +106,50 @@ def _parse_block(block: bytes) -> List[Tuple[bytes, bytes]]:
     return entries
 
 
-def _read_block(data: bytes, offset: int, size: int) -> List[Tuple[bytes, bytes]]:
+# crc32c (Castagnoli 0x1EDC6F41, reflected) — table-driven; blocks in a
+# TF bundle index are KB-sized so a Python loop is fine
+_CRC_TABLE = []
+
+
+def _crc_table():
+    global _CRC_TABLE
+    if not _CRC_TABLE:
+        poly = 0x82F63B78
+        t = []
+        for i in range(256):
+            c = i
+            for _ in range(8):
+                c = (c >> 1) ^ poly if c & 1 else c >> 1
+            t.append(c)
+        _CRC_TABLE = t
+    return _CRC_TABLE
+
+
+def crc32c(buf: bytes, crc: int = 0) -> int:
+    t = _crc_table()
+    c = crc ^ 0xFFFFFFFF
+    for b in buf:
+        c = t[(c ^ b) & 0xFF] ^ (c >> 8)
+    return c ^ 0xFFFFFFFF
+
+
+def _mask_crc(c: int) -> int:
+    return (((c >> 15) | (c << 17)) + 0xA282EAD8) & 0xFFFFFFFF
+
+
+def _read_block(data: bytes, offset: int, size: int,
+                verify: bool = True) -> List[Tuple[bytes, bytes]]:
     body = data[offset:offset + size]
     ctype = data[offset + size]
+    if verify:
+        stored = struct.unpack(
+            "<I", data[offset + size + 1:offset + size + 5])[0]
+        if stored != 0:  # 0 = unchecksummed legacy fixture
+            want = _mask_crc(crc32c(body + bytes([ctype])))
+            if stored != want:
+                raise ValueError(
+                    f"SSTable block crc mismatch at {offset}: "
+                    f"stored {stored:#x} != computed {want:#x}")
     if ctype == 1:
         body = snappy_decompress(body)
     elif ctype != 0:
@@ -138,10 +180,9 @@ def read_table(data: bytes) -> Dict[bytes, bytes]:
 
 # -------------------------------------------------------------- writer
 
-def _crc32c_masked(_body: bytes) -> int:
-    # LevelDB readers that verify checksums would reject 0; TF's reader
-    # only verifies when asked. Fixtures carry 0 (documented).
-    return 0
+def _crc32c_masked(body: bytes, ctype: int = 0) -> int:
+    """LevelDB record crc: crc32c over body + the type byte, masked."""
+    return _mask_crc(crc32c(body + bytes([ctype])))
 
 
 class TableWriter:
@@ -174,13 +215,13 @@ class TableWriter:
         # metaindex (empty)
         metab = self._block([])
         mi_off, mi_size = len(out), len(metab)
-        out += metab + bytes([0]) + struct.pack("<I", 0)
+        out += metab + bytes([0]) + struct.pack("<I", _crc32c_masked(metab))
         # index block: one entry key>=last key -> data handle
         last_key = max((k for k, _ in self.items), default=b"")
         handle = _wv(data_off) + _wv(data_size)
         idxb = self._block([(last_key + b"\x00", handle)])
         idx_off, idx_size = len(out), len(idxb)
-        out += idxb + bytes([0]) + struct.pack("<I", 0)
+        out += idxb + bytes([0]) + struct.pack("<I", _crc32c_masked(idxb))
         footer = _wv(mi_off) + _wv(mi_size) + _wv(idx_off) + _wv(idx_size)
         footer += b"\x00" * (40 - len(footer))
         footer += struct.pack("<Q", MAGIC)
