@@ -40,8 +40,32 @@ def _pick_port() -> int:
     return p
 
 
+def _norm_code(normalize: str) -> int:
+    return {"raw": 0, "unit": 1, "inception": 2}[normalize]
+
+
+def _cpu_resize_shard(local_u8, metp, start, mine, slot, h, w, c, norm):
+    """CPU (gloo) replica resize: half-pixel bilinear to the model input
+    size + normalize — the CPU analogue of csrc/preprocess.hip."""
+    import torch.nn.functional as F
+    out = torch.empty((mine, h, w, c), dtype=torch.float32)
+    for i in range(mine):
+        ih, iw, ic = (int(metp[start + i][0]), int(metp[start + i][1]),
+                      int(metp[start + i][2]))
+        a = local_u8[i * slot:i * slot + ih * iw * ic]
+        img = a.reshape(ih, iw, ic).to(torch.float32).permute(2, 0, 1)
+        r = F.interpolate(img[None], size=(h, w), mode="bilinear",
+                          align_corners=False)[0].permute(1, 2, 0)
+        out[i] = r
+    if norm == 1:
+        out /= 255.0
+    elif norm == 2:
+        out = out / 127.5 - 1.0
+    return out
+
+
 def worker_main(rank: int, world: int, cfg: Config, backend: str, port: int,
-                in_q, out_q, err_q):
+                in_q, out_q, err_q, shm_desc=None):
     """Replica worker entry (separate process)."""
     try:
         import torch.distributed as dist
@@ -60,35 +84,90 @@ def worker_main(rank: int, world: int, cfg: Config, backend: str, port: int,
 
         h, w, c = cfg.input_height, cfg.input_width, cfg.input_channels
         nc = cfg.num_classes
+        norm = _norm_code(cfg.normalize)
+        cdev = device if backend == "nccl" else torch.device("cpu")
+        arena_np = None  # rank0: lazily-attached server decode arena
+        E = None
+        if backend == "nccl" and shm_desc is not None:
+            from ..ops.native import ext
+            E = ext()
         if rank == 0:
             out_q.put(("ready", world))
 
         while True:
+            # header broadcast: [total, mode]; total<0 = shutdown.
+            # mode 0: fp32 model-size batch scattered from rank0.
+            # mode 1: u8 decoded-pixel SLOTS scattered (server's shm
+            #         decode arena mapped zero-copy by rank0); each
+            #         replica resizes its own shard (HIP kernel on
+            #         nccl, torch bilinear on gloo).
+            hdr = torch.zeros(2, dtype=torch.int64, device=cdev)
+            item = None
             if rank == 0:
                 item = in_q.get()
                 if item is None:
-                    n0 = torch.tensor([-1], dtype=torch.int64, device=device
-                                      if backend == "nccl" else "cpu")
-                    dist.broadcast(n0, src=0)
-                    break
-                batch = torch.from_numpy(item)
+                    hdr[0] = -1
+                elif isinstance(item, np.ndarray):
+                    hdr[0], hdr[1] = item.shape[0], 0
+                else:
+                    hdr[0], hdr[1] = item[1], 1
+            dist.broadcast(hdr, src=0)
+            total, mode = int(hdr[0].item()), int(hdr[1].item())
+            if total < 0:
+                break
+            if mode == 0:
+                if rank == 0:
+                    shard = comm._scatter_known(total, (h, w, c),
+                                                torch.from_numpy(item))
+                else:
+                    shard = comm._scatter_known(total, (h, w, c))
             else:
-                batch = None
-            # scatter_batch broadcasts the row count first; -1 = shutdown
-            if rank != 0:
-                n0 = torch.tensor([0], dtype=torch.int64, device=device
-                                  if backend == "nccl" else "cpu")
-                dist.broadcast(n0, src=0)
-                total = int(n0.item())
-                if total < 0:
-                    break
-                shard = comm._scatter_known(total, (h, w, c))
-            else:
-                total = batch.shape[0]
-                n0 = torch.tensor([total], dtype=torch.int64, device=device
-                                  if backend == "nccl" else "cpu")
-                dist.broadcast(n0, src=0)
-                shard = comm._scatter_known(total, (h, w, c), batch)
+                slot = shm_desc["slot_bytes"]
+                per = (total + world - 1) // world
+                metp = torch.zeros((per * world, 3), dtype=torch.int32,
+                                   device=cdev)
+                local = torch.empty(per * slot, dtype=torch.uint8,
+                                    device=cdev)
+                if rank == 0:
+                    _tag, n, win, meta = item
+                    if arena_np is None:
+                        import mmap
+                        f = open(shm_desc["path"], "rb")
+                        mm = mmap.mmap(f.fileno(), 0,
+                                       prot=mmap.PROT_READ)
+                        arena_np = np.frombuffer(mm, dtype=np.uint8)
+                    base = (shm_desc["hdr"] + shm_desc["in_bytes"] + win)
+                    src = torch.from_numpy(
+                        np.ascontiguousarray(
+                            arena_np[base:base + n * slot]))
+                    padded = torch.zeros(per * world * slot,
+                                         dtype=torch.uint8, device=cdev)
+                    padded[:n * slot] = src.to(cdev)
+                    metp[:n] = torch.from_numpy(
+                        meta[:, 1:4].astype(np.int32)).to(cdev)
+                    dist.scatter(local, list(padded.chunk(world)), src=0)
+                else:
+                    dist.scatter(local, None, src=0)
+                dist.broadcast(metp, src=0)
+                start = rank * per
+                mine = max(0, min(per, total - start))
+                if mine > 0 and backend == "nccl":
+                    meta4 = torch.zeros((mine, 4), dtype=torch.int32)
+                    meta4[:, 0] = (torch.arange(mine, dtype=torch.int32)
+                                   * slot)
+                    meta4[:, 1:] = metp[start:start + mine].cpu()
+                    out_t = torch.empty((mine, h, w, c),
+                                        dtype=torch.float32,
+                                        device=device)
+                    E.resize_norm_batch(local[:mine * slot],
+                                        meta4.to(device), out_t, norm)
+                    shard = out_t
+                elif mine > 0:
+                    shard = _cpu_resize_shard(local, metp, start, mine,
+                                              slot, h, w, c, norm)
+                else:
+                    shard = torch.zeros((0, h, w, c),
+                                        dtype=torch.float32, device=cdev)
             if shard.shape[0] > 0:
                 probs = eng.ex.run({cfg.input_name: shard})[0].float()
             else:
@@ -124,6 +203,34 @@ def _scatter_known(self, total, shape_tail, batch=None):
 TorchDistComm._scatter_known = _scatter_known
 
 
+class BlobPre:
+    """Server-side decode stage for DP serving: ShmDecodePool with the
+    GpuPreprocessor.decode_shm interface (no GPU in this process)."""
+
+    def __init__(self, cfg: Config):
+        import os
+
+        from ..ingest.gpu_pipeline import ShmDecodePool
+        nw = os.environ.get("TWD_DECODE_WORKERS")
+        self.pool = ShmDecodePool(
+            cfg.input_channels, workers=int(nw) if nw else None,
+            slot_bytes=1 << 19, slots=max(258, 2 * cfg.max_batch + 2))
+
+    def decode_shm(self, blobs, phase=None):
+        meta = self.pool.decode_batch(blobs, phase=phase)
+        if meta is None:
+            return None
+        return meta, self.pool.last_window_off
+
+    def warm(self):
+        self.pool.warm()
+
+    def desc(self):
+        p = self.pool
+        return {"path": p.path, "slot_bytes": p.slot_bytes,
+                "in_bytes": p.in_bytes, "hdr": p.hdr}
+
+
 class ReplicatedEngine:
     """`infer(batch) -> probs` over N DP replica processes.
 
@@ -134,7 +241,7 @@ class ReplicatedEngine:
 
     def __init__(self, cfg: Config, world: Optional[int] = None,
                  backend: Optional[str] = None, start_timeout: float = 600.0,
-                 shrink_on_failure: bool = True):
+                 shrink_on_failure: bool = True, serve_blobs: bool = False):
         self.cfg = cfg
         self.world = world or cfg.gpus
         if backend is None:
@@ -142,6 +249,14 @@ class ReplicatedEngine:
         self.backend = backend
         self.shrink_on_failure = shrink_on_failure
         self._start_timeout = start_timeout
+        self.pre = None
+        self._shm_desc = None
+        if serve_blobs:
+            # serving fast path: the server decodes into a shm slot
+            # arena; rank0 maps the SAME file (zero-copy handoff) and
+            # u8 slots are scattered so each replica resizes its shard
+            self.pre = BlobPre(cfg)
+            self._shm_desc = self.pre.desc()
         self._spawn(start_timeout)
 
     def _spawn(self, timeout: float):
@@ -154,7 +269,8 @@ class ReplicatedEngine:
         for r in range(self.world):
             p = ctx.Process(target=worker_main,
                             args=(r, self.world, self.cfg, self.backend,
-                                  port, self.in_q, self.out_q, self.err_q),
+                                  port, self.in_q, self.out_q, self.err_q,
+                                  self._shm_desc),
                             daemon=True)
             p.start()
             self.procs.append(p)
@@ -208,6 +324,19 @@ class ReplicatedEngine:
         while True:
             try:
                 tag, payload = self.out_q.get(timeout=5.0)
+                assert tag == "probs"
+                return payload
+            except queue.Empty:
+                self._check_health()
+
+    def infer_decoded(self, decoded, n: int) -> np.ndarray:
+        """Fast-path entry: slots already decoded into self.pre's arena
+        (FastServer ingest thread). Only metadata crosses the queue."""
+        meta, win = decoded
+        self.in_q.put(("slots", n, win, meta))
+        while True:
+            try:
+                tag, payload = self.out_q.get(timeout=10.0)
                 assert tag == "probs"
                 return payload
             except queue.Empty:

@@ -255,7 +255,7 @@ def _default_engine(cfg: Config):
     import torch
     if cfg.gpus > 1:
         from ..parallel.replicas import ReplicatedEngine
-        return ReplicatedEngine(cfg)
+        return ReplicatedEngine(cfg, serve_blobs=True)
     if torch.cuda.is_available():
         from ..engine.service import GpuServingEngine
         return GpuServingEngine(cfg)

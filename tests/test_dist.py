@@ -81,3 +81,44 @@ def test_replica_shrink_serves_on_survivors():
         np.testing.assert_allclose(got, want, rtol=1e-5, atol=1e-6)
     finally:
         eng.shutdown()
+
+
+def test_blob_mode_world2_matches_world1():
+    """DP serving fast path (mode-1 u8 slot scatter + per-replica
+    resize): world=2 must agree exactly with world=1 on the same
+    decoded slots (plumbing test — same resize math both sides)."""
+    import io
+
+    from PIL import Image
+
+    from tensorflow_web_deploy_amd.parallel.replicas import ReplicatedEngine
+
+    cfg = Config(model_name="lenet", device="cpu", input_height=28,
+                 input_width=28, input_channels=1, num_classes=10,
+                 max_batch=16)
+
+    def png(h, w):
+        arr = (np.random.rand(h, w) * 255).astype("uint8")
+        b = io.BytesIO()
+        Image.fromarray(arr, "L").save(b, "PNG")
+        return b.getvalue()
+
+    blobs = [png(40, 40), png(28, 28), png(33, 17)]
+    outs = []
+    for world in (1, 2):
+        eng = ReplicatedEngine(cfg, world=world, backend="gloo",
+                               serve_blobs=True, start_timeout=120.0)
+        try:
+            d = eng.pre.decode_shm(blobs)
+            assert d is not None
+            p = eng.infer_decoded(d, len(blobs))
+            assert p.shape == (3, 10)
+            np.testing.assert_allclose(p.sum(1), 1.0, atol=1e-4)
+            outs.append(p)
+            # mode-0 path still works on the same group
+            x = np.random.rand(4, 28, 28, 1).astype(np.float32)
+            q = eng.infer(x)
+            assert q.shape == (4, 10)
+        finally:
+            eng.shutdown()
+    np.testing.assert_allclose(outs[0], outs[1], atol=1e-5)
