@@ -122,3 +122,41 @@ def test_blob_mode_world2_matches_world1():
         finally:
             eng.shutdown()
     np.testing.assert_allclose(outs[0], outs[1], atol=1e-5)
+
+
+@pytest.mark.gpu
+def test_blob_mode_nccl_world1_matches_local():
+    """mode-1 over nccl world=1: worker maps the server's decode arena,
+    scatters u8 slots, resizes with the HIP kernel. Must match the
+    in-process GpuServingEngine on the same JPEGs."""
+    import io
+
+    from PIL import Image
+
+    from tensorflow_web_deploy_amd.engine.service import GpuServingEngine
+    from tensorflow_web_deploy_amd.parallel.replicas import ReplicatedEngine
+
+    if not torch.cuda.is_available():
+        pytest.skip("needs GPU")
+    cfg = Config(model_name="lenet", device="cuda", input_height=28,
+                 input_width=28, input_channels=1, num_classes=10,
+                 max_batch=16)
+
+    def png(h, w):
+        arr = (np.random.rand(h, w) * 255).astype("uint8")
+        b = io.BytesIO()
+        Image.fromarray(arr, "L").save(b, "PNG")
+        return b.getvalue()
+
+    blobs = [png(40, 40), png(28, 28), png(33, 17)]
+    local = GpuServingEngine(cfg)
+    ref = local.infer_blobs(blobs)
+    eng = ReplicatedEngine(cfg, world=1, backend="nccl",
+                           serve_blobs=True, start_timeout=300.0)
+    try:
+        d = eng.pre.decode_shm(blobs)
+        assert d is not None
+        p = eng.infer_decoded(d, len(blobs))
+        np.testing.assert_allclose(p, ref, atol=2e-3)
+    finally:
+        eng.shutdown()
