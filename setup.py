@@ -39,10 +39,24 @@ httpd = cpp_extension.CppExtension(
     extra_compile_args={"cxx": ["-O2", "-std=c++17"]},
 )
 
+# Baseline JPEG entropy decoder. Plain Extension (NOT CppExtension):
+# it must import without libtorch so the shm-ring decode workers load
+# it cheaply.
+import pybind11  # noqa: E402
+from setuptools import Extension  # noqa: E402
+
+jpegc = Extension(
+    name="tensorflow_web_deploy_amd._jpegc",
+    sources=[os.path.join(CSRC, "jpegc.cpp")],
+    include_dirs=[pybind11.get_include()],
+    extra_compile_args=["-O3", "-std=c++17"],
+    language="c++",
+)
+
 setup(
     name="tensorflow_web_deploy_amd",
     version="0.1.0",
     packages=["tensorflow_web_deploy_amd"],
-    ext_modules=[ext, httpd],
+    ext_modules=[ext, httpd, jpegc],
     cmdclass={"build_ext": cpp_extension.BuildExtension},
 )
