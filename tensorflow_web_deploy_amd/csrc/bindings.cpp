@@ -341,6 +341,20 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
                        (int)sh, (int)sw, (int)pt, (int)pl, (int)act,
                        c10::hip::getCurrentHIPStream());
   });
+  m.def("jpeg_decode_slots", [](torch::Tensor arena, torch::Tensor planes,
+                                torch::Tensor flags, int64_t n,
+                                int64_t slot_bytes) {
+    TORCH_CHECK(arena.scalar_type() == torch::kUInt8 &&
+                arena.is_contiguous() && arena.is_cuda());
+    TORCH_CHECK(planes.scalar_type() == torch::kUInt8 &&
+                planes.is_cuda() &&
+                planes.numel() >= n * slot_bytes);
+    TORCH_CHECK(flags.scalar_type() == torch::kUInt8 && flags.is_cuda() &&
+                flags.numel() >= n);
+    launch_jpeg_decode(arena.data_ptr(), planes.data_ptr(),
+                       flags.data_ptr(), (int)n, slot_bytes,
+                       c10::hip::getCurrentHIPStream());
+  });
   m.def("resize_norm", &resize_norm);
   m.def("resize_norm_batch", &resize_norm_batch);
   // Pin an externally-allocated host buffer (e.g. a /dev/shm mapping the

@@ -24,6 +24,9 @@ void launch_dwconv_bf16(const void* x, const void* w, const float* bias,
                         int sh, int sw, int pt, int pl, int act,
                         hipStream_t stream);
 
+void launch_jpeg_decode(void* arena, void* planes, const void* flags,
+                        int n, int64_t slot_bytes, hipStream_t stream);
+
 // pool_reduce.hip
 void launch_maxpool_bf16(const void* X, void* Y, int Nimg, int H, int W,
                          int C, int kh, int kw, int sh, int sw, int pt,

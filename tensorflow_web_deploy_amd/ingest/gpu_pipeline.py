@@ -74,14 +74,61 @@ def _shm_attach(path):
     return _SHM["mm"]
 
 
+JPEG_MAGIC = 0x4A504731  # SlotHdr.magic (csrc/jpeg_gpu.hip)
+JPEG_HDR_BYTES = 512     # pixels land at slot+512 (header stays intact)
+
+
+def _write_jpeg_slot(buf, off, slot_bytes, info):
+    """Serialize a _jpegc.parse() result as a GPU coeff slot (SlotHdr
+    layout in csrc/jpeg_gpu.hip). Returns (H, W, 3) or None if it does
+    not fit (caller falls back to pixel decode)."""
+    W, H, nc = info["width"], info["height"], info["ncomp"]
+    if nc != 3:
+        return None
+    hdr = np.zeros(23, np.int32)
+    hdr[0] = JPEG_MAGIC
+    hdr[1], hdr[2], hdr[3] = W, H, nc
+    coef_off, plane_off = JPEG_HDR_BYTES, 0
+    for ci in range(nc):
+        bh, bw = info["blocks"][ci]
+        h_, v_ = info["sampling"][ci]
+        hdr[4 + ci], hdr[7 + ci] = h_, v_
+        hdr[10 + ci], hdr[13 + ci] = bh, bw
+        hdr[16 + ci], hdr[19 + ci] = coef_off, plane_off
+        coef_off += bh * bw * 128
+        plane_off += bh * 8 * bw * 8
+    if (coef_off > slot_bytes or plane_off > slot_bytes or
+            JPEG_HDR_BYTES + W * H * 3 > slot_bytes):
+        return None
+    buf[off:off + 92] = hdr.view(np.uint8)
+    qts = np.concatenate([np.asarray(info["qt"][ci], np.uint16)
+                          for ci in range(nc)])
+    buf[off + 92:off + 92 + 384] = qts.view(np.uint8)
+    for ci in range(nc):
+        cb = np.asarray(info["coef"][ci], np.int16).view(np.uint8)
+        o = off + int(hdr[16 + ci])
+        buf[o:o + cb.nbytes] = cb
+    return (H, W, 3)
+
+
 def _ring_worker(path, widx, nworkers, channels, slot_bytes, slots,
-                 in_bytes, wake_rx, done_tx):
+                 in_bytes, wake_rx, done_tx, jpeg=False):
     """Persistent decode worker: wake byte -> decode my stripe -> done
-    byte. Runs in its own process; PIL decode releases the GIL there."""
+    byte. Runs in its own process; PIL decode releases the GIL there.
+    With jpeg=True, baseline JPEGs are entropy-decoded only (_jpegc)
+    and written as COEFF slots (status 3) for the GPU IDCT kernels;
+    anything else falls back to full PIL pixel decode."""
     import gc
     gc.collect()
     gc.freeze()
     from PIL import Image
+    jparse = None
+    if jpeg and channels == 3:
+        try:
+            from tensorflow_web_deploy_amd import _jpegc
+            jparse = _jpegc.parse
+        except Exception:
+            jparse = None
     buf = _shm_attach(path)
     tasks = np.frombuffer(buf[:slots * 16], dtype=np.int64)
     meta = np.frombuffer(buf[slots * 16:slots * 16 + slots * 16],
@@ -100,9 +147,20 @@ def _ring_worker(path, widx, nworkers, channels, slot_bytes, slots,
         for i in range(widx, n, nworkers):
             in_off = int(tasks[2 * (i + 1)])
             in_len = int(tasks[2 * (i + 1) + 1])
+            data = bytes(buf[base_in + in_off:base_in + in_off + in_len])
+            if jparse is not None:
+                try:
+                    info = jparse(data)
+                    sh = _write_jpeg_slot(
+                        buf, base_out + (sbase + i) * slot_bytes,
+                        slot_bytes, info)
+                    if sh is not None:
+                        meta[4 * i:4 * i + 4] = (sh[0], sh[1], sh[2], 3)
+                        continue
+                except ValueError:
+                    pass  # progressive/CMYK/...: PIL pixel path below
             try:
-                img = Image.open(io.BytesIO(
-                    bytes(buf[base_in + in_off:base_in + in_off + in_len])))
+                img = Image.open(io.BytesIO(data))
                 img = img.convert("L" if channels == 1 else "RGB")
                 arr = np.asarray(img, dtype=np.uint8)
                 if arr.ndim == 2:
@@ -134,11 +192,12 @@ class ShmDecodePool:
 
     def __init__(self, channels: int, workers: Optional[int] = None,
                  slot_bytes: int = 1 << 19, slots: int = 256,
-                 in_bytes: int = 64 << 20):
+                 in_bytes: int = 64 << 20, jpeg: bool = False):
         import multiprocessing as mp
         import os
         import tempfile
         self.c = channels
+        self.jpeg = jpeg
         self.slot_bytes = slot_bytes
         self.slots = slots
         self.in_bytes = in_bytes
@@ -170,7 +229,7 @@ class ShmDecodePool:
             pr = ctx.Process(
                 target=_ring_worker,
                 args=(self.path, j, workers, channels, slot_bytes,
-                      slots, in_bytes, wr, dw),
+                      slots, in_bytes, wr, dw, jpeg),
                 daemon=True)
             pr.start()
             wr.close()
@@ -221,12 +280,17 @@ class ShmDecodePool:
         nw = min(self.workers, n)
         for j in range(nw):
             self._wake[j].send_bytes(b"g")
-        for j in range(nw):
-            if not self._done[j].poll(timeout):
-                self._dead = True  # worker gone: never trust the ring
-                return None
-            self._done[j].recv_bytes()
+        try:
+            for j in range(nw):
+                if not self._done[j].poll(timeout):
+                    self._dead = True  # worker gone: don't trust the ring
+                    return None
+                self._done[j].recv_bytes()
+        except (EOFError, OSError):
+            self._dead = True
+            return None
         meta = np.empty((n, 4), np.int32)
+        flags = np.zeros(n, np.uint8)
         m = self._meta
         for i in range(n):
             st = m[4 * i + 3]
@@ -235,9 +299,15 @@ class ShmDecodePool:
             if st == 2:
                 return None  # exceeds a slot: legacy path handles it
             # offsets RELATIVE to the slot area: the H2D copies the
-            # slot prefix only, so the kernel sees that base
-            meta[i] = (i * self.slot_bytes, m[4 * i], m[4 * i + 1],
-                       m[4 * i + 2])
+            # slot prefix only, so the kernel sees that base. Status 3
+            # = COEFF slot (GPU jpeg path): pixels will appear at +512
+            # after the jpeg kernels run.
+            off = i * self.slot_bytes
+            if st == 3:
+                off += JPEG_HDR_BYTES
+                flags[i] = 1
+            meta[i] = (off, m[4 * i], m[4 * i + 1], m[4 * i + 2])
+        self.last_flags = flags
         return meta
 
     def close(self):
@@ -282,6 +352,7 @@ class GpuPreprocessor:
         self._shm_pool: Optional[ShmDecodePool] = None
         self._shm_t = None
         self._shm_pinned = False
+        self._planes = None  # GPU scratch for the jpeg IDCT kernels
 
     def _decode_cpu(self, data: bytes) -> np.ndarray:
         from PIL import Image
@@ -319,10 +390,13 @@ class GpuPreprocessor:
             nw = os.environ.get("TWD_DECODE_WORKERS")
             # half-size (slots-2)//2 must cover max_batch: each
             # pipeline phase decodes a full batch into one half
+            import os as _os
+            jpeg = (_os.environ.get("TWD_GPU_JPEG", "1") != "0"
+                    and self.c == 3)
             self._shm_pool = ShmDecodePool(
                 self.c, workers=int(nw) if nw else None,
                 slot_bytes=1 << 19,
-                slots=max(258, 2 * self.max_batch + 2))
+                slots=max(258, 2 * self.max_batch + 2), jpeg=jpeg)
             self._shm_t = torch.from_numpy(self._shm_pool.buf)
             self._shm_pinned = bool(self.E.host_register(
                 self._shm_t.data_ptr(), self._shm_t.numel()))
@@ -336,20 +410,30 @@ class GpuPreprocessor:
         meta = pool.decode_batch(blobs, phase=phase)
         if meta is None:
             return None
-        return meta, pool.last_window_off
+        return meta, pool.last_window_off, pool.last_flags
 
     def upload_shm(self, decoded, n: int,
                    compute_stream: Optional[torch.cuda.Stream] = None
                    ) -> torch.Tensor:
-        """GPU stage: one pinned DMA of the slot window + one resize
-        kernel on the side stream."""
-        meta, win = decoded
+        """GPU stage: one pinned DMA of the slot window (+ the jpeg
+        dequant/IDCT/color kernels for coeff slots) + one resize kernel,
+        all on the side stream."""
+        meta, win, flags = decoded
         pool = self._shm_pool
         out = self._out[:n]
         lo = pool.hdr + pool.in_bytes + win
         src = self._shm_t[lo:lo + n * pool.slot_bytes]
         with torch.cuda.stream(self.stream):
             arena_gpu = src.to(self.device, non_blocking=True)
+            if flags is not None and flags.any():
+                if self._planes is None:
+                    self._planes = torch.empty(
+                        (pool.half + 1) * pool.slot_bytes,
+                        dtype=torch.uint8, device=self.device)
+                fl = torch.from_numpy(flags).to(self.device,
+                                                non_blocking=True)
+                self.E.jpeg_decode_slots(arena_gpu, self._planes, fl,
+                                         n, pool.slot_bytes)
             meta_gpu = torch.from_numpy(meta).to(self.device,
                                                  non_blocking=True)
             self.E.resize_norm_batch(arena_gpu, meta_gpu, out, self.norm)

@@ -66,3 +66,62 @@ def test_gpu_ingest_feeds_engine():
     torch.cuda.synchronize()
     s = probs.sum(-1).float().cpu().numpy()
     np.testing.assert_allclose(s, 1.0, atol=1e-3)
+
+
+def test_gpu_jpeg_decode_path_matches_pil():
+    """Coeff slots -> HIP dequant/IDCT/upsample/color -> resize: end to
+    end vs the PIL-decode + torch-bilinear reference. Asserts the GPU
+    jpeg path actually ran (worker flags), then pixel agreement."""
+    from PIL import Image
+
+    from tensorflow_web_deploy_amd.ingest.gpu_pipeline import GpuPreprocessor
+
+    rng = np.random.default_rng(3)
+    base = rng.random((8, 8, 3))
+    arr = (np.kron(base, np.ones((40, 40, 1)))[:301, :263]
+           * 255).astype(np.uint8)
+
+    def enc(sub):
+        b = io.BytesIO()
+        Image.fromarray(arr, "RGB").save(b, "JPEG", quality=92,
+                                         subsampling=sub)
+        return b.getvalue()
+
+    blobs = [enc(2), enc(0), _jpeg_bytes(120, 90)]  # 420, 444, PNG mix
+    pre = GpuPreprocessor(224, 224, 3, "unit")
+    pre._ensure_shm()
+    assert pre._shm_pool.jpeg, "GPU jpeg path not enabled"
+    out = pre.process(blobs)
+    torch.cuda.synchronize()
+    assert list(pre._shm_pool.last_flags) == [1, 1, 0]
+    got = out.cpu().numpy()
+    for i, blob in enumerate(blobs):
+        src = np.asarray(Image.open(io.BytesIO(blob)).convert("RGB"),
+                         dtype=np.float32) / 255.0
+        st = torch.from_numpy(src).permute(2, 0, 1)[None]
+        ref = torch.nn.functional.interpolate(
+            st, size=(224, 224), mode="bilinear",
+            align_corners=False)[0].permute(1, 2, 0).numpy()
+        d = np.abs(got[i] - ref)
+        # jpeg decode itself agrees with PIL to <=2/255; resize is exact
+        assert d.max() < 0.03 and d.mean() < 0.005, (i, d.max(), d.mean())
+
+
+def test_gpu_jpeg_disabled_fallback():
+    """TWD_GPU_JPEG=0 keeps the pixel path (flags stay 0)."""
+    import os
+
+    from tensorflow_web_deploy_amd.ingest.gpu_pipeline import GpuPreprocessor
+    os.environ["TWD_GPU_JPEG"] = "0"
+    try:
+        pre = GpuPreprocessor(64, 64, 3, "unit")
+        from PIL import Image
+        arr = (np.random.rand(80, 80, 3) * 255).astype("uint8")
+        b = io.BytesIO()
+        Image.fromarray(arr, "RGB").save(b, "JPEG", quality=90)
+        out = pre.process([b.getvalue()] * 6)
+        torch.cuda.synchronize()
+        assert not pre._shm_pool.jpeg
+        assert out.shape == (6, 64, 64, 3)
+    finally:
+        os.environ.pop("TWD_GPU_JPEG")
