@@ -118,6 +118,14 @@ def _ring_worker(path, widx, nworkers, channels, slot_bytes, slots,
     With jpeg=True, baseline JPEGs are entropy-decoded only (_jpegc)
     and written as COEFF slots (status 3) for the GPU IDCT kernels;
     anything else falls back to full PIL pixel decode."""
+    import sys
+    import traceback
+
+    def _die(exc):
+        traceback.print_exc(file=sys.stderr)
+        sys.stderr.flush()
+        raise exc
+
     import gc
     gc.collect()
     gc.freeze()
@@ -178,6 +186,17 @@ def _ring_worker(path, widx, nworkers, channels, slot_bytes, slots,
             done_tx.send_bytes(b"d")
         except (BrokenPipeError, OSError):
             return
+
+
+def _ring_worker_wrap(*a, **kw):
+    try:
+        _ring_worker_inner(*a, **kw)
+    except Exception:
+        import sys
+        import traceback
+        traceback.print_exc(file=sys.stderr)
+        sys.stderr.flush()
+        raise
 
 
 class ShmDecodePool:
@@ -283,11 +302,22 @@ class ShmDecodePool:
         try:
             for j in range(nw):
                 if not self._done[j].poll(timeout):
+                    import sys
                     self._dead = True  # worker gone: don't trust the ring
+                    print(f"[ShmDecodePool] worker {j} timeout "
+                          f"({timeout}s); falling back", file=sys.stderr,
+                          flush=True)
                     return None
                 self._done[j].recv_bytes()
-        except (EOFError, OSError):
+        except (EOFError, OSError) as e:
+            import sys
             self._dead = True
+            dead = [j for j, pr in enumerate(self._procs)
+                    if not pr.is_alive()]
+            codes = [self._procs[j].exitcode for j in dead]
+            print(f"[ShmDecodePool] worker pipe error {e!r}; dead "
+                  f"workers {dead} exitcodes {codes}; falling back to "
+                  f"the pipe decode path", file=sys.stderr, flush=True)
             return None
         meta = np.empty((n, 4), np.int32)
         flags = np.zeros(n, np.uint8)

@@ -87,13 +87,16 @@ def test_gpu_jpeg_decode_path_matches_pil():
                                          subsampling=sub)
         return b.getvalue()
 
-    blobs = [enc(2), enc(0), _jpeg_bytes(120, 90)]  # 420, 444, PNG mix
+    # >4 blobs: batches of <=4 take the direct small-batch path and
+    # never touch the shm ring at all
+    blobs = [enc(2), enc(0), _jpeg_bytes(120, 90), enc(2), enc(0),
+             _jpeg_bytes(77, 133)]  # 420/444/PNG mix
     pre = GpuPreprocessor(224, 224, 3, "unit")
     pre._ensure_shm()
     assert pre._shm_pool.jpeg, "GPU jpeg path not enabled"
     out = pre.process(blobs)
     torch.cuda.synchronize()
-    assert list(pre._shm_pool.last_flags) == [1, 1, 0]
+    assert list(pre._shm_pool.last_flags) == [1, 1, 0, 1, 1, 0]
     got = out.cpu().numpy()
     for i, blob in enumerate(blobs):
         src = np.asarray(Image.open(io.BytesIO(blob)).convert("RGB"),
@@ -122,6 +125,7 @@ def test_gpu_jpeg_disabled_fallback():
         out = pre.process([b.getvalue()] * 6)
         torch.cuda.synchronize()
         assert not pre._shm_pool.jpeg
+        assert not pre._shm_pool.last_flags.any()
         assert out.shape == (6, 64, 64, 3)
     finally:
         os.environ.pop("TWD_GPU_JPEG")
