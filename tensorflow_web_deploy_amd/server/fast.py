@@ -154,21 +154,30 @@ class FastServer:
         A while batch N's DMA from it is still in flight). Main thread:
         upload + model + respond. Fallback paths never touch the ring
         (it is single-decoder by design)."""
+        import os
         import queue
+        import time as _t
         q = queue.Queue()
         free_q = queue.Queue()
         free_q.put(0)
         free_q.put(1)
         eng = self.engine
+        stats = ({"wait": 0.0, "decode": 0.0, "token": 0.0, "gpu": 0.0,
+                  "json": 0.0, "respond": 0.0, "batches": 0, "imgs": 0}
+                 if os.environ.get("TWD_FAST_STATS") else None)
+        self.stats = stats
 
         def ingest():
             while not self._stop.is_set():
+                t0 = _t.perf_counter()
                 batch = self.fe.wait_batch(self.cfg.max_batch,
                                            self.cfg.max_wait_ms)
                 if not batch:
                     continue
+                t1 = _t.perf_counter()
                 bodies = [b for (_, b) in batch]
                 ph = free_q.get()
+                t2 = _t.perf_counter()
                 try:
                     decoded = eng.pre.decode_shm(bodies, phase=ph)
                 except Exception:
@@ -176,6 +185,13 @@ class FastServer:
                 if decoded is None:
                     free_q.put(ph)  # half unused
                     ph = None
+                if stats is not None:
+                    t3 = _t.perf_counter()
+                    stats["wait"] += t1 - t0
+                    stats["token"] += t2 - t1
+                    stats["decode"] += t3 - t2
+                    stats["batches"] += 1
+                    stats["imgs"] += len(batch)
                 q.put((batch, bodies, decoded, ph))
             q.put(None)
 
@@ -191,13 +207,18 @@ class FastServer:
                 if item is None:
                     return
                 batch, probs, results = item
+                t0 = _t.perf_counter()
                 if results is None:
                     payloads = self._payloads(probs)
                     results = [(200, pl) for pl in payloads]
+                t1 = _t.perf_counter()
                 self.fe.respond_many([
                     (rid, st, body.encode())
                     for (rid, _), (st, body) in zip(batch, results)
                 ])
+                if stats is not None:
+                    stats["json"] += t1 - t0
+                    stats["respond"] += _t.perf_counter() - t1
 
         t = threading.Thread(target=ingest, daemon=True)
         t.start()
@@ -211,7 +232,10 @@ class FastServer:
                 batch, bodies, decoded, ph = item
                 if decoded is not None:
                     try:
+                        t0 = _t.perf_counter()
                         probs = eng.infer_decoded(decoded, len(bodies))
+                        if stats is not None:
+                            stats["gpu"] += _t.perf_counter() - t0
                         respq.put((batch, probs, None))
                         continue
                     except Exception:
@@ -223,6 +247,14 @@ class FastServer:
                 respq.put((batch, None, results))
         finally:
             respq.put(None)
+            if stats is not None and stats["batches"]:
+                import sys
+                b = stats["batches"]
+                print("[fast stats] " + " ".join(
+                    f"{k}={1000 * v / b:.2f}ms" for k, v in stats.items()
+                    if isinstance(v, float)) +
+                    f" batches={b} avg_bs={stats['imgs'] / b:.1f}",
+                    file=sys.stderr, flush=True)
             self.fe.stop()
 
     def start_background(self) -> threading.Thread:
