@@ -134,6 +134,31 @@ class GpuServingEngine(LocalEngine):
         return out.float().cpu().numpy()[:n]
 
     @torch.inference_mode()
+    def upload_decoded(self, decoded, n: int, phase):
+        """Ingest-thread half of the pipelined fast path: enqueue
+        H2D + jpeg + resize + bucket-pad on the side stream; returns
+        (batch_tensor, event, n) for infer_uploaded in the GPU thread.
+        Nothing here touches the default stream."""
+        out, ev = self.pre.upload_shm(decoded, n, phase=phase,
+                                      sync=False)
+        nb = _bucket(n)
+        if nb != n:
+            full = self.pre._out_buf(phase)[:nb]
+            with torch.cuda.stream(self.pre.stream):
+                full[n:].zero_()
+            ev = torch.cuda.Event()
+            ev.record(self.pre.stream)
+            out = full
+        return out, ev, n
+
+    @torch.inference_mode()
+    def infer_uploaded(self, handle) -> np.ndarray:
+        batch, ev, n = handle
+        torch.cuda.current_stream().wait_event(ev)
+        out = self.ex.run({self.cfg.input_name: batch})[0]
+        return out.float().cpu().numpy()[:n]
+
+    @torch.inference_mode()
     def infer_decoded(self, decoded, n: int) -> np.ndarray:
         """GPU stage of the pipelined fast server: slot window ->
         H2D + resize (side stream) -> model -> probs. The CPU decode

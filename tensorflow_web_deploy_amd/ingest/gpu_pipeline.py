@@ -383,6 +383,7 @@ class GpuPreprocessor:
         self._shm_t = None
         self._shm_pinned = False
         self._planes = None  # GPU scratch for the jpeg IDCT kernels
+        self._out_b = None   # second output batch (pipeline phase 1)
 
     def _decode_cpu(self, data: bytes) -> np.ndarray:
         from PIL import Image
@@ -442,15 +443,27 @@ class GpuPreprocessor:
             return None
         return meta, pool.last_window_off, pool.last_flags
 
+    def _out_buf(self, phase):
+        """Double-buffered output batches: the pipelined server uploads
+        batch N+1 (side stream) while batch N's forward still reads its
+        output tensor; phase tokens gate reuse."""
+        if phase is None or phase == 0:
+            return self._out
+        if self._out_b is None:
+            self._out_b = torch.empty_like(self._out)
+        return self._out_b
+
     def upload_shm(self, decoded, n: int,
-                   compute_stream: Optional[torch.cuda.Stream] = None
-                   ) -> torch.Tensor:
+                   compute_stream: Optional[torch.cuda.Stream] = None,
+                   phase: Optional[int] = None, sync: bool = True):
         """GPU stage: one pinned DMA of the slot window (+ the jpeg
         dequant/IDCT/color kernels for coeff slots) + one resize kernel,
-        all on the side stream."""
+        all on the side stream. sync=False returns (out, event) with NO
+        wait enqueued — the consumer thread waits on the event itself
+        (cross-thread default-stream waits would order arbitrarily)."""
         meta, win, flags = decoded
         pool = self._shm_pool
-        out = self._out[:n]
+        out = self._out_buf(phase)[:n]
         lo = pool.hdr + pool.in_bytes + win
         src = self._shm_t[lo:lo + n * pool.slot_bytes]
         with torch.cuda.stream(self.stream):
@@ -469,6 +482,8 @@ class GpuPreprocessor:
             self.E.resize_norm_batch(arena_gpu, meta_gpu, out, self.norm)
         ev = torch.cuda.Event()
         ev.record(self.stream)
+        if not sync:
+            return out, ev
         target = compute_stream or torch.cuda.current_stream()
         target.wait_event(ev)
         return out

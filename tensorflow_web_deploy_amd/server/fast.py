@@ -167,6 +167,11 @@ class FastServer:
                  if os.environ.get("TWD_FAST_STATS") else None)
         self.stats = stats
 
+        # engines that can split upload (side stream, ingest thread)
+        # from forward (default stream, this thread) overlap the H2D +
+        # preprocess of batch N+1 with the forward of batch N
+        can_upload = hasattr(eng, "upload_decoded")
+
         def ingest():
             while not self._stop.is_set():
                 t0 = _t.perf_counter()
@@ -178,8 +183,12 @@ class FastServer:
                 bodies = [b for (_, b) in batch]
                 ph = free_q.get()
                 t2 = _t.perf_counter()
+                handle = None
                 try:
                     decoded = eng.pre.decode_shm(bodies, phase=ph)
+                    if decoded is not None and can_upload:
+                        handle = eng.upload_decoded(decoded,
+                                                    len(bodies), ph)
                 except Exception:
                     decoded = None  # bad image: isolate downstream
                 if decoded is None:
@@ -192,7 +201,7 @@ class FastServer:
                     stats["decode"] += t3 - t2
                     stats["batches"] += 1
                     stats["imgs"] += len(batch)
-                q.put((batch, bodies, decoded, ph))
+                q.put((batch, bodies, decoded, handle, ph))
             q.put(None)
 
         # stage 3: top-k + JSON + respond off the GPU thread (the
@@ -229,11 +238,15 @@ class FastServer:
                 item = q.get()
                 if item is None:
                     break
-                batch, bodies, decoded, ph = item
+                batch, bodies, decoded, handle, ph = item
                 if decoded is not None:
                     try:
                         t0 = _t.perf_counter()
-                        probs = eng.infer_decoded(decoded, len(bodies))
+                        if handle is not None:
+                            probs = eng.infer_uploaded(handle)
+                        else:
+                            probs = eng.infer_decoded(decoded,
+                                                      len(bodies))
                         if stats is not None:
                             stats["gpu"] += _t.perf_counter() - t0
                         respq.put((batch, probs, None))
@@ -241,7 +254,7 @@ class FastServer:
                     except Exception:
                         results = self._infer_isolating(bodies)
                     finally:
-                        free_q.put(ph)  # H2D synced inside infer_decoded
+                        free_q.put(ph)  # forward synced: half reusable
                 else:
                     results = self._infer_isolating(bodies)
                 respq.put((batch, None, results))
