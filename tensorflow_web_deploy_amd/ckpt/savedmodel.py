@@ -79,21 +79,39 @@ def parse_savedmodel_pb(data: bytes):
 
 def _materialize_variables(g: gd.GraphDef, bundle: BundleReader) -> None:
     """Rewrite VariableV2/VarHandleOp nodes into Const nodes with the
-    bundle values; collapse ReadVariableOp onto them."""
+    bundle values; collapse ReadVariableOp onto them.
+
+    Key resolution, in order: (1) the node name itself (TF1 name-based
+    checkpoints), (2) name + '/.ATTRIBUTES/VARIABLE_VALUE' (the common
+    TF2 layout when object paths mirror variable names), (3) the
+    checkpoint's TrackableObjectGraph (`ckpt/object_graph.py`) — TF2
+    object-based checkpoints record full_name -> checkpoint_key there,
+    which covers renamed/nested objects."""
+    from . import object_graph as og
     var_nodes = {}
     for n in g.node:
         if n.op in ("VariableV2", "Variable", "VarHandleOp"):
             var_nodes[n.name] = n
+    og_index = None
+    og_loaded = False
     for name, n in var_nodes.items():
         key = name
         if key not in bundle.entries:
-            # TF2-style keys can carry a suffix
             alt = name + "/.ATTRIBUTES/VARIABLE_VALUE"
             if alt in bundle.entries:
                 key = alt
             else:
-                raise KeyError(f"variable {name} not in bundle "
-                               f"(keys: {list(bundle.keys())[:5]}...)")
+                if not og_loaded:
+                    og_index = og.load_index(bundle)
+                    og_loaded = True
+                resolved = (og_index.resolve(name)
+                            if og_index is not None else None)
+                if resolved is not None and resolved in bundle.entries:
+                    key = resolved
+                else:
+                    raise KeyError(
+                        f"variable {name} not in bundle "
+                        f"(keys: {list(bundle.keys())[:5]}...)")
         arr = bundle.get(key)
         n.op = "Const"
         n.input = []
