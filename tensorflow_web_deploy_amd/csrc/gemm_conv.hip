@@ -1144,6 +1144,22 @@ static void dispatch(const void* A, const void* B, void* C, const float* bias,
   // cross-block overlap that was hiding half the staging latency):
   // stem 537->351 TF, s1 3x3 500->325. Keep DEPTH=2 + 2 blocks/CU.
   if (N <= 96) {
+    // A/B harness for the wait-bound s1 class: (128,64) halves the
+    // tile so 2-3 blocks/CU overlap more staging (TWD_S1=1: DEPTH 3 at
+    // 2 blocks/CU; TWD_S1=2: DEPTH 2 at 3 blocks/CU)
+    const char* s1 = getenv("TWD_S1");
+    if (s1 && s1[0] == '1') {
+      launch_tile<128, 64, MODE, 3>(A, B, C, bias, residual, M, N, K, Kp,
+                                    a_bytes, act, cp, stream, ws, splitk,
+                                    ldc);
+      return;
+    }
+    if (s1 && s1[0] == '2') {
+      launch_tile<128, 64, MODE, 2>(A, B, C, bias, residual, M, N, K, Kp,
+                                    a_bytes, act, cp, stream, ws, splitk,
+                                    ldc);
+      return;
+    }
     launch_tile<256, 64, MODE, 2>(A, B, C, bias, residual, M, N, K, Kp,
                                   a_bytes, act, cp, stream, ws, splitk, ldc);
     return;
