@@ -26,6 +26,7 @@ ext = cpp_extension.CUDAExtension(
         os.path.join(CSRC, "attn_fused.hip"),
         os.path.join(CSRC, "depthwise.hip"),
         os.path.join(CSRC, "jpeg_gpu.hip"),
+        os.path.join(CSRC, "gemm_8p.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

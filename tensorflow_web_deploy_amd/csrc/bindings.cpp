@@ -341,6 +341,23 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
                        (int)sh, (int)sw, (int)pt, (int)pl, (int)act,
                        c10::hip::getCurrentHIPStream());
   });
+  m.def("gemm8p_bf16", [](torch::Tensor A, torch::Tensor B,
+                          torch::Tensor C,
+                          c10::optional<torch::Tensor> bias,
+                          c10::optional<torch::Tensor> residual,
+                          int64_t K, int64_t act) {
+    check_bf16(A, "A");
+    check_bf16(B, "B");
+    int ldc = check_out2d(C, "C");
+    int64_t M = C.size(0), N = C.size(1), Kp = B.size(1);
+    const float* bptr = nullptr;
+    if (bias) bptr = bias->data_ptr<float>();
+    const void* rptr = nullptr;
+    if (residual) rptr = residual->data_ptr();
+    launch_gemm_8p(A.data_ptr(), B.data_ptr(), C.data_ptr(), bptr, rptr,
+                   (int)M, (int)N, (int)K, (int)Kp, (int)act, ldc,
+                   c10::hip::getCurrentHIPStream());
+  });
   m.def("jpeg_decode_slots", [](torch::Tensor arena, torch::Tensor planes,
                                 torch::Tensor flags, int64_t n,
                                 int64_t slot_bytes) {

@@ -27,6 +27,11 @@ void launch_dwconv_bf16(const void* x, const void* w, const float* bias,
 void launch_jpeg_decode(void* arena, void* planes, const void* flags,
                         int n, int64_t slot_bytes, hipStream_t stream);
 
+void launch_gemm_8p(const void* A, const void* B, void* C,
+                    const float* bias, const void* residual, int M,
+                    int N, int K, int Kp, int act, int ldc,
+                    hipStream_t stream);
+
 // pool_reduce.hip
 void launch_maxpool_bf16(const void* X, void* Y, int Nimg, int H, int W,
                          int C, int kh, int kw, int sh, int sw, int pt,
