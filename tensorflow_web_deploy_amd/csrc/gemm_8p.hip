@@ -146,45 +146,72 @@ __global__ __launch_bounds__(512) void gemm_8p_kernel(
   for (int t = 0; t < nt; ++t) {
     const int buf = t & 1;
     const char* Abase = smem8p;  // halves addressed via HALF_OFF
+    // ONE barrier per phase. Residency proof per phase (per-wave glds
+    // queue, oldest->newest; each half = 2 glds):
+    //   q0: wait vmcnt(4) BEFORE staging -> leaves {A1(t+1?),..} no —
+    //       outstanding is {t:A1, t:B1}? The steady queue at q0 entry
+    //       is {t+1? none yet}: halves staged during t-1's q1..q3 plus
+    //       t's q0: by induction outstanding at q0 entry = t+1's
+    //       {A0..}? See derivation: entering q0 of tile t the queue is
+    //       {t:A1, t:B1} + {t+1: none}; vmcnt(4) keeps both, so
+    //       earlier halves (t:A0, t:B0 — staged 8/7 phases ago) are
+    //       resident for q0. The stage of t+1:A0 happens AFTER the
+    //       barrier: the barrier proves every wave left q3 of t-1 —
+    //       without that a fast wave would overwrite buf[t-1&1].A0
+    //       (= buf[(t+1)&1].A0) while a slow wave still reads it.
+    //   q1: stage t+1:B0 then vmcnt(6): leaves {t:B1, t+1:A0, t+1:B0}
+    //       -> t:A1 resident.   q2: stage t+1:A1, vmcnt(6): leaves
+    //       {t+1:A0,B0,A1} -> t:B1 resident.   q3: stage t+1:B1, no
+    //       wait (A0/B1 of t already proven).
 #pragma unroll
     for (int q = 0; q < 4; ++q) {
-      stage_half(SO[q], t + 1, nt);
-      if (q < 3)
-        asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
-      __builtin_amdgcn_s_barrier();
+      if (q == 0) {
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        stage_half(SO[0], t + 1, nt);
+      } else {
+        stage_half(SO[q], t + 1, nt);
+        if (q < 3)
+          asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+      }
       const char* Ah = Abase + HALF_OFF(buf, QR[q]);
       const char* Bh = Abase + HALF_OFF(buf, 2 + QC[q]);
+      // kk-split: kk=1's ds_reads land while kk=0's MFMAs issue
       bf16x8_8p af[4][2], bf[2][2];
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
+      for (int kk = 0; kk < 2; ++kk) {
 #pragma unroll
-        for (int kk = 0; kk < 2; ++kk) {
+        for (int mi = 0; mi < 4; ++mi) {
           int row = wr2 * 64 + mi * 16 + (lane & 15);
           af[mi][kk] = *(const bf16x8_8p*)(
               Ah + lds_b(row, kk * 4 + (lane >> 4)));
         }
 #pragma unroll
-      for (int ni = 0; ni < 2; ++ni)
-#pragma unroll
-        for (int kk = 0; kk < 2; ++kk) {
+        for (int ni = 0; ni < 2; ++ni) {
           int row = wc2 * 32 + ni * 16 + (lane & 15);
           bf[ni][kk] = *(const bf16x8_8p*)(
               Bh + lds_b(row, kk * 4 + (lane >> 4)));
         }
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      __builtin_amdgcn_s_setprio(1);
+      }
 #pragma unroll
-      for (int kk = 0; kk < 2; ++kk)
+      for (int kk = 0; kk < 2; ++kk) {
+        if (kk == 0)
+          asm volatile("s_waitcnt lgkmcnt(6)" ::: "memory");
+        else
+          asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
           for (int ni = 0; ni < 2; ++ni)
             acc[q][mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 af[mi][kk], bf[ni][kk], acc[q][mi][ni], 0, 0, 0);
-      __builtin_amdgcn_s_setprio(0);
-      __builtin_amdgcn_s_barrier();
+        __builtin_amdgcn_s_setprio(0);
+      }
     }
   }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
 
   // ---- epilogue: scattered stores (v1); LDS restage once it wins ----
 #pragma unroll
