@@ -143,76 +143,89 @@ __global__ __launch_bounds__(512) void gemm_8p_kernel(
   stage_half(1, 0, nt);
   stage_half(3, 0, nt);
 
+  // A-frags are read ONE PHASE AHEAD (double-buffered; their 8
+  // ds_reads hide under the previous phase's MFMA burst); B-frags are
+  // read in-phase after the drain (4 ds_reads of exposed latency —
+  // the lgkm counter's 15-op ceiling forbids pipelining both).
+  bf16x8_8p af[2][4][2], bf[2][2];
+
+  auto issue_a = [&](int pipe, int buf, int q) {
+    const char* Ah = smem8p + HALF_OFF(buf, QR[q]);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) {
+        int row = wr2 * 64 + mi * 16 + (lane & 15);
+        af[pipe][mi][kk] = *(const bf16x8_8p*)(
+            Ah + lds_b(row, kk * 4 + (lane >> 4)));
+      }
+  };
+  auto issue_b = [&](int buf, int q) {
+    const char* Bh = smem8p + HALF_OFF(buf, 2 + QC[q]);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni) {
+        int row = wc2 * 32 + ni * 16 + (lane & 15);
+        bf[ni][kk] = *(const bf16x8_8p*)(
+            Bh + lds_b(row, kk * 4 + (lane >> 4)));
+      }
+  };
+
+  // prologue: verify tile0 A0/B0 (vmcnt(4): A1,B1 may still fly),
+  // sync, pre-issue q0's A-frags
+  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  issue_a(0, 0, 0);
+
+  // Residency schedule (per-wave glds queue; sim in git history):
+  //   q0: vmcnt(2) pre-barrier (verifies t:A1 for q1's reads); stage
+  //       of t+1:A0 goes AFTER the barrier — it writes buf[t+1] but
+  //       the NEXT tile's q0 stage would hit the buffer the slowest
+  //       wave may still read, so the barrier must separate them.
+  //   q1: stage t+1:B0; vmcnt(4) (verifies t:B1 for q2's reads).
+  //   q2: stage t+1:A1; no wait (q3 needs A0/B1, both proven).
+  //   q3: stage t+1:B1; vmcnt(4) (verifies t+1:A0,B0 for next q0's
+  //       read-ahead from the other buffer).
   for (int t = 0; t < nt; ++t) {
     const int buf = t & 1;
-    const char* Abase = smem8p;  // halves addressed via HALF_OFF
-    // ONE barrier per phase. Residency proof per phase (per-wave glds
-    // queue, oldest->newest; each half = 2 glds):
-    //   q0: wait vmcnt(4) BEFORE staging -> leaves {A1(t+1?),..} no —
-    //       outstanding is {t:A1, t:B1}? The steady queue at q0 entry
-    //       is {t+1? none yet}: halves staged during t-1's q1..q3 plus
-    //       t's q0: by induction outstanding at q0 entry = t+1's
-    //       {A0..}? See derivation: entering q0 of tile t the queue is
-    //       {t:A1, t:B1} + {t+1: none}; vmcnt(4) keeps both, so
-    //       earlier halves (t:A0, t:B0 — staged 8/7 phases ago) are
-    //       resident for q0. The stage of t+1:A0 happens AFTER the
-    //       barrier: the barrier proves every wave left q3 of t-1 —
-    //       without that a fast wave would overwrite buf[t-1&1].A0
-    //       (= buf[(t+1)&1].A0) while a slow wave still reads it.
-    //   q1: stage t+1:B0 then vmcnt(6): leaves {t:B1, t+1:A0, t+1:B0}
-    //       -> t:A1 resident.   q2: stage t+1:A1, vmcnt(6): leaves
-    //       {t+1:A0,B0,A1} -> t:B1 resident.   q3: stage t+1:B1, no
-    //       wait (A0/B1 of t already proven).
 #pragma unroll
     for (int q = 0; q < 4; ++q) {
       if (q == 0) {
-        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
         __builtin_amdgcn_s_barrier();
         stage_half(SO[0], t + 1, nt);
       } else {
         stage_half(SO[q], t + 1, nt);
-        if (q < 3)
-          asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+        if (q != 2)
+          asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
         __builtin_amdgcn_s_barrier();
       }
-      const char* Ah = Abase + HALF_OFF(buf, QR[q]);
-      const char* Bh = Abase + HALF_OFF(buf, 2 + QC[q]);
-      // kk-split: kk=1's ds_reads land while kk=0's MFMAs issue
-      bf16x8_8p af[4][2], bf[2][2];
+      // B of this phase after the drain; A of the NEXT phase before
+      // the MFMAs (q3 pre-reads tile t+1's q0 A from buf^1; on the
+      // last tile those values are never consumed)
+      if ((q & 1) == 0)  // q1/q3 reuse the pair's B-half frags
+        issue_b(buf, q);
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      if (q < 3)
+        issue_a((q + 1) & 1, buf, q + 1);
+      else
+        issue_a(0, buf ^ 1, 0);
+      __builtin_amdgcn_s_setprio(1);
+      const int pipe = q & 1;
 #pragma unroll
-      for (int kk = 0; kk < 2; ++kk) {
-#pragma unroll
-        for (int mi = 0; mi < 4; ++mi) {
-          int row = wr2 * 64 + mi * 16 + (lane & 15);
-          af[mi][kk] = *(const bf16x8_8p*)(
-              Ah + lds_b(row, kk * 4 + (lane >> 4)));
-        }
-#pragma unroll
-        for (int ni = 0; ni < 2; ++ni) {
-          int row = wc2 * 32 + ni * 16 + (lane & 15);
-          bf[ni][kk] = *(const bf16x8_8p*)(
-              Bh + lds_b(row, kk * 4 + (lane >> 4)));
-        }
-      }
-#pragma unroll
-      for (int kk = 0; kk < 2; ++kk) {
-        if (kk == 0)
-          asm volatile("s_waitcnt lgkmcnt(6)" ::: "memory");
-        else
-          asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        __builtin_amdgcn_s_setprio(1);
+      for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
           for (int ni = 0; ni < 2; ++ni)
             acc[q][mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                af[mi][kk], bf[ni][kk], acc[q][mi][ni], 0, 0, 0);
-        __builtin_amdgcn_s_setprio(0);
-      }
+                af[pipe][mi][kk], bf[ni][kk], acc[q][mi][ni], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_barrier();
     }
   }
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-
+  asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
   // ---- epilogue: scattered stores (v1); LDS restage once it wins ----
 #pragma unroll
   for (int q = 0; q < 4; ++q) {
