@@ -143,25 +143,27 @@ __global__ __launch_bounds__(512) void gemm_8p_kernel(
   stage_half(1, 0, nt);
   stage_half(3, 0, nt);
 
-  // A-frags are read ONE PHASE AHEAD (double-buffered; their 8
-  // ds_reads hide under the previous phase's MFMA burst); B-frags are
-  // read in-phase after the drain (4 ds_reads of exposed latency —
-  // the lgkm counter's 15-op ceiling forbids pipelining both).
-  bf16x8_8p af[2][4][2], bf[2][2];
+  // Fragment reuse across the quadrant order (0,0),(1,0),(1,1),(0,1):
+  // q0 and q3 use the SAME A-frags (half 0, same wave rows), q1/q2
+  // share half 1; q0/q1 share B-pair 0, q2/q3 pair 1. So each A set
+  // and B pair is read ONCE per K-tile (24 ds_reads, not 40) and held
+  // in registers. A sets are issued where the previous phase's MFMA
+  // burst hides them; B pairs are the only exposed reads (4 each).
+  bf16x8_8p af[2][4][2], bf[2][2];  // af[set][mi][kk], bf[ni][kk]
 
-  auto issue_a = [&](int pipe, int buf, int q) {
-    const char* Ah = smem8p + HALF_OFF(buf, QR[q]);
+  auto issue_a = [&](int set, int buf) {
+    const char* Ah = smem8p + HALF_OFF(buf, set);
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi) {
         int row = wr2 * 64 + mi * 16 + (lane & 15);
-        af[pipe][mi][kk] = *(const bf16x8_8p*)(
+        af[set][mi][kk] = *(const bf16x8_8p*)(
             Ah + lds_b(row, kk * 4 + (lane >> 4)));
       }
   };
-  auto issue_b = [&](int buf, int q) {
-    const char* Bh = smem8p + HALF_OFF(buf, 2 + QC[q]);
+  auto issue_b = [&](int pair, int buf) {
+    const char* Bh = smem8p + HALF_OFF(buf, 2 + pair);
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
@@ -172,58 +174,68 @@ __global__ __launch_bounds__(512) void gemm_8p_kernel(
       }
   };
 
-  // prologue: verify tile0 A0/B0 (vmcnt(4): A1,B1 may still fly),
-  // sync, pre-issue q0's A-frags
+  auto mfma_q = [&](int q, int set) {
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+          acc[q][mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[set][mi][kk], bf[ni][kk], acc[q][mi][ni], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+  };
+
+  // prologue: verify tile0 A0/B0, sync, pre-read A set 0
   asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
   __builtin_amdgcn_s_barrier();
-  issue_a(0, 0, 0);
+  issue_a(0, 0);
 
-  // Residency schedule (per-wave glds queue; sim in git history):
-  //   q0: vmcnt(2) pre-barrier (verifies t:A1 for q1's reads); stage
-  //       of t+1:A0 goes AFTER the barrier — it writes buf[t+1] but
-  //       the NEXT tile's q0 stage would hit the buffer the slowest
-  //       wave may still read, so the barrier must separate them.
-  //   q1: stage t+1:B0; vmcnt(4) (verifies t:B1 for q2's reads).
-  //   q2: stage t+1:A1; no wait (q3 needs A0/B1, both proven).
-  //   q3: stage t+1:B1; vmcnt(4) (verifies t+1:A0,B0 for next q0's
-  //       read-ahead from the other buffer).
+  // Waits as derived for the v3 schedule (sim in git history):
+  //   q0: vmcnt(2) pre-barrier (t:A1 proven for q1); stage t+1:A0
+  //       AFTER the barrier (its next-tile counterpart would overwrite
+  //       the half the slowest wave still reads).
+  //   q1: stage t+1:B0; vmcnt(4) (t:B1 proven for q2).
+  //   q2: stage t+1:A1; no wait.
+  //   q3: stage t+1:B1; vmcnt(4) (t+1:A0 proven -> pre-read next A0
+  //       from the other buffer under q3's MFMAs).
   for (int t = 0; t < nt; ++t) {
     const int buf = t & 1;
-#pragma unroll
-    for (int q = 0; q < 4; ++q) {
-      if (q == 0) {
-        asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
-        __builtin_amdgcn_s_barrier();
-        stage_half(SO[0], t + 1, nt);
-      } else {
-        stage_half(SO[q], t + 1, nt);
-        if (q != 2)
-          asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-        __builtin_amdgcn_s_barrier();
-      }
-      // B of this phase after the drain; A of the NEXT phase before
-      // the MFMAs (q3 pre-reads tile t+1's q0 A from buf^1; on the
-      // last tile those values are never consumed)
-      if ((q & 1) == 0)  // q1/q3 reuse the pair's B-half frags
-        issue_b(buf, q);
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      if (q < 3)
-        issue_a((q + 1) & 1, buf, q + 1);
-      else
-        issue_a(0, buf ^ 1, 0);
-      __builtin_amdgcn_s_setprio(1);
-      const int pipe = q & 1;
-#pragma unroll
-      for (int kk = 0; kk < 2; ++kk)
-#pragma unroll
-        for (int mi = 0; mi < 4; ++mi)
-#pragma unroll
-          for (int ni = 0; ni < 2; ++ni)
-            acc[q][mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                af[pipe][mi][kk], bf[ni][kk], acc[q][mi][ni], 0, 0, 0);
-      __builtin_amdgcn_s_setprio(0);
-      __builtin_amdgcn_s_barrier();
-    }
+    // ---- q0 ----
+    asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    stage_half(0, t + 1, nt);
+    issue_b(0, buf);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    issue_a(1, buf);  // half 1 for q1/q2, hidden under q0's MFMAs
+    mfma_q(0, 0);
+    __builtin_amdgcn_s_barrier();
+    // ---- q1 ----
+    stage_half(2, t + 1, nt);
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    mfma_q(1, 1);
+    __builtin_amdgcn_s_barrier();
+    // ---- q2 ----
+    stage_half(1, t + 1, nt);
+    __builtin_amdgcn_s_barrier();
+    issue_b(1, buf);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    mfma_q(2, 1);
+    __builtin_amdgcn_s_barrier();
+    // ---- q3 ----
+    stage_half(3, t + 1, nt);
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    mfma_q(3, 0);
+    // next tile's A0 AFTER q3's MFMAs consume set 0 (prefetching it
+    // above would create a register WAR the scheduler resolves by
+    // serializing — wrong data cannot occur, but the MFMAs would wait
+    // on next-tile loads; a 3rd A set doesn't fit in 256 VGPRs)
+    issue_a(0, buf ^ 1);
+    __builtin_amdgcn_s_barrier();
   }
   asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
   // ---- epilogue: scattered stores (v1); LDS restage once it wins ----
