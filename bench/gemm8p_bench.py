@@ -1,3 +1,5 @@
+"""Numerics + perf harness for the experimental 8-phase 256^2 GEMM
+(csrc/gemm_8p.hip, unrouted — see NOTES_ROUND2.md item 2)."""
 import sys
 sys.path.insert(0, "/root/repo")
 import torch
