@@ -229,20 +229,26 @@ class FastServer:
                     stats["json"] += t1 - t0
                     stats["respond"] += _t.perf_counter() - t1
 
-        t = threading.Thread(target=ingest, daemon=True)
-        t.start()
-        rt = threading.Thread(target=responder, daemon=True)
-        rt.start()
-        try:
-            while not self._stop.is_set():
+        def gpu_loop(ex, stream):
+            import torch
+            while True:
                 item = q.get()
                 if item is None:
-                    break
+                    q.put(None)  # let sibling workers exit too
+                    return
                 batch, bodies, decoded, handle, ph = item
                 if decoded is not None:
                     try:
                         t0 = _t.perf_counter()
-                        if handle is not None:
+                        if handle is not None and ex is not None:
+                            bt, ev, n = handle
+                            with torch.inference_mode(), \
+                                    torch.cuda.stream(stream):
+                                stream.wait_event(ev)
+                                out = ex.run(
+                                    {self.cfg.input_name: bt})[0]
+                                probs = out.float().cpu().numpy()[:n]
+                        elif handle is not None:
                             probs = eng.infer_uploaded(handle)
                         else:
                             probs = eng.infer_decoded(decoded,
@@ -252,13 +258,57 @@ class FastServer:
                         respq.put((batch, probs, None))
                         continue
                     except Exception:
-                        results = self._infer_isolating(bodies)
+                        with self._fb_lock:
+                            results = self._infer_isolating(bodies)
                     finally:
                         free_q.put(ph)  # forward synced: half reusable
                 else:
-                    results = self._infer_isolating(bodies)
+                    with self._fb_lock:
+                        results = self._infer_isolating(bodies)
                 respq.put((batch, None, results))
+
+        t = threading.Thread(target=ingest, daemon=True)
+        t.start()
+        rt = threading.Thread(target=responder, daemon=True)
+        rt.start()
+        try:
+            if n_gpu > 1:
+                import torch
+
+                from ..engine.gpu_executor import GpuExecutor
+                workers = []
+                for g in range(n_gpu):
+                    if g == 0:
+                        ex = eng.ex
+                    else:
+                        ex = GpuExecutor(eng.graph, device=eng.device,
+                                         cfg=eng.cfg)
+                        # pre-compile/capture the bucket plans so the
+                        # first live batches don't pay graph capture
+                        b = 1
+                        cfgv = self.cfg
+                        while True:
+                            z = torch.zeros(
+                                (b, cfgv.input_height, cfgv.input_width,
+                                 cfgv.input_channels))
+                            with torch.inference_mode():
+                                for _ in range(2):  # warm + capture
+                                    ex.run({cfgv.input_name: z})
+                            if b >= cfgv.max_batch:
+                                break
+                            b = min(b * 2, cfgv.max_batch)
+                    st = torch.cuda.Stream()
+                    th = threading.Thread(target=gpu_loop,
+                                          args=(ex, st), daemon=True)
+                    th.start()
+                    workers.append(th)
+                while (not self._stop.is_set()
+                       and any(w.is_alive() for w in workers)):
+                    self._stop.wait(0.5)
+            else:
+                gpu_loop(None, None)
         finally:
+            q.put(None)
             respq.put(None)
             if stats is not None and stats["batches"]:
                 import sys

@@ -130,3 +130,44 @@ def test_stats_and_stop():
     srv.stop()
     t.join(timeout=10)
     assert not t.is_alive()
+
+
+@pytest.mark.gpu
+def test_fast_server_gpu_end_to_end():
+    """Real GpuServingEngine through the whole fast stack (C++ frontend,
+    shm-ring decode incl the GPU-jpeg path, dual-stream pipeline):
+    responses must match in-process infer_blobs on the same JPEGs."""
+    import torch
+    if not torch.cuda.is_available():
+        pytest.skip("needs GPU")
+    from tensorflow_web_deploy_amd.engine.service import GpuServingEngine
+    from tensorflow_web_deploy_amd.server.fast import FastServer
+
+    from tensorflow_web_deploy_amd.conf.config import Config as Cfg
+    cfg = Cfg(model_name="lenet", device="cuda", host="127.0.0.1",
+              input_height=28, input_width=28, input_channels=1,
+              num_classes=10, max_batch=16, max_wait_ms=3)
+    eng = GpuServingEngine(cfg)
+    srv = FastServer(cfg, engine=eng, workers=2, port=0)
+    srv.start_background()
+    try:
+        imgs = [_png_bytes() for _ in range(12)]
+        ref = eng.infer_blobs(imgs)  # in-process reference
+        got = {}
+
+        def post(i):
+            status, j = _post(srv.port, imgs[i])
+            assert status == 200
+            got[i] = j["predictions"][0]["class_id"]
+
+        ts = [threading.Thread(target=post, args=(i,))
+              for i in range(len(imgs))]
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join(timeout=60)
+        assert len(got) == len(imgs)
+        for i in range(len(imgs)):
+            assert got[i] == int(ref[i].argmax()), i
+    finally:
+        srv.stop()
