@@ -171,6 +171,16 @@ class FastServer:
         # from forward (default stream, this thread) overlap the H2D +
         # preprocess of batch N+1 with the forward of batch N
         can_upload = hasattr(eng, "upload_decoded")
+        # dual GPU streams (opt-in TWD_GPU_STREAMS=2): a second executor
+        # (own weights + HIP graphs + torch stream) runs alternate
+        # batches so forward N+1 overlaps batch N's D2H/epilogue and
+        # inter-launch Python. Off by default pending live-load
+        # validation (the first rollout crashed on an unrelated NameError
+        # and burned the round's A/B budget — see NOTES_ROUND2.md).
+        n_gpu = 1
+        if can_upload and hasattr(eng, "ex"):
+            n_gpu = int(os.environ.get("TWD_GPU_STREAMS", "1"))
+        self._fb_lock = threading.Lock()
 
         def ingest():
             while not self._stop.is_set():

@@ -171,3 +171,37 @@ def test_fast_server_gpu_end_to_end():
             assert got[i] == int(ref[i].argmax()), i
     finally:
         srv.stop()
+
+
+def test_pipelined_loop_cpu_via_replicas():
+    """_serve_pipelined on CPU: ReplicatedEngine(gloo, world=1,
+    serve_blobs=True) has infer_decoded + pre.decode_shm, so the full
+    three-stage pipeline (incl the n_gpu dispatch block) runs without
+    a GPU — this is the test that would have caught the NameError that
+    killed the first dual-stream rollout."""
+    from tensorflow_web_deploy_amd.parallel.replicas import ReplicatedEngine
+    from tensorflow_web_deploy_amd.server.fast import FastServer
+    cfg = Config(model_name="lenet", device="cpu", host="127.0.0.1",
+                 input_height=28, input_width=28, input_channels=1,
+                 num_classes=10, max_batch=8, max_wait_ms=3)
+    eng = ReplicatedEngine(cfg, world=1, backend="gloo",
+                           serve_blobs=True, start_timeout=120.0)
+    srv = FastServer(cfg, engine=eng, workers=2, port=0)
+    srv.start_background()
+    try:
+        oks = []
+
+        def post():
+            status, j = _post(srv.port, _png_bytes(), timeout=30)
+            assert status == 200 and len(j["predictions"]) == 5
+            oks.append(1)
+
+        ts = [threading.Thread(target=post) for _ in range(10)]
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join(timeout=60)
+        assert len(oks) == 10
+    finally:
+        srv.stop()
+        eng.shutdown()
