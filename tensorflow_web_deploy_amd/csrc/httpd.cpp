@@ -45,6 +45,7 @@
 #include <unordered_map>
 #include <vector>
 
+#include <pybind11/numpy.h>
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
@@ -221,6 +222,58 @@ class Frontend {
       if (w >= nworkers_) continue;
       byw[w].push_back(
           Resp{id, std::get<1>(t), std::string(std::get<2>(t))});
+    }
+    py::gil_scoped_release nogil;
+    for (int w = 0; w < nworkers_; ++w) {
+      if (byw[w].empty()) continue;
+      {
+        std::lock_guard<std::mutex> lk(ws_[w]->mu);
+        for (auto& r : byw[w]) ws_[w]->responses.push_back(std::move(r));
+      }
+      ding(*ws_[w]);
+    }
+  }
+
+  void set_labels(const std::vector<std::string>& labels,
+                  const std::string& model) {
+    labels_ = labels;
+    model_ = model;
+  }
+
+  // Build the /predict JSON for a whole batch in C++ (the Python
+  // responder's per-row json.dumps measured ~1.4 ms per 64-batch of
+  // pure-GIL time) and enqueue the responses. topk/probs: [n][k].
+  void respond_topk(const std::vector<uint64_t>& ids,
+                    py::array_t<int32_t> topk,
+                    py::array_t<float> probs) {
+    auto ti = topk.unchecked<2>();
+    auto pi = probs.unchecked<2>();
+    const size_t n = ids.size();
+    const py::ssize_t k = ti.shape(1);
+    std::vector<Resp> byw[16];
+    for (size_t i = 0; i < n; ++i) {
+      std::string body;
+      body.reserve(64 + 64 * k);
+      body += "{\"predictions\":[";
+      for (py::ssize_t j = 0; j < k; ++j) {
+        int cls = ti(i, j);
+        if (j) body += ',';
+        body += "{\"label\":\"";
+        if (cls >= 0 && (size_t)cls < labels_.size())
+          json_escape(labels_[cls], body);
+        body += "\",\"class_id\":";
+        body += std::to_string(cls);
+        char buf[32];
+        snprintf(buf, sizeof(buf), ",\"prob\":%.6g}", (double)pi(i, j));
+        body += buf;
+      }
+      body += "],\"model\":\"";
+      json_escape(model_, body);
+      body += "\"}";
+      uint64_t id = ids[i];
+      int w = (int)(id >> 60);
+      if (w >= nworkers_) continue;
+      byw[w].push_back(Resp{id, 200, std::move(body)});
     }
     py::gil_scoped_release nogil;
     for (int w = 0; w < nworkers_; ++w) {
@@ -551,6 +604,23 @@ class Frontend {
 
   std::atomic<uint64_t> accepted_{0}, requests_{0}, responded_{0},
       rejected_{0}, dropped_{0};
+  std::vector<std::string> labels_;
+  std::string model_;
+
+  static void json_escape(const std::string& in, std::string& out) {
+    for (char c : in) {
+      if (c == '"' || c == '\\') {
+        out += '\\';
+        out += c;
+      } else if ((unsigned char)c < 0x20) {
+        char buf[8];
+        snprintf(buf, sizeof(buf), "\\u%04x", c);
+        out += buf;
+      } else {
+        out += c;
+      }
+    }
+  }
 };
 
 }  // namespace
@@ -565,6 +635,8 @@ PYBIND11_MODULE(_httpd, m) {
       .def("wait_batch", &Frontend::wait_batch, py::arg("max_batch"),
            py::arg("max_wait_ms"))
       .def("respond_many", &Frontend::respond_many)
+      .def("set_labels", &Frontend::set_labels)
+      .def("respond_topk", &Frontend::respond_topk)
       .def("stats", &Frontend::stats)
       .def("stop", &Frontend::stop);
 }

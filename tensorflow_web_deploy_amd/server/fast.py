@@ -38,6 +38,10 @@ class FastServer:
         self.port = self.fe.port()
         self._stop = threading.Event()
         self.model_name = cfg.model_name or cfg.model_path
+        # C-side /predict JSON: the frontend holds labels + model name
+        # and formats whole batches from (ids, topk, probs) arrays
+        self.fe.set_labels([str(x) for x in self.labels],
+                           str(self.model_name))
 
     # -- one batch ----------------------------------------------------
     def _infer(self, bodies):
@@ -82,8 +86,9 @@ class FastServer:
                     out[i] = (500, json.dumps({"error": str(e)}))
         return out
 
-    def _payloads(self, probs) -> list:
-        """Vectorized top-k for a whole (n, classes) batch."""
+    def _topk(self, probs):
+        """Vectorized top-k for a whole (n, classes) batch ->
+        (idx int32 [n,k], vals f32 [n,k])."""
         n, nc = probs.shape
         k = min(self.cfg.top_k, nc)
         if k < nc:
@@ -92,20 +97,10 @@ class FastServer:
             order = np.argsort(vals, axis=1)[:, ::-1]
             top = np.take_along_axis(part, order, axis=1)
         else:
-            top = np.argsort(probs, axis=1)[:, ::-1]
-        out = []
-        labels = self.labels
-        name = self.model_name
-        for r in range(n):
-            pr = probs[r]
-            out.append(json.dumps({
-                "predictions": [
-                    {"label": labels[int(i)], "class_id": int(i),
-                     "prob": float(pr[int(i)])} for i in top[r]
-                ],
-                "model": name,
-            }))
-        return out
+            top = np.argsort(probs, axis=1)[:, ::-1][:, :k]
+        vals = np.take_along_axis(probs, top, axis=1)
+        return (np.ascontiguousarray(top, dtype=np.int32),
+                np.ascontiguousarray(vals, dtype=np.float32))
 
     def _payload(self, probs) -> str:
         k = min(self.cfg.top_k, len(probs))
@@ -228,13 +223,17 @@ class FastServer:
                 batch, probs, results = item
                 t0 = _t.perf_counter()
                 if results is None:
-                    payloads = self._payloads(probs)
-                    results = [(200, pl) for pl in payloads]
-                t1 = _t.perf_counter()
-                self.fe.respond_many([
-                    (rid, st, body.encode())
-                    for (rid, _), (st, body) in zip(batch, results)
-                ])
+                    # top-k in numpy, JSON + enqueue entirely in C++
+                    top, vals = self._topk(probs)
+                    t1 = _t.perf_counter()
+                    self.fe.respond_topk([rid for (rid, _) in batch],
+                                         top, vals)
+                else:
+                    t1 = _t.perf_counter()
+                    self.fe.respond_many([
+                        (rid, st, body.encode())
+                        for (rid, _), (st, body) in zip(batch, results)
+                    ])
                 if stats is not None:
                     stats["json"] += t1 - t0
                     stats["respond"] += _t.perf_counter() - t1

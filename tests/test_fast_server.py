@@ -194,6 +194,13 @@ def test_pipelined_loop_cpu_via_replicas():
         def post():
             status, j = _post(srv.port, _png_bytes(), timeout=30)
             assert status == 200 and len(j["predictions"]) == 5
+            # this path responds through the C++ respond_topk JSON
+            assert set(j) == {"predictions", "model"}
+            for pr in j["predictions"]:
+                assert set(pr) == {"label", "class_id", "prob"}
+                assert isinstance(pr["class_id"], int)
+            probs = [pr["prob"] for pr in j["predictions"]]
+            assert probs == sorted(probs, reverse=True)
             oks.append(1)
 
         ts = [threading.Thread(target=post) for _ in range(10)]
@@ -205,3 +212,18 @@ def test_pipelined_loop_cpu_via_replicas():
     finally:
         srv.stop()
         eng.shutdown()
+
+
+def test_c_json_schema_matches_python(server):
+    """respond_topk's C-formatted JSON must carry the same schema and
+    ordering as the Flask/Python payloads (label escaping included)."""
+    status, j = _post(server.port, _png_bytes())
+    assert status == 200
+    assert set(j) == {"predictions", "model"}
+    for p in j["predictions"]:
+        assert set(p) == {"label", "class_id", "prob"}
+        assert isinstance(p["label"], str)
+        assert isinstance(p["class_id"], int)
+        assert 0.0 <= p["prob"] <= 1.0
+    probs = [p["prob"] for p in j["predictions"]]
+    assert probs == sorted(probs, reverse=True)
