@@ -387,6 +387,68 @@ def _shape(node, ins, consts):
     return torch.tensor(list(ins[0].shape), dtype=torch.int32)
 
 
+@cpu_op("Maximum")
+def _maximum(node, ins, consts):
+    return torch.maximum(ins[0], ins[1])
+
+
+@cpu_op("Minimum")
+def _minimum(node, ins, consts):
+    return torch.minimum(ins[0], ins[1])
+
+
+@cpu_op("LeakyRelu")
+def _leaky_relu(node, ins, consts):
+    return F.leaky_relu(ins[0], float(node.attrs.get("alpha", 0.2)))
+
+
+@cpu_op("Elu")
+def _elu(node, ins, consts):
+    return F.elu(ins[0])
+
+
+@cpu_op("Softplus")
+def _softplus(node, ins, consts):
+    return F.softplus(ins[0])
+
+
+@cpu_op("Square")
+def _square(node, ins, consts):
+    return ins[0] * ins[0]
+
+
+@cpu_op("Neg")
+def _neg(node, ins, consts):
+    return -ins[0]
+
+
+@cpu_op("Abs")
+def _abs(node, ins, consts):
+    return torch.abs(ins[0])
+
+
+@cpu_op("Log")
+def _log(node, ins, consts):
+    return torch.log(ins[0])
+
+
+@cpu_op("Slice")
+def _slice(node, ins, consts):
+    # inputs: x, begin, size (size -1 = to end), both const or computed
+    x = ins[0]
+    begin = [int(v) for v in ins[1].tolist()]
+    size = [int(v) for v in ins[2].tolist()]
+    idx = tuple(slice(b, x.shape[d] if sz == -1 else b + sz)
+                for d, (b, sz) in enumerate(zip(begin, size)))
+    return x[idx]
+
+
+@cpu_op("Fill")
+def _fill(node, ins, consts):
+    shape = [int(v) for v in ins[0].tolist()]
+    return torch.full(shape, float(ins[1].reshape(-1)[0].item()))
+
+
 @cpu_op("Cast")
 def _cast(node, ins, consts):
     from ..ckpt import graphdef as g

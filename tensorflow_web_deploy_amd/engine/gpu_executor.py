@@ -662,6 +662,32 @@ class GpuExecutor:
                 srcs = [bufs[i] for i in ins]
                 steps.append(lambda Y=Y, srcs=srcs, axis=axis:
                              torch.cat(srcs, dim=axis, out=Y))
+            elif op in ("Maximum", "Minimum", "LeakyRelu", "Elu",
+                        "Softplus", "Square", "Neg", "Abs", "Log"):
+                # rare glue ops from real TF exports: torch eager on the
+                # bf16 buffers (capture-safe; none appear in the
+                # built-in model families' hot paths)
+                xn = node.inputs[0]
+                Y = alloc(node.name, shapes[xn])
+                if op in ("Maximum", "Minimum"):
+                    other = node.inputs[1]
+                    fn = torch.maximum if op == "Maximum" else torch.minimum
+                    steps.append(lambda X=bufs[xn], O=bufs[other], Y=Y,
+                                 fn=fn: Y.copy_(fn(X, O)))
+                else:
+                    import torch.nn.functional as _F
+                    alpha = float(node.attrs.get("alpha", 0.2))
+                    unary = {
+                        "LeakyRelu": lambda v: _F.leaky_relu(v, alpha),
+                        "Elu": _F.elu,
+                        "Softplus": _F.softplus,
+                        "Square": lambda v: v * v,
+                        "Neg": torch.neg,
+                        "Abs": torch.abs,
+                        "Log": torch.log,
+                    }[op]
+                    steps.append(lambda X=bufs[xn], Y=Y, u=unary:
+                                 Y.copy_(u(X.float()).to(Y.dtype)))
             else:
                 raise NotImplementedError(
                     f"GPU path: op {op} (node {node.name})")

@@ -66,3 +66,32 @@ def test_same_padding_asymmetric():
     assert p.shape == (1, 1, 5, 5)
     assert p[0, 0, 4].sum() == 0 and p[0, 0, :, 4].sum() == 0  # pad at end
     assert p[0, 0, 0, 0] == 0.0 and p[0, 0, 0, 1] == 1.0  # no pad at start
+
+
+def test_glue_ops_cpu_reference():
+    """Rare glue ops from real TF exports run through the CPU executor
+    with TF semantics (Slice with size=-1, Fill, activations)."""
+    import torch
+
+    from tensorflow_web_deploy_amd.engine.executor import _CPU_OPS as CPU_OPS
+    from tensorflow_web_deploy_amd.engine.ir import Node
+
+    x = torch.tensor([[-2.0, -0.5, 0.0, 1.5]])
+
+    def run(op, ins, **attrs):
+        return CPU_OPS[op](Node("n", op, [], attrs), ins, {})
+
+    assert torch.equal(run("Maximum", [x, torch.zeros_like(x)]),
+                       torch.clamp(x, min=0))
+    assert torch.equal(run("Minimum", [x, torch.zeros_like(x)]),
+                       torch.clamp(x, max=0))
+    lr = run("LeakyRelu", [x], alpha=0.1)
+    assert torch.allclose(lr, torch.nn.functional.leaky_relu(x, 0.1))
+    assert torch.allclose(run("Square", [x]), x * x)
+    assert torch.allclose(run("Abs", [x]), x.abs())
+    y = torch.arange(24.0).reshape(2, 3, 4)
+    s = run("Slice", [y, torch.tensor([0, 1, 2]),
+                      torch.tensor([2, -1, 2])])
+    assert torch.equal(s, y[0:2, 1:, 2:4])
+    f = run("Fill", [torch.tensor([2, 2]), torch.tensor(3.5)])
+    assert torch.equal(f, torch.full((2, 2), 3.5))
