@@ -449,6 +449,25 @@ def _fill(node, ins, consts):
     return torch.full(shape, float(ins[1].reshape(-1)[0].item()))
 
 
+@cpu_op("ResizeBilinear")
+def _resize_bilinear(node, ins, consts):
+    # TF semantics: align_corners / half_pixel_centers attrs
+    x = ins[0].permute(0, 3, 1, 2)
+    size = [int(v) for v in ins[1].tolist()]
+    ac = bool(node.attrs.get("align_corners"))
+    y = F.interpolate(x, size=tuple(size), mode="bilinear",
+                      align_corners=ac)
+    return y.permute(0, 2, 3, 1)
+
+
+@cpu_op("ResizeNearestNeighbor")
+def _resize_nearest(node, ins, consts):
+    x = ins[0].permute(0, 3, 1, 2)
+    size = [int(v) for v in ins[1].tolist()]
+    y = F.interpolate(x, size=tuple(size), mode="nearest")
+    return y.permute(0, 2, 3, 1)
+
+
 @cpu_op("Cast")
 def _cast(node, ins, consts):
     from ..ckpt import graphdef as g

@@ -688,6 +688,26 @@ class GpuExecutor:
                     }[op]
                     steps.append(lambda X=bufs[xn], Y=Y, u=unary:
                                  Y.copy_(u(X.float()).to(Y.dtype)))
+            elif op in ("ResizeBilinear", "ResizeNearestNeighbor"):
+                xn = node.inputs[0]
+                N_, H, W, C = shapes[xn]
+                size = [int(v) for v in
+                        self.consts[node.inputs[1]].reshape(-1)]
+                Y = alloc(node.name, (N_, size[0], size[1], C))
+                ac = bool(node.attrs.get("align_corners"))
+                mode = ("bilinear" if op == "ResizeBilinear"
+                        else "nearest")
+                import torch.nn.functional as _F
+
+                def _rs(X=bufs[xn], Y=Y, size=tuple(size), mode=mode,
+                        ac=ac):
+                    x4 = X.float().permute(0, 3, 1, 2)
+                    r = _F.interpolate(
+                        x4, size=size, mode=mode,
+                        align_corners=(ac if mode == "bilinear"
+                                       else None))
+                    Y.copy_(r.permute(0, 2, 3, 1).to(Y.dtype))
+                steps.append(_rs)
             else:
                 raise NotImplementedError(
                     f"GPU path: op {op} (node {node.name})")

@@ -95,3 +95,19 @@ def test_glue_ops_cpu_reference():
     assert torch.equal(s, y[0:2, 1:, 2:4])
     f = run("Fill", [torch.tensor([2, 2]), torch.tensor(3.5)])
     assert torch.equal(f, torch.full((2, 2), 3.5))
+
+
+def test_resize_bilinear_cpu():
+    import torch
+
+    from tensorflow_web_deploy_amd.engine.executor import _CPU_OPS
+    from tensorflow_web_deploy_amd.engine.ir import Node
+    x = torch.rand(1, 6, 8, 3)
+    out = _CPU_OPS["ResizeBilinear"](
+        Node("r", "ResizeBilinear", [], {"align_corners": False}),
+        [x, torch.tensor([12, 16])], {})
+    assert out.shape == (1, 12, 16, 3)
+    ref = torch.nn.functional.interpolate(
+        x.permute(0, 3, 1, 2), size=(12, 16), mode="bilinear",
+        align_corners=False).permute(0, 2, 3, 1)
+    assert torch.allclose(out, ref)
