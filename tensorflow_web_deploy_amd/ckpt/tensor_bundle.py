@@ -8,7 +8,7 @@ Index = SSTable (ckpt/sstable.py) mapping:
 Data shards are raw little-endian tensor bytes at [offset, offset+size).
 
 The writer produces single-shard bundles for fixtures and for our own
-SavedModel export (tools/export_savedmodel.py).
+SavedModel export (`tools/make_model.py --savedmodel`).
 """
 from __future__ import annotations
 
