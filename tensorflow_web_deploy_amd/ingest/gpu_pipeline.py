@@ -258,6 +258,10 @@ class ShmDecodePool:
             self._procs.append(pr)
         self._dead = False
         self._warm = False
+        # /dev/shm files outlive the process: without this a crashed or
+        # un-close()d server leaks a ~200-300 MB mapping per run
+        import atexit
+        atexit.register(self.close)
 
     def warm(self, timeout: float = 60.0):
         if self._warm:
@@ -342,6 +346,9 @@ class ShmDecodePool:
 
     def close(self):
         import os
+        if getattr(self, "_closed", False):
+            return
+        self._closed = True
         for w in self._wake:
             try:
                 w.send_bytes(b"q")
