@@ -59,6 +59,7 @@ constexpr int kMaxEvents = 128;
 struct Conn {
   int fd = -1;
   uint32_t gen = 0;
+  int64_t last_ms = 0;  // last read/write activity (idle sweep)
   std::string in;
   std::string out;
   size_t out_off = 0;
@@ -119,9 +120,10 @@ bool iequal(const char* a, const char* b, size_t n) {
 
 class Frontend {
  public:
-  Frontend(const std::string& host, int port, int workers, int backlog)
+  Frontend(const std::string& host, int port, int workers, int backlog,
+           int idle_timeout_s = 120)
       : nworkers_(workers < 1 ? 1 : (workers > 15 ? 15 : workers)),
-        backlog_(backlog) {
+        backlog_(backlog), idle_timeout_ms_(idle_timeout_s * 1000LL) {
     listen_fd_ = ::socket(AF_INET, SOCK_STREAM | SOCK_NONBLOCK, 0);
     if (listen_fd_ < 0) throw std::runtime_error("socket() failed");
     int one = 1;
@@ -473,11 +475,25 @@ class Frontend {
   void worker_loop(int widx) {
     Worker& w = *ws_[widx];
     epoll_event evs[kMaxEvents];
+    int64_t next_sweep = now_ms() + 5000;
     while (running_.load()) {
       int n = ::epoll_wait(w.epfd, evs, kMaxEvents, 200);
       if (n < 0) {
         if (errno == EINTR) continue;
         break;
+      }
+      const int64_t now = now_ms();
+      if (now >= next_sweep) {
+        // reap connections idle past the timeout UNLESS a request is
+        // in flight (its response is still coming from Python)
+        next_sweep = now + 5000;
+        std::vector<int> dead;
+        for (auto& kv : w.conns) {
+          Conn& c = *kv.second;
+          if (!c.awaiting && now - c.last_ms > idle_timeout_ms_)
+            dead.push_back(kv.first);
+        }
+        for (int fd : dead) close_conn(w, fd);
       }
       for (int i = 0; i < n; ++i) {
         int fd = evs[i].data.fd;
@@ -498,6 +514,7 @@ class Frontend {
             continue;
           }
           bool alive = true;
+          c.last_ms = now;
           if (e & EPOLLIN) {
             char buf[64 * 1024];
             for (;;) {
@@ -566,6 +583,7 @@ class Frontend {
       auto c = std::make_unique<Conn>();
       c->fd = fd;
       c->gen = w.gen_ctr++;
+      c->last_ms = now_ms();
       epoll_event ev{};
       ev.events = EPOLLIN | EPOLLRDHUP;
       ev.data.fd = fd;
@@ -591,10 +609,17 @@ class Frontend {
     }
   }
 
+  static int64_t now_ms() {
+    return std::chrono::duration_cast<std::chrono::milliseconds>(
+               std::chrono::steady_clock::now().time_since_epoch())
+        .count();
+  }
+
   int listen_fd_ = -1;
   int port_ = 0;
   int nworkers_;
   int backlog_;
+  int64_t idle_timeout_ms_;
   std::vector<std::unique_ptr<Worker>> ws_;
   std::atomic<bool> running_{false};
 
@@ -628,9 +653,9 @@ class Frontend {
 PYBIND11_MODULE(_httpd, m) {
   m.doc() = "epoll HTTP frontend: request handling without the GIL";
   py::class_<Frontend>(m, "Frontend")
-      .def(py::init<const std::string&, int, int, int>(), py::arg("host"),
-           py::arg("port"), py::arg("workers") = 3,
-           py::arg("backlog") = 8192)
+      .def(py::init<const std::string&, int, int, int, int>(),
+           py::arg("host"), py::arg("port"), py::arg("workers") = 3,
+           py::arg("backlog") = 8192, py::arg("idle_timeout_s") = 120)
       .def("port", &Frontend::port)
       .def("wait_batch", &Frontend::wait_batch, py::arg("max_batch"),
            py::arg("max_wait_ms"))

@@ -227,3 +227,34 @@ def test_c_json_schema_matches_python(server):
         assert 0.0 <= p["prob"] <= 1.0
     probs = [p["prob"] for p in j["predictions"]]
     assert probs == sorted(probs, reverse=True)
+
+
+def test_idle_connection_reaped():
+    """A connection that sends nothing is closed by the idle sweep; an
+    active request is never reaped mid-flight."""
+    import socket
+    import time
+
+    from tensorflow_web_deploy_amd.server.fast import FastServer
+    from tensorflow_web_deploy_amd._httpd import Frontend
+    cfg = Config(model_name="lenet", device="cpu", host="127.0.0.1",
+                 max_batch=4, max_wait_ms=2)
+    srv = FastServer(cfg, engine=FakeEngine(10), workers=1, port=0)
+    # swap in a frontend with a 1 s idle timeout for the test
+    srv.fe.stop()
+    srv.fe = Frontend("127.0.0.1", 0, 1, 1024, idle_timeout_s=1)
+    srv.fe.set_labels([str(i) for i in range(10)], "lenet")
+    srv.port = srv.fe.port()
+    srv.start_background()
+    try:
+        s = socket.create_connection(("127.0.0.1", srv.port), timeout=10)
+        time.sleep(7)  # > timeout + sweep period
+        # the server should have closed us: recv returns EOF
+        s.settimeout(3)
+        assert s.recv(64) == b""
+        s.close()
+        # normal requests still work afterwards
+        status, j = _post(srv.port, _png_bytes())
+        assert status == 200 and len(j["predictions"]) == 5
+    finally:
+        srv.stop()
