@@ -449,6 +449,35 @@ def _fill(node, ins, consts):
     return torch.full(shape, float(ins[1].reshape(-1)[0].item()))
 
 
+@cpu_op("Split")
+def _split(node, ins, consts):
+    # inputs: axis (scalar const), value; num_split attr
+    axis = int(ins[0].reshape(-1)[0].item())
+    n = int(node.attrs.get("num_split", 1))
+    return tuple(torch.chunk(ins[1], n, dim=axis))
+
+
+@cpu_op("SplitV")
+def _splitv(node, ins, consts):
+    # inputs: value, size_splits, axis
+    sizes = [int(v) for v in ins[1].tolist()]
+    axis = int(ins[2].reshape(-1)[0].item())
+    return tuple(torch.split(ins[0], sizes, dim=axis))
+
+
+@cpu_op("Unpack")
+def _unpack(node, ins, consts):
+    axis = int(node.attrs.get("axis", 0))
+    return tuple(t.squeeze(axis) for t in
+                 torch.split(ins[0], 1, dim=axis))
+
+
+@cpu_op("Pack")
+def _pack(node, ins, consts):
+    axis = int(node.attrs.get("axis", 0))
+    return torch.stack(list(ins), dim=axis)
+
+
 @cpu_op("ResizeBilinear")
 def _resize_bilinear(node, ins, consts):
     # TF semantics: align_corners / half_pixel_centers attrs
@@ -512,5 +541,12 @@ class Executor:
             fn = self.ops.get(node.op)
             if fn is None:
                 raise NotImplementedError(f"op {node.op} (node {node.name})")
-            env[node.name] = fn(node, ins, self.consts)
+            out = fn(node, ins, self.consts)
+            if isinstance(out, (tuple, list)):
+                # multi-output op (Split/Unstack): TF refs are name:k
+                env[node.name] = out[0]
+                for k, o in enumerate(out):
+                    env[f"{node.name}:{k}"] = o
+            else:
+                env[node.name] = out
         return [env[o] for o in outputs]

@@ -111,3 +111,26 @@ def test_resize_bilinear_cpu():
         x.permute(0, 3, 1, 2), size=(12, 16), mode="bilinear",
         align_corners=False).permute(0, 2, 3, 1)
     assert torch.allclose(out, ref)
+
+
+def test_multi_output_split_through_executor():
+    """Split/Unpack produce name:k tensors that downstream nodes can
+    reference (real exported graphs use this)."""
+    import numpy as np
+    import torch
+
+    from tensorflow_web_deploy_amd.engine.executor import Executor
+    from tensorflow_web_deploy_amd.engine.ir import Graph, Node
+    consts = {"axis": np.array(1, np.int32)}
+    nodes = [
+        Node("x", "Placeholder", [], {}),
+        Node("sp", "Split", ["axis", "x"], {"num_split": 2}),
+        Node("sum", "Add", ["sp", "sp:1"], {}),  # sp == sp:0
+        Node("packed", "Pack", ["sp:0", "sp:1"], {"axis": 0}),
+    ]
+    g = Graph(nodes, consts, ["x"], ["sum", "packed"])
+    ex = Executor(g)
+    x = torch.arange(8.0).reshape(2, 4)
+    out, packed = ex.run({"x": x})
+    assert torch.equal(out, x[:, :2] + x[:, 2:])
+    assert packed.shape == (2, 2, 2)
