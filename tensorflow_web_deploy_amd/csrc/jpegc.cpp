@@ -203,12 +203,17 @@ py::dict jpeg_parse(py::bytes data_b) {
       }
       if (m == 0xD9) break;  // EOI
       int len = rd16(pos + 2);
+      if (len < 2 || pos + 2 + (size_t)len > n)
+        throw std::invalid_argument("truncated segment");
       size_t seg = pos + 4;
       if (m == 0xDB) {  // DQT
         size_t p = seg;
-        while (p < pos + 2 + len) {
+        const size_t segend = pos + 2 + len;
+        while (p < segend) {
           int pq = d[p] >> 4, tq = d[p] & 15;
           ++p;
+          if (p + (pq ? 128 : 64) > segend)
+            throw std::invalid_argument("truncated DQT");
           for (int i = 0; i < 64; ++i) {
             qt[tq][kZigzag[i]] = pq ? rd16(p + 2 * i) : d[p + i];
           }
@@ -217,9 +222,12 @@ py::dict jpeg_parse(py::bytes data_b) {
         }
       } else if (m == 0xC4) {  // DHT
         size_t p = seg;
-        while (p < pos + 2 + len) {
+        const size_t segend = pos + 2 + len;
+        while (p < segend) {
           int tc = d[p] >> 4, th = d[p] & 15;
           ++p;
+          if (p + 16 > segend)
+            throw std::invalid_argument("truncated DHT");
           uint8_t counts[16];
           int nsym = 0;
           for (int i = 0; i < 16; ++i) {
@@ -227,16 +235,23 @@ py::dict jpeg_parse(py::bytes data_b) {
             nsym += counts[i];
           }
           p += 16;
+          if (nsym > 256 || p + nsym > segend)
+            throw std::invalid_argument("bad DHT symbol count");
           (tc ? hac[th] : hdc[th]).build(counts, d + p, nsym);
           p += nsym;
         }
       } else if (m == 0xC0 || m == 0xC1) {  // SOF0/1 baseline
+        if (len < 8) throw std::invalid_argument("truncated SOF");
         if (d[seg] != 8) throw std::invalid_argument("not 8-bit");
         H = rd16(seg + 1);
         W = rd16(seg + 3);
         int nc = d[seg + 5];
         if (nc != 1 && nc != 3)
           throw std::invalid_argument("1- or 3-component only");
+        if (len < 8 + 3 * nc)
+          throw std::invalid_argument("truncated SOF comps");
+        if (W <= 0 || H <= 0 || (int64_t)W * H > (64 << 20))
+          throw std::invalid_argument("unreasonable dimensions");
         comps.resize(nc);
         for (int c = 0; c < nc; ++c) {
           comps[c].id = d[seg + 6 + 3 * c];
@@ -257,6 +272,8 @@ py::dict jpeg_parse(py::bytes data_b) {
       } else if (m == 0xDA) {  // SOS
         if (!sof_seen) throw std::invalid_argument("SOS before SOF");
         int ns = d[seg];
+        if (len < 6 + 2 * ns)
+          throw std::invalid_argument("truncated SOS");
         for (int s2 = 0; s2 < ns; ++s2) {
           int cid = d[seg + 1 + 2 * s2];
           int tt = d[seg + 2 + 2 * s2];
@@ -300,6 +317,8 @@ py::dict jpeg_parse(py::bytes data_b) {
         }
         ++mcu_count;
         for (auto& c : comps) {
+          if (!hdc[c.td].present || !hac[c.ta].present)
+            throw std::invalid_argument("missing huffman table");
           for (int by = 0; by < c.v; ++by) {
             for (int bx = 0; bx < c.h; ++bx) {
               int16_t* blk =
