@@ -21,9 +21,12 @@
 //     and the ds_read address — the LDS image itself stays linear per
 //     instruction (rule 21 generalized).
 //
-// v1 scope: MODE=linear A (dense/1x1), bias + act + optional residual
-// in a scattered epilogue; routed for large square-ish shapes behind
-// TWD_8P=1 until it beats gemm_conv_kernel_256 (then dispatch flips).
+// Scope: MODE=linear A (dense/1x1), bias + act + optional residual,
+// LDS-restage epilogue. UNROUTED: measured faster than
+// gemm_conv_kernel_256 only at 8192^3 (1006 vs 954 TF) and slower at
+// 4096^3 (875 vs 1130) and on model shapes — NOTES_ROUND2.md item 2
+// records the five-variant exploration. bench/gemm8p_bench.py
+// reproduces numerics + perf via the gemm8p_bf16 binding.
 #include <hip/hip_bf16.h>
 #include <hip/hip_runtime.h>
 

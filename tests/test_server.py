@@ -112,3 +112,19 @@ def test_debug_trace_endpoint():
         assert j["config"]["max_batch"] == 8
     finally:
         batcher.stop()
+
+
+def test_conf_presets_load():
+    """Every shipped YAML preset parses into a Config."""
+    import glob
+    import os
+
+    from tensorflow_web_deploy_amd.conf.config import Config, load_config
+    root = os.path.join(os.path.dirname(__file__), "..",
+                        "tensorflow_web_deploy_amd", "conf", "presets")
+    paths = sorted(glob.glob(os.path.join(root, "*.yaml")))
+    assert len(paths) == 5
+    for p in paths:
+        cfg = load_config(p)
+        assert isinstance(cfg, Config)
+        assert cfg.model_name or cfg.model_path
