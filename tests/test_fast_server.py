@@ -258,3 +258,70 @@ def test_idle_connection_reaped():
         assert status == 200 and len(j["predictions"]) == 5
     finally:
         srv.stop()
+
+
+def test_stress_mixed_traffic(server):
+    """Race-class stress: concurrent keep-alive and one-shot clients,
+    interleaved bad payloads and healthz probes; every request gets a
+    correct, well-formed answer and the frontend counters reconcile."""
+    import random
+
+    errors = []
+    ok = []
+    lock = threading.Lock()
+    rng = random.Random(7)
+
+    def keepalive_client(n):
+        conn = http.client.HTTPConnection("127.0.0.1", server.port,
+                                          timeout=15)
+        try:
+            for _ in range(n):
+                bad = rng.random() < 0.2
+                body = b"garbage-bytes" if bad else _png_bytes()
+                conn.request("POST", "/predict", body=body)
+                r = conn.getresponse()
+                data = r.read()
+                j = json.loads(data)
+                if bad:
+                    assert r.status == 400 and "error" in j, j
+                else:
+                    assert r.status == 200, j
+                    assert len(j["predictions"]) == 5
+                with lock:
+                    ok.append(1)
+        except Exception as e:  # pragma: no cover
+            with lock:
+                errors.append(repr(e))
+        finally:
+            conn.close()
+
+    def oneshot_client(n):
+        for _ in range(n):
+            try:
+                if rng.random() < 0.3:
+                    with urllib.request.urlopen(
+                            f"http://127.0.0.1:{server.port}/healthz",
+                            timeout=15) as r:
+                        assert json.loads(r.read())["status"] == "ok"
+                else:
+                    status, j = _post(server.port, _png_bytes(),
+                                      timeout=15)
+                    assert status == 200
+                with lock:
+                    ok.append(1)
+            except Exception as e:  # pragma: no cover
+                with lock:
+                    errors.append(repr(e))
+
+    ts = ([threading.Thread(target=keepalive_client, args=(8,))
+           for _ in range(6)] +
+          [threading.Thread(target=oneshot_client, args=(8,))
+           for _ in range(6)])
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join(timeout=60)
+    assert not errors, errors[:3]
+    assert len(ok) == 12 * 8
+    st = server.fe.stats()
+    assert st["responded"] + st["rejected_503"] >= 50
