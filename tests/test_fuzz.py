@@ -91,3 +91,41 @@ def test_jpeg_truncation_rejected(seed, cut_pct):
         _jpegc.parse(blob[:cut])
     except (ValueError, RuntimeError):
         pass
+
+
+@given(st.binary(max_size=300))
+@settings(max_examples=200)
+def test_graphdef_garbage_rejected(data):
+    try:
+        gd.parse_graphdef(data)
+    except Exception:
+        pass  # reject is fine; crash/hang is the failure
+
+
+@given(st.integers(min_value=0, max_value=99))
+def test_graphdef_roundtrip_random(seed):
+    rng = np.random.default_rng(seed)
+    g = gd.GraphDef()
+    nodes = []
+    for i in range(int(rng.integers(1, 6))):
+        n = gd.NodeDef(name=f"n{i}", op=rng.choice(
+            ["Const", "Conv2D", "Relu", "Weird/Op-Name_2"]))
+        if i:
+            n.input = [f"n{rng.integers(0, i)}"]
+        if rng.random() < 0.7:
+            arr = rng.standard_normal(
+                tuple(rng.integers(1, 4, size=2))).astype(np.float32)
+            n.attr["value"] = gd.AttrValue(tensor=arr)
+        n.attr["s"] = gd.AttrValue(s=bytes(rng.integers(
+            0, 256, size=int(rng.integers(0, 10))).astype(np.uint8)))
+        nodes.append(n)
+    g.node = nodes
+    g2 = gd.parse_graphdef(gd.encode_graphdef(g))
+    assert [n.name for n in g2.node] == [n.name for n in nodes]
+    assert [n.op for n in g2.node] == [n.op for n in nodes]
+    for a, b in zip(nodes, g2.node):
+        assert a.input == b.input
+        assert b.attr["s"].s == a.attr["s"].s
+        if "value" in a.attr:
+            np.testing.assert_array_equal(b.attr["value"].tensor,
+                                          a.attr["value"].tensor)
