@@ -89,6 +89,7 @@ const char* status_text(int s) {
     case 200: return "OK";
     case 400: return "Bad Request";
     case 404: return "Not Found";
+    case 411: return "Length Required";
     case 413: return "Payload Too Large";
     case 503: return "Service Unavailable";
     default: return "Error";
@@ -398,6 +399,7 @@ class Frontend {
       size_t content_len = 0;
       bool keep_alive = line.find("HTTP/1.1") != std::string::npos;
       bool expect_100 = false;
+      bool chunked = false;
       for (size_t p = line_end + 2; p < hdr_end;) {
         size_t e = c.in.find("\r\n", p);
         if (e == std::string::npos || e > hdr_end) e = hdr_end;
@@ -413,6 +415,9 @@ class Frontend {
             keep_alive = !iequal(c.in.c_str() + v, "close", 5);
           } else if (nlen == 6 && iequal(name, "expect", 6)) {
             expect_100 = true;
+          } else if (nlen == 17 &&
+                     iequal(name, "transfer-encoding", 17)) {
+            chunked = true;  // not supported: answered with 411 below
           }
         }
         p = e + 2;
@@ -421,6 +426,12 @@ class Frontend {
 
       if (content_len > kMaxBody) {
         send_inline(w, c, 413, "{\"error\":\"payload too large\"}");
+        c.closing = true;
+        break;
+      }
+      if (chunked) {
+        send_inline(w, c, 411,
+                    "{\"error\":\"length required (no chunked)\"}");
         c.closing = true;
         break;
       }
@@ -456,7 +467,22 @@ class Frontend {
         }
       } else {
         c.in.erase(0, body_start + content_len);
-        if (method == "GET" && path == "/healthz") {
+        if (method == "GET" && (path == "/" || path == "/index.html")) {
+          // reference-parity upload form (the Flask app serves the
+          // full template; this is the fast path's minimal equivalent)
+          std::string html =
+              "<!doctype html><title>predict</title><h3>image "
+              "classification</h3><form method=post action=/predict "
+              "enctype=multipart/form-data><input type=file "
+              "name=file><input type=submit></form>";
+          c.out += "HTTP/1.1 200 OK\r\nContent-Type: text/html\r\n"
+                   "Content-Length: " +
+                   std::to_string(html.size()) +
+                   (c.keep_alive ? "\r\nConnection: keep-alive\r\n\r\n"
+                                 : "\r\nConnection: close\r\n\r\n") +
+              html;
+          if (!c.keep_alive) c.closing = true;
+        } else if (method == "GET" && path == "/healthz") {
           send_inline(w, c, 200, "{\"status\":\"ok\"}");
         } else if (method == "GET" && path == "/stats") {
           std::string s = "{\"requests\":" +

@@ -56,6 +56,22 @@ def test_predict_roundtrip(server):
     assert probs == sorted(probs, reverse=True)
 
 
+def test_index_form_and_chunked_rejected(server):
+    with urllib.request.urlopen(
+            f"http://127.0.0.1:{server.port}/", timeout=10) as r:
+        body = r.read()
+        assert r.status == 200 and b"form" in body
+    # chunked request bodies are answered with 411, not mis-parsed
+    import socket
+    s = socket.create_connection(("127.0.0.1", server.port), timeout=10)
+    s.sendall(b"POST /predict HTTP/1.1\r\nHost: x\r\n"
+              b"Transfer-Encoding: chunked\r\n\r\n"
+              b"5\r\nhello\r\n0\r\n\r\n")
+    resp = s.recv(4096)
+    assert b"411" in resp.split(b"\r\n", 1)[0]
+    s.close()
+
+
 def test_healthz_inline_and_404(server):
     with urllib.request.urlopen(
             f"http://127.0.0.1:{server.port}/healthz", timeout=10) as r:
