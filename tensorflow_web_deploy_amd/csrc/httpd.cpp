@@ -400,6 +400,7 @@ class Frontend {
       bool keep_alive = line.find("HTTP/1.1") != std::string::npos;
       bool expect_100 = false;
       bool chunked = false;
+      std::string boundary;  // multipart/form-data (the upload form)
       for (size_t p = line_end + 2; p < hdr_end;) {
         size_t e = c.in.find("\r\n", p);
         if (e == std::string::npos || e > hdr_end) e = hdr_end;
@@ -418,6 +419,15 @@ class Frontend {
           } else if (nlen == 17 &&
                      iequal(name, "transfer-encoding", 17)) {
             chunked = true;  // not supported: answered with 411 below
+          } else if (nlen == 12 && iequal(name, "content-type", 12)) {
+            std::string val = c.in.substr(v, e - v);
+            size_t bp = val.find("boundary=");
+            if (bp != std::string::npos &&
+                val.compare(0, 19, "multipart/form-data") == 0) {
+              boundary = val.substr(bp + 9);
+              if (!boundary.empty() && boundary.front() == '"')
+                boundary = boundary.substr(1, boundary.size() - 2);
+            }
           }
         }
         p = e + 2;
@@ -448,6 +458,29 @@ class Frontend {
       if (method == "POST" &&
           (path == "/predict" || path == "/api/predict")) {
         std::string body = c.in.substr(body_start, content_len);
+        if (!boundary.empty()) {
+          // extract the first file part's payload (what the Flask
+          // surface does with request.files['file'])
+          std::string delim = "--" + boundary;
+          size_t p0 = body.find(delim);
+          std::string part;
+          while (p0 != std::string::npos) {
+            size_t hdr_beg = p0 + delim.size();
+            size_t hdr_end2 = body.find("\r\n\r\n", hdr_beg);
+            if (hdr_end2 == std::string::npos) break;
+            size_t nxt = body.find(delim, hdr_end2);
+            if (nxt == std::string::npos) break;
+            std::string phdr = body.substr(hdr_beg, hdr_end2 - hdr_beg);
+            size_t data_beg = hdr_end2 + 4;
+            size_t data_end = nxt >= 2 ? nxt - 2 : nxt;  // strip CRLF
+            if (phdr.find("filename=") != std::string::npos ||
+                part.empty())
+              part = body.substr(data_beg, data_end - data_beg);
+            if (phdr.find("filename=") != std::string::npos) break;
+            p0 = nxt;
+          }
+          if (!part.empty()) body = std::move(part);
+        }
         c.in.erase(0, body_start + content_len);
         requests_.fetch_add(1);
         bool enq = false;

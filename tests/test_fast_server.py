@@ -341,3 +341,27 @@ def test_stress_mixed_traffic(server):
     assert len(ok) == 12 * 8
     st = server.fe.stats()
     assert st["responded"] + st["rejected_503"] >= 50
+
+
+def test_multipart_upload(server):
+    """The HTML form posts multipart/form-data; the frontend extracts
+    the file part in C++ (same contract as Flask request.files)."""
+    img = _png_bytes()
+    boundary = "----twdboundary123"
+    body = (
+        f"--{boundary}\r\n"
+        "Content-Disposition: form-data; name=\"note\"\r\n\r\n"
+        "hello\r\n"
+        f"--{boundary}\r\n"
+        "Content-Disposition: form-data; name=\"file\"; "
+        "filename=\"x.png\"\r\n"
+        "Content-Type: image/png\r\n\r\n"
+    ).encode() + img + f"\r\n--{boundary}--\r\n".encode()
+    req = urllib.request.Request(
+        f"http://127.0.0.1:{server.port}/predict", data=body,
+        headers={"Content-Type":
+                 f"multipart/form-data; boundary={boundary}"})
+    with urllib.request.urlopen(req, timeout=15) as r:
+        j = json.loads(r.read())
+        assert r.status == 200
+        assert len(j["predictions"]) == 5
