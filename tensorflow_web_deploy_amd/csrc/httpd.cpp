@@ -523,6 +523,29 @@ class Frontend {
                           ",\"responded\":" +
                           std::to_string(responded_.load()) + "}";
           send_inline(w, c, 200, s);
+        } else if (method == "GET" && path == "/metrics") {
+          // prometheus exposition (parity with the Flask /metrics)
+          std::string m;
+          m += "# TYPE twd_requests_total counter\n";
+          m += "twd_requests_total " +
+               std::to_string(requests_.load()) + "\n";
+          m += "# TYPE twd_responses_total counter\n";
+          m += "twd_responses_total " +
+               std::to_string(responded_.load()) + "\n";
+          m += "# TYPE twd_rejected_total counter\n";
+          m += "twd_rejected_total " +
+               std::to_string(rejected_.load()) + "\n";
+          m += "# TYPE twd_connections_accepted_total counter\n";
+          m += "twd_connections_accepted_total " +
+               std::to_string(accepted_.load()) + "\n";
+          c.out += "HTTP/1.1 200 OK\r\nContent-Type: text/plain; "
+                   "version=0.0.4\r\nContent-Length: " +
+                   std::to_string(m.size()) +
+                   (c.keep_alive
+                        ? "\r\nConnection: keep-alive\r\n\r\n"
+                        : "\r\nConnection: close\r\n\r\n") +
+               m;
+          if (!c.keep_alive) c.closing = true;
         } else {
           send_inline(w, c, 404, "{\"error\":\"not found\"}");
         }

@@ -381,8 +381,13 @@ def main(argv=None):
     gc.collect()
     gc.freeze()
     gc.set_threshold(200000, 100, 100)
+    import signal
+    signal.signal(signal.SIGTERM, lambda *_: srv.stop())
     print(f"[fast] serving {srv.model_name} on :{srv.port}", flush=True)
-    srv.serve_forever()
+    try:
+        srv.serve_forever()
+    except KeyboardInterrupt:
+        srv.stop()
 
 
 if __name__ == "__main__":

@@ -365,3 +365,13 @@ def test_multipart_upload(server):
         j = json.loads(r.read())
         assert r.status == 200
         assert len(j["predictions"]) == 5
+
+
+def test_metrics_prometheus(server):
+    _post(server.port, _png_bytes())
+    with urllib.request.urlopen(
+            f"http://127.0.0.1:{server.port}/metrics", timeout=10) as r:
+        text = r.read().decode()
+        assert r.status == 200
+        assert "twd_requests_total" in text
+        assert "twd_responses_total" in text
