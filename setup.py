@@ -34,18 +34,20 @@ ext = cpp_extension.CUDAExtension(
     },
 )
 
-# Pure-C++ epoll HTTP frontend (no torch/HIP): serving fast path
-httpd = cpp_extension.CppExtension(
-    name="tensorflow_web_deploy_amd._httpd",
-    sources=[os.path.join(CSRC, "httpd.cpp")],
-    extra_compile_args={"cxx": ["-O2", "-std=c++17"]},
-)
-
-# Baseline JPEG entropy decoder. Plain Extension (NOT CppExtension):
-# it must import without libtorch so the shm-ring decode workers load
-# it cheaply.
+# Torch-free native modules (plain Extension, NOT CppExtension: they
+# must import without libtorch — the server CLI loads the frontend
+# before torch, and the shm-ring decode workers load the JPEG decoder
+# without torch at all).
 import pybind11  # noqa: E402
 from setuptools import Extension  # noqa: E402
+
+httpd = Extension(
+    name="tensorflow_web_deploy_amd._httpd",
+    sources=[os.path.join(CSRC, "httpd.cpp")],
+    include_dirs=[pybind11.get_include()],
+    extra_compile_args=["-O2", "-std=c++17"],
+    language="c++",
+)
 
 jpegc = Extension(
     name="tensorflow_web_deploy_amd._jpegc",

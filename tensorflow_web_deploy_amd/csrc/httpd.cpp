@@ -191,9 +191,16 @@ class Frontend {
     {
       py::gil_scoped_release nogil;
       std::unique_lock<std::mutex> lk(qmu_);
-      // block (100 ms heartbeat so stop() is honored) for the 1st req
-      while (running_.load() && reqq_.empty())
+      // block for the 1st request, but give the GIL back after ~1 s
+      // idle: Python signal handlers (SIGTERM/SIGINT) only run between
+      // bytecodes of the main thread, so an unbounded C++ wait here
+      // makes the server unkillable while idle
+      auto idle_deadline = std::chrono::steady_clock::now() +
+                           std::chrono::milliseconds(1000);
+      while (running_.load() && reqq_.empty()) {
+        if (std::chrono::steady_clock::now() >= idle_deadline) break;
         qcv_.wait_for(lk, std::chrono::milliseconds(100));
+      }
       if (!reqq_.empty()) {
         auto deadline = std::chrono::steady_clock::now() +
                         std::chrono::duration<double, std::milli>(max_wait_ms);

@@ -9,7 +9,7 @@ Python ever runs, which removes the werkzeug/batcher GIL contention
 measured in BASELINE.md (batch work 0.3-3 ms alone vs 73-91 ms in situ
 under Flask).
 
-  python -m tensorflow_web_deploy_amd.server.fast --model resnet50 ...
+  python -m tensorflow_web_deploy_amd.server.fast --model-name resnet50
 """
 from __future__ import annotations
 
@@ -382,7 +382,10 @@ def main(argv=None):
     gc.freeze()
     gc.set_threshold(200000, 100, 100)
     import signal
-    signal.signal(signal.SIGTERM, lambda *_: srv.stop())
+    try:
+        signal.signal(signal.SIGTERM, lambda *_: srv.stop())
+    except ValueError:
+        pass  # not the main thread (embedded use)
     print(f"[fast] serving {srv.model_name} on :{srv.port}", flush=True)
     try:
         srv.serve_forever()
