@@ -84,6 +84,20 @@ def _convblock(node, ins, consts):
     return y
 
 
+@cpu_op("DepthwiseConv2dNative")
+def _dwconv_raw(node, ins, consts):
+    # unfused form (graphs where block fusion did not engage)
+    x, w = ins[0], ins[1]
+    kh, kw, c, mult = w.shape
+    wt = w.permute(2, 3, 0, 1).reshape(c * mult, 1, kh, kw)
+    sh, sw = int(node.attrs["strides"][1]), int(node.attrs["strides"][2])
+    xc = x.permute(0, 3, 1, 2).contiguous()
+    if node.attrs.get("padding", "VALID") == "SAME":
+        xc = tf_same_pad(xc, kh, kw, sh, sw)
+    y = F.conv2d(xc, wt, stride=(sh, sw), groups=c)
+    return y.permute(0, 2, 3, 1).contiguous()
+
+
 @cpu_op("DepthwiseConvBlock")
 def _dwconvblock(node, ins, consts):
     x, w = ins[0], ins[1]

@@ -1,5 +1,6 @@
 """IR lowering-pass tests: optimized graph must match the raw graph."""
 import numpy as np
+import pytest
 import torch
 
 from tensorflow_web_deploy_amd.ckpt import graphdef as gd
@@ -134,3 +135,28 @@ def test_multi_output_split_through_executor():
     out, packed = ex.run({"x": x})
     assert torch.equal(out, x[:, :2] + x[:, 2:])
     assert packed.shape == (2, 2, 2)
+
+
+@pytest.mark.parametrize("family", ["lenet", "resnet50", "mobilenet_v2"])
+def test_optimize_preserves_semantics(family):
+    """optimize() (BN fold, block fusion, concat elimination, ...) must
+    not change the function: raw-graph execution == optimized-graph
+    execution on the same random input."""
+    import numpy as np
+    import torch
+
+    from tensorflow_web_deploy_amd.engine import ir
+    from tensorflow_web_deploy_amd.engine.executor import Executor
+    from tensorflow_web_deploy_amd.models import lenet, mobilenet_v2, resnet50
+    build = {"lenet": lenet.build_lenet,
+             "resnet50": resnet50.build_resnet50,
+             "mobilenet_v2": mobilenet_v2.build_mobilenet_v2}[family]
+    g = build(seed=3)
+    raw = ir.from_graphdef(g, outputs=["softmax"])
+    opt = ir.optimize(ir.from_graphdef(g, outputs=["softmax"]))
+    shape = (1, 28, 28, 1) if family == "lenet" else (1, 224, 224, 3)
+    x = torch.from_numpy(
+        np.random.default_rng(1).random(shape, dtype=np.float32))
+    p_raw = Executor(raw).run({"input": x})[0]
+    p_opt = Executor(opt).run({"input": x})[0]
+    torch.testing.assert_close(p_raw, p_opt, atol=2e-4, rtol=2e-4)
