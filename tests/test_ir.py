@@ -137,7 +137,8 @@ def test_multi_output_split_through_executor():
     assert packed.shape == (2, 2, 2)
 
 
-@pytest.mark.parametrize("family", ["lenet", "resnet50", "mobilenet_v2"])
+@pytest.mark.parametrize("family", ["lenet", "resnet50", "mobilenet_v2",
+                                    "inception_v3", "vit_b16"])
 def test_optimize_preserves_semantics(family):
     """optimize() (BN fold, block fusion, concat elimination, ...) must
     not change the function: raw-graph execution == optimized-graph
@@ -147,14 +148,20 @@ def test_optimize_preserves_semantics(family):
 
     from tensorflow_web_deploy_amd.engine import ir
     from tensorflow_web_deploy_amd.engine.executor import Executor
-    from tensorflow_web_deploy_amd.models import lenet, mobilenet_v2, resnet50
+    from tensorflow_web_deploy_amd.models import (inception_v3, lenet,
+                                                  mobilenet_v2, resnet50,
+                                                  vit)
     build = {"lenet": lenet.build_lenet,
              "resnet50": resnet50.build_resnet50,
-             "mobilenet_v2": mobilenet_v2.build_mobilenet_v2}[family]
+             "mobilenet_v2": mobilenet_v2.build_mobilenet_v2,
+             "inception_v3": inception_v3.build_inception_v3,
+             "vit_b16": vit.build_vit_b16}[family]
     g = build(seed=3)
     raw = ir.from_graphdef(g, outputs=["softmax"])
     opt = ir.optimize(ir.from_graphdef(g, outputs=["softmax"]))
-    shape = (1, 28, 28, 1) if family == "lenet" else (1, 224, 224, 3)
+    shape = {"lenet": (1, 28, 28, 1),
+             "inception_v3": (1, 299, 299, 3)}.get(family,
+                                                   (1, 224, 224, 3))
     x = torch.from_numpy(
         np.random.default_rng(1).random(shape, dtype=np.float32))
     p_raw = Executor(raw).run({"input": x})[0]
