@@ -92,8 +92,10 @@ def main():
     p.add_argument("--max-wait-ms", type=float, default=2.0)
     p.add_argument("--http-workers", type=int, default=0,
                    help=">0: multi-process SO_REUSEPORT serving stack")
-    p.add_argument("--fast", action="store_true",
-                   help="C++ epoll frontend + batch loop (server/fast.py)")
+    p.add_argument("--fast", action="store_true", default=True,
+                   help="C++ epoll frontend + batch loop (default)")
+    p.add_argument("--flask", dest="fast", action="store_false",
+                   help="measure the Flask/werkzeug stack instead")
     args = p.parse_args()
 
     import torch
