@@ -186,9 +186,13 @@ void bias_act_bf16(torch::Tensor X, c10::optional<torch::Tensor> bias,
                    torch::Tensor Y, int64_t act) {
   check_bf16(X, "X");
   check_bf16(Y, "Y");
-  const float* bptr = nullptr;
-  if (bias) bptr = bias->data_ptr<float>();
   int C = X.size(-1);
+  const float* bptr = nullptr;
+  if (bias) {
+    TORCH_CHECK(bias->numel() == C, "bias size ", bias->numel(),
+                " != channels ", C);
+    bptr = bias->data_ptr<float>();
+  }
   launch_bias_act_bf16(X.data_ptr(), bptr, Y.data_ptr(), X.numel() / C, C,
                        (int)act, cur_stream());
 }
@@ -198,6 +202,8 @@ void scale_shift_bf16(torch::Tensor X, torch::Tensor scale,
   check_bf16(X, "X");
   check_bf16(Y, "Y");
   int C = X.size(-1);
+  TORCH_CHECK(scale.numel() == C && shift.numel() == C,
+              "scale/shift size mismatch vs channels ", C);
   launch_scale_shift_bf16(X.data_ptr(), scale.data_ptr<float>(),
                           shift.data_ptr<float>(), Y.data_ptr(),
                           X.numel() / C, C, relu, cur_stream());

@@ -13,7 +13,8 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 __device__ __forceinline__ float bf2f(__bf16 v) { return (float)v; }
 
 static inline int grid_for(int64_t work) {
-  return (int)std::min<int64_t>((work + 255) / 256, 2048);
+  return (int)std::max<int64_t>(
+      1, std::min<int64_t>((work + 255) / 256, 2048));
 }
 
 // -------------------------------------------------------------- casts
@@ -117,7 +118,12 @@ void launch_cast_f32_pad_bf16(const float* X, void* Y, int64_t rows,
 __global__ void add_relu_kernel(const __bf16* __restrict__ A,
                                 const __bf16* __restrict__ B,
                                 __bf16* __restrict__ Y, int64_t n8,
-                                int relu) {
+                                int64_t n, int relu) {
+  if (blockIdx.x == 0 && threadIdx.x == 0)
+    for (int64_t i = n8 * 8; i < n; ++i) {
+      float v = bf2f(A[i]) + bf2f(B[i]);
+      Y[i] = (__bf16)(relu && v < 0.f ? 0.f : v);
+    }
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
        i += (int64_t)gridDim.x * blockDim.x) {
     bf16x8 a = *(const bf16x8*)(A + i * 8);
@@ -135,9 +141,9 @@ __global__ void add_relu_kernel(const __bf16* __restrict__ A,
 void launch_add_relu_bf16(const void* A, const void* B, void* Y, int64_t n,
                           bool relu, hipStream_t stream) {
   int64_t n8 = n / 8;
-  hipLaunchKernelGGL(add_relu_kernel, dim3(grid_for(n8)), dim3(256), 0,
+  hipLaunchKernelGGL(add_relu_kernel, dim3(grid_for(n8 + 1)), dim3(256), 0,
                      stream, (const __bf16*)A, (const __bf16*)B, (__bf16*)Y,
-                     n8, relu ? 1 : 0);
+                     n8, n, relu ? 1 : 0);
 }
 
 // ------------------------------------------------ bias + activation
@@ -148,11 +154,13 @@ __global__ void bias_act_kernel(const __bf16* __restrict__ X,
                                 __bf16* __restrict__ Y, int64_t rows, int C,
                                 int act) {
   int c8 = C >> 3;
+  int ct = C & 7;  // tail channels when C % 8 != 0
   int64_t total = rows * c8;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (int64_t)gridDim.x * blockDim.x) {
     int cg = (int)(i % c8);
-    bf16x8 v = *(const bf16x8*)(X + i * 8);
+    int64_t off = (i / c8) * C + (int64_t)cg * 8;
+    bf16x8 v = *(const bf16x8*)(X + off);
     f32x4 bl = bias ? *(const f32x4*)(bias + cg * 8) : f32x4{0, 0, 0, 0};
     f32x4 bh = bias ? *(const f32x4*)(bias + cg * 8 + 4) : f32x4{0, 0, 0, 0};
     bf16x8 o;
@@ -168,13 +176,30 @@ __global__ void bias_act_kernel(const __bf16* __restrict__ X,
       }
       o[e] = (__bf16)f;
     }
-    *(bf16x8*)(Y + i * 8) = o;
+    *(bf16x8*)(Y + off) = o;
+  }
+  if (ct) {
+    int64_t ntail = rows * ct;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < ntail; i += (int64_t)gridDim.x * blockDim.x) {
+      int c = c8 * 8 + (int)(i % ct);
+      int64_t off = (i / ct) * C + c;
+      float f = bf2f(X[off]) + (bias ? bias[c] : 0.f);
+      switch (act) {
+        case 1: f = fmaxf(f, 0.f); break;
+        case 2: f = fminf(fmaxf(f, 0.f), 6.f); break;
+        case 3: f = tanhf(f); break;
+        case 4: f = 1.f / (1.f + __expf(-f)); break;
+        case 5: f = 0.5f * f * (1.f + erff(f * 0.70710678f)); break;
+      }
+      Y[off] = (__bf16)f;
+    }
   }
 }
 
 void launch_bias_act_bf16(const void* X, const float* bias, void* Y,
                           int64_t rows, int C, int act, hipStream_t stream) {
-  int64_t total = rows * (C >> 3);
+  int64_t total = rows * ((C >> 3) + (C & 7 ? 1 : 0));
   hipLaunchKernelGGL(bias_act_kernel, dim3(grid_for(total)), dim3(256), 0,
                      stream, (const __bf16*)X, bias, (__bf16*)Y, rows, C,
                      act);
@@ -188,25 +213,37 @@ __global__ void scale_shift_kernel(const __bf16* __restrict__ X,
                                    __bf16* __restrict__ Y, int64_t rows,
                                    int C, int relu) {
   int c8 = C >> 3;
+  int ct = C & 7;
   int64_t total = rows * c8;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (int64_t)gridDim.x * blockDim.x) {
     int cg = (int)(i % c8);
-    bf16x8 v = *(const bf16x8*)(X + i * 8);
+    int64_t off = (i / c8) * C + (int64_t)cg * 8;
+    bf16x8 v = *(const bf16x8*)(X + off);
     bf16x8 o;
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
       float f = bf2f(v[e]) * scale[cg * 8 + e] + shift[cg * 8 + e];
       o[e] = (__bf16)(relu && f < 0.f ? 0.f : f);
     }
-    *(bf16x8*)(Y + i * 8) = o;
+    *(bf16x8*)(Y + off) = o;
+  }
+  if (ct) {
+    int64_t ntail = rows * ct;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < ntail; i += (int64_t)gridDim.x * blockDim.x) {
+      int c = c8 * 8 + (int)(i % ct);
+      int64_t off = (i / ct) * C + c;
+      float f = bf2f(X[off]) * scale[c] + shift[c];
+      Y[off] = (__bf16)(relu && f < 0.f ? 0.f : f);
+    }
   }
 }
 
 void launch_scale_shift_bf16(const void* X, const float* scale,
                              const float* shift, void* Y, int64_t rows, int C,
                              bool relu, hipStream_t stream) {
-  int64_t total = rows * (C >> 3);
+  int64_t total = rows * ((C >> 3) + (C & 7 ? 1 : 0));
   hipLaunchKernelGGL(scale_shift_kernel, dim3(grid_for(total)), dim3(256), 0,
                      stream, (const __bf16*)X, scale, shift, (__bf16*)Y, rows,
                      C, relu ? 1 : 0);

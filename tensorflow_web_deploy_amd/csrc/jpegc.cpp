@@ -211,6 +211,7 @@ py::dict jpeg_parse(py::bytes data_b) {
         const size_t segend = pos + 2 + len;
         while (p < segend) {
           int pq = d[p] >> 4, tq = d[p] & 15;
+          if (tq > 3) throw std::invalid_argument("bad DQT table id");
           ++p;
           if (p + (pq ? 128 : 64) > segend)
             throw std::invalid_argument("truncated DQT");
@@ -225,6 +226,8 @@ py::dict jpeg_parse(py::bytes data_b) {
         const size_t segend = pos + 2 + len;
         while (p < segend) {
           int tc = d[p] >> 4, th = d[p] & 15;
+          if (tc > 1 || th > 3)
+            throw std::invalid_argument("bad DHT table id");
           ++p;
           if (p + 16 > segend)
             throw std::invalid_argument("truncated DHT");
@@ -258,6 +261,8 @@ py::dict jpeg_parse(py::bytes data_b) {
           comps[c].h = d[seg + 7 + 3 * c] >> 4;
           comps[c].v = d[seg + 7 + 3 * c] & 15;
           comps[c].tq = d[seg + 8 + 3 * c];
+          if (comps[c].tq > 3)
+            throw std::invalid_argument("bad SOF quant table id");
           if (comps[c].h < 1 || comps[c].h > 2 || comps[c].v < 1 ||
               comps[c].v > 2)
             throw std::invalid_argument("unsupported sampling");
@@ -281,6 +286,8 @@ py::dict jpeg_parse(py::bytes data_b) {
             if (c.id == cid) {
               c.td = tt >> 4;
               c.ta = tt & 15;
+              if (c.td > 3 || c.ta > 3)
+                throw std::invalid_argument("bad SOS huffman table id");
             }
         }
         scan_off = seg + 1 + 2 * ns + 3;

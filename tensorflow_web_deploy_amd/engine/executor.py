@@ -492,22 +492,59 @@ def _pack(node, ins, consts):
     return torch.stack(list(ins), dim=axis)
 
 
+def tf_resize_bilinear(x_nchw, size, align_corners, half_pixel):
+    """TF-semantics bilinear resize on an NCHW tensor.
+
+    TF1's three coordinate mappings (ref: tf.image.resize_bilinear):
+      align_corners=True          src = dst * (in-1)/(out-1)   == torch ac=True
+      half_pixel_centers=True     src = (dst+.5)*scale - .5    == torch ac=False
+      default (both False)        src = dst * scale            (asymmetric;
+                                  no torch equivalent — computed explicitly)
+    """
+    oh, ow = int(size[0]), int(size[1])
+    if align_corners:
+        return F.interpolate(x_nchw, size=(oh, ow), mode="bilinear",
+                             align_corners=True)
+    if half_pixel:
+        return F.interpolate(x_nchw, size=(oh, ow), mode="bilinear",
+                             align_corners=False)
+    N, C, H, W = x_nchw.shape
+    dev = x_nchw.device
+    x = x_nchw.float()
+
+    def _axis(n_in, n_out):
+        f = torch.arange(n_out, device=dev, dtype=torch.float32) \
+            * (n_in / n_out)
+        i0 = f.floor().long().clamp_(0, n_in - 1)
+        i1 = (i0 + 1).clamp_(max=n_in - 1)
+        return i0, i1, (f - i0.float())
+
+    y0, y1, wy = _axis(H, oh)
+    x0, x1, wx = _axis(W, ow)
+    top = x.index_select(2, y0)
+    bot = x.index_select(2, y1)
+    xr = top + wy.view(1, 1, -1, 1) * (bot - top)
+    left = xr.index_select(3, x0)
+    right = xr.index_select(3, x1)
+    return left + wx.view(1, 1, 1, -1) * (right - left)
+
+
 @cpu_op("ResizeBilinear")
 def _resize_bilinear(node, ins, consts):
-    # TF semantics: align_corners / half_pixel_centers attrs
     x = ins[0].permute(0, 3, 1, 2)
     size = [int(v) for v in ins[1].tolist()]
     ac = bool(node.attrs.get("align_corners"))
-    y = F.interpolate(x, size=tuple(size), mode="bilinear",
-                      align_corners=ac)
-    return y.permute(0, 2, 3, 1)
+    hp = bool(node.attrs.get("half_pixel_centers"))
+    return tf_resize_bilinear(x, size, ac, hp).permute(0, 2, 3, 1)
 
 
 @cpu_op("ResizeNearestNeighbor")
 def _resize_nearest(node, ins, consts):
     x = ins[0].permute(0, 3, 1, 2)
     size = [int(v) for v in ins[1].tolist()]
-    y = F.interpolate(x, size=tuple(size), mode="nearest")
+    hp = bool(node.attrs.get("half_pixel_centers"))
+    y = F.interpolate(x, size=tuple(size),
+                      mode="nearest-exact" if hp else "nearest")
     return y.permute(0, 2, 3, 1)
 
 

@@ -599,8 +599,18 @@ class GpuExecutor:
                     cb = self.consts[b].astype(np.float32)
                     sh_a = shapes[a]
                     Y = alloc(node.name, sh_a)
-                    if cb.ndim <= 1 or cb.size == sh_a[-1]:
-                        bias = self._bias(b, cb.size)
+                    C = int(sh_a[-1])
+                    if cb.size == 1 or cb.size == C:
+                        # bias_act's kernel reads bias[0..C-1]: expand a
+                        # scalar const to C on the host before upload
+                        key = b + f"/biasf32x{C}"
+                        if key not in self._weights:
+                            bexp = np.broadcast_to(
+                                cb.reshape(-1) if cb.size == C else cb.ravel(),
+                                (C,)).astype(np.float32)
+                            self._weights[key] = torch.from_numpy(
+                                np.ascontiguousarray(bexp)).to(dev)
+                        bias = self._weights[key]
                         steps.append(lambda X=bufs[a], Y=Y, bias=bias:
                                      E.bias_act_bf16(X, bias, Y, ACT_NONE))
                     else:
@@ -695,18 +705,27 @@ class GpuExecutor:
                         self.consts[node.inputs[1]].reshape(-1)]
                 Y = alloc(node.name, (N_, size[0], size[1], C))
                 ac = bool(node.attrs.get("align_corners"))
-                mode = ("bilinear" if op == "ResizeBilinear"
-                        else "nearest")
-                import torch.nn.functional as _F
+                hp = bool(node.attrs.get("half_pixel_centers"))
+                if op == "ResizeBilinear":
+                    from .executor import tf_resize_bilinear
 
-                def _rs(X=bufs[xn], Y=Y, size=tuple(size), mode=mode,
-                        ac=ac):
-                    x4 = X.float().permute(0, 3, 1, 2)
-                    r = _F.interpolate(
-                        x4, size=size, mode=mode,
-                        align_corners=(ac if mode == "bilinear"
-                                       else None))
-                    Y.copy_(r.permute(0, 2, 3, 1).to(Y.dtype))
+                    def _rs(X=bufs[xn], Y=Y, size=tuple(size), ac=ac,
+                            hp=hp):
+                        x4 = X.float().permute(0, 3, 1, 2)
+                        r = tf_resize_bilinear(x4, size, ac, hp)
+                        Y.copy_(r.permute(0, 2, 3, 1).to(Y.dtype))
+                else:
+                    # TF nearest default (src = floor(dst*scale)) matches
+                    # torch mode="nearest"; half_pixel matches
+                    # "nearest-exact"
+                    import torch.nn.functional as _F
+
+                    def _rs(X=bufs[xn], Y=Y, size=tuple(size), hp=hp):
+                        x4 = X.float().permute(0, 3, 1, 2)
+                        r = _F.interpolate(
+                            x4, size=size,
+                            mode="nearest-exact" if hp else "nearest")
+                        Y.copy_(r.permute(0, 2, 3, 1).to(Y.dtype))
                 steps.append(_rs)
             else:
                 raise NotImplementedError(

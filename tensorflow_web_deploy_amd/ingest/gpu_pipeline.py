@@ -165,8 +165,10 @@ def _ring_worker(path, widx, nworkers, channels, slot_bytes, slots,
                     if sh is not None:
                         meta[4 * i:4 * i + 4] = (sh[0], sh[1], sh[2], 3)
                         continue
-                except ValueError:
-                    pass  # progressive/CMYK/...: PIL pixel path below
+                except (ValueError, RuntimeError):
+                    # invalid_argument (progressive/CMYK/...) or runtime_error
+                    # (corrupt entropy stream): PIL pixel path below
+                    pass
             try:
                 img = Image.open(io.BytesIO(data))
                 img = img.convert("L" if channels == 1 else "RGB")
@@ -188,15 +190,6 @@ def _ring_worker(path, widx, nworkers, channels, slot_bytes, slots,
             return
 
 
-def _ring_worker_wrap(*a, **kw):
-    try:
-        _ring_worker_inner(*a, **kw)
-    except Exception:
-        import sys
-        import traceback
-        traceback.print_exc(file=sys.stderr)
-        sys.stderr.flush()
-        raise
 
 
 class ShmDecodePool:

@@ -104,14 +104,47 @@ def test_resize_bilinear_cpu():
     from tensorflow_web_deploy_amd.engine.executor import _CPU_OPS
     from tensorflow_web_deploy_amd.engine.ir import Node
     x = torch.rand(1, 6, 8, 3)
+    # half_pixel_centers=True == torch align_corners=False
     out = _CPU_OPS["ResizeBilinear"](
-        Node("r", "ResizeBilinear", [], {"align_corners": False}),
+        Node("r", "ResizeBilinear", [],
+             {"align_corners": False, "half_pixel_centers": True}),
         [x, torch.tensor([12, 16])], {})
     assert out.shape == (1, 12, 16, 3)
     ref = torch.nn.functional.interpolate(
         x.permute(0, 3, 1, 2), size=(12, 16), mode="bilinear",
         align_corners=False).permute(0, 2, 3, 1)
     assert torch.allclose(out, ref)
+
+
+def test_resize_bilinear_tf1_default_asymmetric():
+    """TF1 default (align_corners=False, half_pixel_centers=False) maps
+    src = dst * in/out — NOT half-pixel (ADVICE round-1 low). Check
+    against a scalar reference implementation."""
+    import torch
+
+    from tensorflow_web_deploy_amd.engine.executor import _CPU_OPS
+    from tensorflow_web_deploy_amd.engine.ir import Node
+    torch.manual_seed(0)
+    x = torch.rand(2, 5, 7, 3)
+    out = _CPU_OPS["ResizeBilinear"](
+        Node("r", "ResizeBilinear", [], {"align_corners": False}),
+        [x, torch.tensor([9, 4])], {})
+    H, W, oh, ow = 5, 7, 9, 4
+    ref = torch.empty(2, oh, ow, 3)
+    for oy in range(oh):
+        fy = oy * H / oh
+        y0 = min(int(fy), H - 1)
+        y1 = min(y0 + 1, H - 1)
+        wy = fy - y0
+        for ox in range(ow):
+            fx = ox * W / ow
+            x0 = min(int(fx), W - 1)
+            x1 = min(x0 + 1, W - 1)
+            wx = fx - x0
+            top = x[:, y0, x0] + wx * (x[:, y0, x1] - x[:, y0, x0])
+            bot = x[:, y1, x0] + wx * (x[:, y1, x1] - x[:, y1, x0])
+            ref[:, oy, ox] = top + wy * (bot - top)
+    assert torch.allclose(out, ref, atol=1e-5)
 
 
 def test_multi_output_split_through_executor():

@@ -95,3 +95,69 @@ def test_progressive_rejected(img):
 def test_garbage_rejected():
     with pytest.raises(ValueError):
         _jpegc.parse(b"definitely not a jpeg")
+
+def _patch_marker(data: bytes, marker: int, offset: int, value: int) -> bytes:
+    """Return a copy of `data` with byte `offset` (relative to the first
+    `marker` segment's payload start, i.e. after the 2-byte length) set to
+    `value`."""
+    i = 2
+    b = bytearray(data)
+    while i + 4 <= len(b):
+        assert b[i] == 0xFF
+        m = b[i + 1]
+        seglen = (b[i + 2] << 8) | b[i + 3]
+        if m == marker:
+            b[i + 4 + offset] = value
+            return bytes(b)
+        i += 2 + seglen
+    raise AssertionError(f"marker {marker:#x} not found")
+
+
+def test_bad_dqt_table_id_rejected(img):
+    # ADVICE round-1 high: 4-bit table ids up to 15 indexed qt[4] —
+    # OOB stack write. Must raise, not crash.
+    data = _jpeg(img, "RGB", quality=90)
+    bad = _patch_marker(data, 0xDB, 0, 0x0F)  # tq=15
+    with pytest.raises(ValueError, match="DQT"):
+        _jpegc.parse(bad)
+
+
+def test_bad_dht_table_id_rejected(img):
+    data = _jpeg(img, "RGB", quality=90)
+    bad = _patch_marker(data, 0xC4, 0, 0x0F)  # tc=0, th=15
+    with pytest.raises(ValueError, match="DHT"):
+        _jpegc.parse(bad)
+
+
+def test_bad_sof_quant_id_rejected(img):
+    data = _jpeg(img, "RGB", quality=90)
+    # SOF0 payload: precision(1) H(2) W(2) nc(1) then per-comp id,hv,tq
+    bad = _patch_marker(data, 0xC0, 8, 0x0F)  # comp0 tq=15
+    with pytest.raises(ValueError, match="quant"):
+        _jpegc.parse(bad)
+
+
+def test_bad_sos_huff_id_rejected(img):
+    data = _jpeg(img, "RGB", quality=90)
+    # SOS payload: ns(1) then per-comp id,tdta
+    bad = _patch_marker(data, 0xDA, 2, 0xFF)  # comp0 td=15, ta=15
+    with pytest.raises(ValueError, match="SOS"):
+        _jpegc.parse(bad)
+
+
+def test_corrupt_entropy_raises_runtime(img):
+    # truncate/flip bits mid-scan until the huffman decoder objects; the
+    # ingest worker must survive this (RuntimeError is caught there).
+    data = bytearray(_jpeg(img, "RGB", quality=90))
+    sos = data.find(b"\xff\xda")
+    assert sos > 0
+    raised = False
+    for off in range(sos + 20, min(sos + 220, len(data) - 2)):
+        trial = bytes(data[:off]) + bytes([data[off] ^ 0xFF]) + bytes(
+            data[off + 1:])
+        try:
+            _jpegc.parse(trial)
+        except (RuntimeError, ValueError):
+            raised = True
+            break
+    assert raised

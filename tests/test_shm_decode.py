@@ -61,3 +61,36 @@ def test_reuse_across_batches(pool):
         meta = pool.decode_batch([_jpeg(50, 40) for _ in range(8)])
         assert meta is not None and len(meta) == 8
         assert all(tuple(m[1:]) == (50, 40, 3) for m in meta)
+
+
+def test_corrupt_entropy_does_not_kill_workers(pool):
+    """ADVICE round-1 medium: a corrupt-but-wellformed JPEG raised
+    RuntimeError past the worker's ValueError-only catch, killing the
+    worker and permanently disabling the shm fast path. The worker must
+    fall through to PIL (bad image -> per-index error) and stay alive."""
+    good = _jpeg(32, 32)
+    data = bytearray(_jpeg(64, 64))
+    sos = data.find(b"\xff\xda")
+    corrupt = None
+    from tensorflow_web_deploy_amd import _jpegc
+    for off in range(sos + 20, min(sos + 220, len(data) - 2)):
+        trial = bytes(data[:off]) + bytes([data[off] ^ 0xFF]) + bytes(
+            data[off + 1:])
+        try:
+            _jpegc.parse(trial)
+        except RuntimeError:
+            corrupt = trial
+            break
+        except ValueError:
+            continue
+    if corrupt is None:
+        pytest.skip("could not craft a RuntimeError-raising JPEG")
+    # PIL may decode it (sloppily) or reject it; either way the pool
+    # must survive and keep serving subsequent batches.
+    try:
+        pool.decode_batch([good, corrupt, good])
+    except ValueError:
+        pass
+    assert not pool._dead
+    meta = pool.decode_batch([good, good])
+    assert meta is not None
