@@ -84,6 +84,57 @@ class TorchDistComm:
         self.dist.barrier()
 
 
+def broadcast_graph(graph, rank: int, device: torch.device, dist):
+    """Rank0 fans the loaded model out to every replica (SURVEY §2b row
+    "model load"): ONE disk read on rank0, then (1) the graph skeleton +
+    const metadata via broadcast_object_list, (2) all const/weight bytes
+    packed into a single flat tensor broadcast once — on the nccl(=RCCL)
+    backend that is one xGMI fan-out (~51 MB for bf16 ResNet-50) instead
+    of 8 independent disk loads / random inits.
+
+    Rank0 passes its Graph and gets it back unchanged; other ranks pass
+    None and receive a fully-populated copy.
+    """
+    import numpy as np
+
+    from ..engine.ir import Graph
+    if rank == 0:
+        names = sorted(graph.consts)
+        metas = [(n, graph.consts[n].dtype.str,
+                  tuple(graph.consts[n].shape)) for n in names]
+        skeleton = Graph(graph.nodes, {}, graph.inputs, graph.outputs)
+        obj = [skeleton, metas]
+    else:
+        obj = [None, None]
+    dist.broadcast_object_list(obj, src=0)
+    skeleton, metas = obj
+    sizes = [int(np.prod(s)) * np.dtype(d).itemsize if len(s)
+             else np.dtype(d).itemsize for (_, d, s) in metas]
+    total = int(sum(sizes))
+    if total == 0:
+        return graph if rank == 0 else skeleton
+    buf = torch.empty(total, dtype=torch.uint8, device=device)
+    if rank == 0:
+        packed = np.empty(total, dtype=np.uint8)
+        off = 0
+        for (n, _, _), nb in zip(metas, sizes):
+            packed[off:off + nb] = np.frombuffer(
+                np.ascontiguousarray(graph.consts[n]).tobytes(), np.uint8)
+            off += nb
+        buf.copy_(torch.from_numpy(packed))
+    dist.broadcast(buf, src=0)
+    if rank == 0:
+        return graph
+    host = buf.cpu().numpy()
+    consts = {}
+    off = 0
+    for (n, d, s), nb in zip(metas, sizes):
+        consts[n] = host[off:off + nb].copy().view(np.dtype(d)).reshape(s)
+        off += nb
+    skeleton.consts = consts
+    return skeleton
+
+
 class LoopbackComm:
     """In-process world_size=1 fake with the same interface (CPU tests of
     the replica manager / sharding math without a process group)."""

@@ -79,8 +79,19 @@ def worker_main(rank: int, world: int, cfg: Config, backend: str, port: int,
         dist.init_process_group(backend=backend, rank=rank, world_size=world)
         comm = TorchDistComm(rank, world, device)
 
-        from ..engine.service import LocalEngine
-        eng = LocalEngine(cfg.replace(device=str(device)))
+        from ..engine.service import LocalEngine, build_graph
+        bdev = device if backend == "nccl" else torch.device("cpu")
+        if world > 1:
+            # rank0 reads the checkpoint ONCE; weights fan out over
+            # RCCL/xGMI (SURVEY §2b "model load") — never N disk loads,
+            # and random-init models are identical across ranks by
+            # construction, not by seed discipline.
+            from .comm import broadcast_graph
+            g = build_graph(cfg) if rank == 0 else None
+            g = broadcast_graph(g, rank, bdev, dist)
+        else:
+            g = build_graph(cfg)
+        eng = LocalEngine(cfg.replace(device=str(device)), graph=g)
 
         h, w, c = cfg.input_height, cfg.input_width, cfg.input_channels
         nc = cfg.num_classes
