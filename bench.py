@@ -8,6 +8,11 @@ images, random-init weights, through the native HIP-kernel engine
   python bench.py --gpus N --steps K --warmup W
 For N>1 the driver launches via torch.distributed.run (one rank per GPU,
 RCCL); rank 0 prints ONE JSON line.
+
+`--device cpu` is a rehearsal mode (gloo backend, fp32 CPU executor) so
+the torchrun rendezvous / barrier / all-reduce path can be exercised at
+world>1 on machines with no GPU; it changes the backend, never the
+measurement protocol.
 """
 import argparse
 import json
@@ -18,14 +23,23 @@ import time
 ROOT = os.path.dirname(os.path.abspath(__file__))
 sys.path.insert(0, ROOT)
 
+# default steps sized so the 1-GPU ResNet-50 timed region is >= 1 s
+# (~2.1 ms/step measured round 1) — a region the driver's SMI sampler
+# can actually see (VERDICT round-1 weak #6).
+DEFAULT_STEPS = 500
+DEFAULT_WARMUP = 20
+
 
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=20)
-    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--steps", type=int, default=DEFAULT_STEPS)
+    p.add_argument("--warmup", type=int, default=DEFAULT_WARMUP)
     p.add_argument("--batch", type=int, default=64)
     p.add_argument("--model", default="resnet50")
+    p.add_argument("--device", default="cuda",
+                   choices=["cuda", "cpu"],
+                   help="cpu = gloo rehearsal mode (no GPU needed)")
     args = p.parse_args()
 
     import numpy as np
@@ -34,30 +48,44 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    on_gpu = args.device == "cuda"
     dist = None
     if world > 1:
         import torch.distributed as dist_mod
         dist = dist_mod
-        torch.cuda.set_device(local_rank)
-        dist.init_process_group(backend="nccl")  # RCCL on ROCm
+        if on_gpu:
+            torch.cuda.set_device(local_rank)
+        # nccl IS RCCL on ROCm; gloo for the CPU rehearsal
+        dist.init_process_group(backend="nccl" if on_gpu else "gloo")
 
-    assert torch.cuda.is_available(), "bench needs a GPU"
-    device = f"cuda:{local_rank}"
-    torch.cuda.set_device(device)
+    if on_gpu:
+        assert torch.cuda.is_available(), "bench needs a GPU"
+        device = f"cuda:{local_rank}"
+        torch.cuda.set_device(device)
+    else:
+        device = "cpu"
 
     from tensorflow_web_deploy_amd.conf.config import Config
     from tensorflow_web_deploy_amd.engine.service import LocalEngine
-    from tensorflow_web_deploy_amd.ops.native import require_native_on_gpu
-    require_native_on_gpu()
+    if on_gpu:
+        from tensorflow_web_deploy_amd.ops.native import \
+            require_native_on_gpu
+        require_native_on_gpu()
 
-    size = 299 if args.model == "inception_v3" else 224
+    shapes = {"lenet": (28, 1, 10)}
+    size, chans, ncls = shapes.get(
+        args.model, (299 if args.model == "inception_v3" else 224, 3, 1000))
     cfg = Config(model_name=args.model, device=device, input_height=size,
-                 input_width=size, input_channels=3, num_classes=1000)
+                 input_width=size, input_channels=chans, num_classes=ncls)
     eng = LocalEngine(cfg)
+
+    def sync():
+        if on_gpu:
+            torch.cuda.synchronize()
 
     # synthetic images of the benchmark shape (no network: BASELINE.json)
     rng = np.random.default_rng(1234 + rank)
-    x = rng.random((args.batch, size, size, 3), dtype=np.float32)
+    x = rng.random((args.batch, size, size, chans), dtype=np.float32)
     xt = torch.from_numpy(x).to(device)
 
     feeds = {cfg.input_name: xt}
@@ -65,23 +93,24 @@ def main():
     # warmup (also compiles + captures the HIP graph)
     for _ in range(args.warmup):
         eng.ex.run(feeds)
-    torch.cuda.synchronize()
+    sync()
     if dist:
         dist.barrier()
-        torch.cuda.synchronize()
+        sync()
 
     t0 = time.perf_counter()
     for _ in range(args.steps):
         eng.ex.run(feeds)
-    torch.cuda.synchronize()
+    sync()
     if dist:
         dist.barrier()
-        torch.cuda.synchronize()
+        sync()
     t1 = time.perf_counter()
 
     elapsed = t1 - t0
     if dist:
-        t = torch.tensor([elapsed], device=device)
+        t = torch.tensor([elapsed],
+                         device=device if on_gpu else "cpu")
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
@@ -101,7 +130,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16",
+            "dtype": "bf16" if on_gpu else "fp32",
             "data": "synthetic",
             "config": {
                 "model": args.model,
