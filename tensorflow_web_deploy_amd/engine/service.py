@@ -159,6 +159,17 @@ class GpuServingEngine(LocalEngine):
         return out.float().cpu().numpy()[:n]
 
     @torch.inference_mode()
+    def infer_uploaded_topk(self, handle, k: int):
+        """Forward + top-k on the GPU: the D2H is 2x[n,k] instead of
+        [n,classes] and the host never runs an argpartition."""
+        batch, ev, n = handle
+        torch.cuda.current_stream().wait_event(ev)
+        out = self.ex.run({self.cfg.input_name: batch})[0]
+        v, i = torch.topk(out.float()[:n], min(k, out.shape[-1]), dim=1)
+        return (np.ascontiguousarray(i.to(torch.int32).cpu().numpy()),
+                np.ascontiguousarray(v.cpu().numpy()))
+
+    @torch.inference_mode()
     def infer_decoded(self, decoded, n: int) -> np.ndarray:
         """GPU stage of the pipelined fast server: slot window ->
         H2D + resize (side stream) -> model -> probs. The CPU decode

@@ -204,7 +204,8 @@ class ShmDecodePool:
 
     def __init__(self, channels: int, workers: Optional[int] = None,
                  slot_bytes: int = 1 << 19, slots: int = 256,
-                 in_bytes: int = 64 << 20, jpeg: bool = False):
+                 in_bytes: int = 64 << 20, jpeg: bool = False,
+                 phases: int = 2):
         import multiprocessing as mp
         import os
         import tempfile
@@ -215,10 +216,12 @@ class ShmDecodePool:
         self.in_bytes = in_bytes
         self.hdr = slots * 32
         total = self.hdr + in_bytes + slots * slot_bytes
-        # slot area is double-buffered: a pipelined server decodes
-        # batch N+1 into one half while batch N's half is still being
-        # DMA'd / consumed by the GPU stage
-        self.half = (slots - 2) // 2
+        # slot area is N-way buffered: a pipelined server decodes batch
+        # N+k into a free window while batch N's window is still being
+        # DMA'd / consumed by the GPU stage (window tokens gate reuse;
+        # >2 phases keep the decode stage from stalling on the forward)
+        self.phases = max(2, phases)
+        self.half = (slots - 2) // self.phases  # slots per window
         self._phase = 0
         self.last_window_off = 0  # byte offset of the active half
         fd, self.path = tempfile.mkstemp(prefix="twd_dec_",
@@ -280,7 +283,7 @@ class ShmDecodePool:
             return None
         if phase is None:
             phase = self._phase
-            self._phase ^= 1
+            self._phase = (self._phase + 1) % self.phases
         self.last_window_off = phase * self.half * self.slot_bytes
         self._tasks[0] = n
         self._tasks[1] = phase * self.half  # slot index base for workers
@@ -383,7 +386,11 @@ class GpuPreprocessor:
         self._shm_t = None
         self._shm_pinned = False
         self._planes = None  # GPU scratch for the jpeg IDCT kernels
-        self._out_b = None   # second output batch (pipeline phase 1)
+        import os as _os
+        # pipeline depth: decode windows + per-phase GPU output batches
+        # (3 = decode can run two batches ahead of the forward)
+        self.phases = max(2, int(_os.environ.get("TWD_PIPE_PHASES", "3")))
+        self._out_phases = [self._out]  # lazily extended per phase
 
     def _decode_cpu(self, data: bytes) -> np.ndarray:
         from PIL import Image
@@ -427,7 +434,9 @@ class GpuPreprocessor:
             self._shm_pool = ShmDecodePool(
                 self.c, workers=int(nw) if nw else None,
                 slot_bytes=1 << 19,
-                slots=max(258, 2 * self.max_batch + 2), jpeg=jpeg)
+                slots=max(self.phases * 129 + 2,
+                          self.phases * self.max_batch + 2),
+                jpeg=jpeg, phases=self.phases)
             self._shm_t = torch.from_numpy(self._shm_pool.buf)
             self._shm_pinned = bool(self.E.host_register(
                 self._shm_t.data_ptr(), self._shm_t.numel()))
@@ -444,14 +453,14 @@ class GpuPreprocessor:
         return meta, pool.last_window_off, pool.last_flags
 
     def _out_buf(self, phase):
-        """Double-buffered output batches: the pipelined server uploads
-        batch N+1 (side stream) while batch N's forward still reads its
+        """Per-phase output batches: the pipelined server uploads batch
+        N+k (side stream) while batch N's forward still reads its own
         output tensor; phase tokens gate reuse."""
-        if phase is None or phase == 0:
-            return self._out
-        if self._out_b is None:
-            self._out_b = torch.empty_like(self._out)
-        return self._out_b
+        if phase is None:
+            phase = 0
+        while len(self._out_phases) <= phase:
+            self._out_phases.append(torch.empty_like(self._out))
+        return self._out_phases[phase]
 
     def upload_shm(self, decoded, n: int,
                    compute_stream: Optional[torch.cuda.Stream] = None,

@@ -223,9 +223,13 @@ class BlobPre:
 
         from ..ingest.gpu_pipeline import ShmDecodePool
         nw = os.environ.get("TWD_DECODE_WORKERS")
+        self.phases = max(2, int(os.environ.get("TWD_PIPE_PHASES", "3")))
         self.pool = ShmDecodePool(
             cfg.input_channels, workers=int(nw) if nw else None,
-            slot_bytes=1 << 19, slots=max(258, 2 * cfg.max_batch + 2))
+            slot_bytes=1 << 19,
+            slots=max(self.phases * 129 + 2,
+                      self.phases * cfg.max_batch + 2),
+            phases=self.phases)
 
     def decode_shm(self, blobs, phase=None):
         meta = self.pool.decode_batch(blobs, phase=phase)

@@ -154,9 +154,12 @@ class FastServer:
         import time as _t
         q = queue.Queue()
         free_q = queue.Queue()
-        free_q.put(0)
-        free_q.put(1)
         eng = self.engine
+        # one token per decode window (N-phase pipeline: decode runs up
+        # to N-1 batches ahead of the forward)
+        nph = getattr(getattr(eng, "pre", None), "phases", 2)
+        for _ph in range(nph):
+            free_q.put(_ph)
         stats = ({"wait": 0.0, "decode": 0.0, "token": 0.0, "gpu": 0.0,
                   "json": 0.0, "respond": 0.0, "batches": 0, "imgs": 0}
                  if os.environ.get("TWD_FAST_STATS") else None)
@@ -220,11 +223,17 @@ class FastServer:
                 item = respq.get()
                 if item is None:
                     return
-                batch, probs, results = item
+                batch, kind, payload = item
                 t0 = _t.perf_counter()
-                if results is None:
-                    # top-k in numpy, JSON + enqueue entirely in C++
-                    top, vals = self._topk(probs)
+                if kind == "topk":
+                    # top-k already computed on the GPU; JSON + enqueue
+                    # entirely in C++
+                    top, vals = payload
+                    t1 = _t.perf_counter()
+                    self.fe.respond_topk([rid for (rid, _) in batch],
+                                         top, vals)
+                elif kind == "probs":
+                    top, vals = self._topk(payload)
                     t1 = _t.perf_counter()
                     self.fe.respond_topk([rid for (rid, _) in batch],
                                          top, vals)
@@ -232,11 +241,13 @@ class FastServer:
                     t1 = _t.perf_counter()
                     self.fe.respond_many([
                         (rid, st, body.encode())
-                        for (rid, _), (st, body) in zip(batch, results)
+                        for (rid, _), (st, body) in zip(batch, payload)
                     ])
                 if stats is not None:
                     stats["json"] += t1 - t0
                     stats["respond"] += _t.perf_counter() - t1
+
+        k = min(self.cfg.top_k, self.cfg.num_classes)
 
         def gpu_loop(ex, stream):
             import torch
@@ -256,15 +267,28 @@ class FastServer:
                                 stream.wait_event(ev)
                                 out = ex.run(
                                     {self.cfg.input_name: bt})[0]
-                                probs = out.float().cpu().numpy()[:n]
+                                # top-k ON the GPU: D2H shrinks from
+                                # [n,classes] to 2x[n,k] and the numpy
+                                # top-k leaves the responder stage
+                                v, i = torch.topk(out.float()[:n], k, 1)
+                                payload = ("topk",
+                                           (i.to(torch.int32)
+                                            .cpu().numpy(),
+                                            v.cpu().numpy()))
+                        elif (handle is not None
+                              and hasattr(eng, "infer_uploaded_topk")):
+                            payload = ("topk",
+                                       eng.infer_uploaded_topk(handle, k))
                         elif handle is not None:
-                            probs = eng.infer_uploaded(handle)
+                            payload = ("probs",
+                                       eng.infer_uploaded(handle))
                         else:
-                            probs = eng.infer_decoded(decoded,
-                                                      len(bodies))
+                            payload = ("probs",
+                                       eng.infer_decoded(decoded,
+                                                         len(bodies)))
                         if stats is not None:
                             stats["gpu"] += _t.perf_counter() - t0
-                        respq.put((batch, probs, None))
+                        respq.put((batch, payload[0], payload[1]))
                         continue
                     except Exception:
                         with self._fb_lock:
@@ -274,7 +298,7 @@ class FastServer:
                 else:
                     with self._fb_lock:
                         results = self._infer_isolating(bodies)
-                respq.put((batch, None, results))
+                respq.put((batch, "results", results))
 
         t = threading.Thread(target=ingest, daemon=True)
         t.start()
