@@ -129,3 +129,61 @@ def test_graphdef_roundtrip_random(seed):
         if "value" in a.attr:
             np.testing.assert_array_equal(b.attr["value"].tensor,
                                           a.attr["value"].tensor)
+
+
+@given(st.binary(max_size=64))
+@settings(max_examples=200, deadline=None)
+def test_ordered_code_decode_never_hangs(data):
+    """Slice-key decoders on arbitrary bytes: parse or raise ValueError
+    (never hang / never index out of range)."""
+    from tensorflow_web_deploy_amd.ckpt import ordered_code as oc
+    for fn in (oc.read_signed_num_increasing, oc.read_num_increasing,
+               oc.read_string):
+        try:
+            fn(data, 0)
+        except (ValueError, IndexError):
+            pass
+    try:
+        oc.decode_tensor_name_slice(data)
+    except (ValueError, IndexError, UnicodeDecodeError):
+        pass
+
+
+@given(st.integers(min_value=0, max_value=2**32), st.integers(0, 100))
+@settings(max_examples=100, deadline=None)
+def test_sliced_bundle_truncation_rejected(seed, cut_pct):
+    """Truncating either bundle file of a sliced checkpoint must raise,
+    not return silently-wrong weights."""
+    import struct
+    import tempfile
+
+    from tensorflow_web_deploy_amd.ckpt.tensor_bundle import (BundleReader,
+                                                              write_bundle)
+    rng = np.random.default_rng(seed)
+    full = rng.standard_normal((6, 4)).astype(np.float32)
+    with tempfile.TemporaryDirectory() as td:
+        prefix = td + "/variables"
+        write_bundle(prefix, {},
+                     sliced={"w": ([6, 4], [
+                         ([(0, 3), (-1, -1)], full[:3]),
+                         ([(3, 3), (-1, -1)], full[3:]),
+                     ])})
+        for suffix in (".index", ".data-00000-of-00001"):
+            path = prefix + suffix
+            with open(path, "rb") as fh:
+                blob = fh.read()
+            cut = len(blob) * cut_pct // 100
+            if cut == len(blob):
+                continue
+            with open(path, "wb") as fh:
+                fh.write(blob[:cut])
+            try:
+                r = BundleReader(prefix)
+                got = r.get("w")
+                # ok only if the cut did not reach this tensor's bytes
+                np.testing.assert_array_equal(got, full)
+            except (ValueError, KeyError, EOFError, struct.error,
+                    FileNotFoundError, IndexError):
+                pass
+            with open(path, "wb") as fh:
+                fh.write(blob)
