@@ -732,7 +732,13 @@ class GpuExecutor:
                     f"GPU path: op {op} (node {node.name})")
 
         outs = [bufs[o] for o in g.outputs]
-        return _Plan(x_f32, steps, outs, self.use_graphs)
+        plan = _Plan(x_f32, steps, outs, self.use_graphs)
+        # per-node buffers kept for the layer-by-layer numerics harness
+        # (engine/layercheck.py): every node owns its tensor, so after a
+        # run each holds that node's value
+        plan.bufs = bufs
+        plan.shapes = shapes
+        return plan
 
     @torch.inference_mode()
     def run(self, feeds: Dict[str, torch.Tensor],
