@@ -105,7 +105,7 @@ def test_blob_mode_world2_matches_world1():
 
     blobs = [png(40, 40), png(28, 28), png(33, 17)]
     outs = []
-    for world in (1, 2):
+    for world in (1, 2, 4):
         eng = ReplicatedEngine(cfg, world=world, backend="gloo",
                                serve_blobs=True, start_timeout=120.0)
         try:
@@ -121,7 +121,10 @@ def test_blob_mode_world2_matches_world1():
             assert q.shape == (4, 10)
         finally:
             eng.shutdown()
+    # world=4 with n=3 is RAGGED (rank 3 gets an empty shard): all
+    # worlds must agree bit-for-bit
     np.testing.assert_allclose(outs[0], outs[1], atol=1e-5)
+    np.testing.assert_allclose(outs[0], outs[2], atol=1e-5)
 
 
 @pytest.mark.gpu
