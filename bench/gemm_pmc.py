@@ -20,6 +20,13 @@ if args.shape == "square":
     C = torch.zeros(n, n, dtype=torch.bfloat16, device="cuda")
     for _ in range(args.iters):
         E.gemm_bf16(A, B, C, None, None, n, False)
+elif args.shape == "8p":
+    n = args.n
+    A = torch.randn(n, n, device="cuda").bfloat16()
+    B = torch.randn(n, n, device="cuda").bfloat16()
+    C = torch.zeros(n, n, dtype=torch.bfloat16, device="cuda")
+    for _ in range(args.iters):
+        E.gemm8p_bf16(A, B, C, None, None, n, 0)
 elif args.shape == "ln":
     X = torch.randn(12544, 768, device="cuda").bfloat16()
     g = torch.rand(768, device="cuda")

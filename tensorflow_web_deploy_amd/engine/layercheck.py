@@ -49,8 +49,8 @@ DEFAULT_BUDGET = 0.02
 
 class LayerReport:
     def __init__(self):
-        self.rows: List[Tuple[str, str, int, float, float]] = []
-        # (name, op, depth, rel_err, budget)
+        # (name, op, depth, local_err, budget, global_err)
+        self.rows: List[Tuple[str, str, int, float, float, float]] = []
 
     def worst(self, k: int = 10):
         return sorted(self.rows, key=lambda r: r[3] / r[4],
@@ -61,10 +61,11 @@ class LayerReport:
 
     def summary(self) -> str:
         out = [f"{len(self.rows)} nodes compared; "
-               f"{len(self.violations())} over budget"]
-        for name, op, depth, err, budget in self.worst(8):
-            out.append(f"  {err:8.5f} (budget {budget:.4f}) "
-                       f"d{depth:<3} {op:<16} {name}")
+               f"{len(self.violations())} over budget "
+               f"(local-err gate; global shown for context)"]
+        for name, op, depth, err, budget, gerr in self.worst(8):
+            out.append(f"  local {err:8.5f} (budget {budget:.4f}) "
+                       f"global {gerr:8.5f} d{depth:<3} {op:<16} {name}")
         return "\n".join(out)
 
 
@@ -101,27 +102,76 @@ def per_layer_compare(graph, cfg, x: np.ndarray, device: str = "cuda",
             d += 1
         depth_of[n.name] = d
 
-    for name, ref in zip(want, refs):
-        got_t = plan.bufs[name]
-        got = got_t.float().cpu().numpy()
-        refn = ref.float().cpu().numpy()
-        if got.size == 0:
-            continue
-        # fused-concat members may carry padded/logical shape mismatch;
-        # compare the overlapping region
-        if got.shape != refn.shape:
-            if got.size == refn.size:
-                got = got.reshape(refn.shape)
+    def _logical(name: str) -> Optional[np.ndarray]:
+        """GPU buffer -> fp32 numpy at the node's LOGICAL shape
+        (buffers may be channel/row padded)."""
+        t = plan.bufs.get(name)
+        if t is None:
+            return None
+        a = t.float().cpu().numpy()
+        sh = plan.shapes.get(name)
+        if sh is not None and tuple(a.shape) != tuple(sh):
+            if a.size == int(np.prod(sh)):
+                a = a.reshape(sh)
             else:
-                sl = tuple(slice(0, min(a, b))
-                           for a, b in zip(got.shape, refn.shape))
-                got, refn = got[sl], refn[sl]
+                a = a[tuple(slice(0, s) for s in sh)]
+        return a
+
+    from .executor import _CPU_OPS
+
+    for name, ref in zip(want, refs):
+        got = _logical(name)
+        refn = ref.float().cpu().numpy()
+        if got is None or got.size == 0:
+            continue
+        if got.shape != refn.shape:
+            sl = tuple(slice(0, min(a, b))
+                       for a, b in zip(got.shape, refn.shape))
+            got, refn = got[sl], refn[sl]
         scale = float(np.abs(refn).max()) + 1e-6
-        err = float(np.abs(got - refn).max()) / scale
-        op = node_by_name[name].op
+        gerr = float(np.abs(got - refn).max()) / scale
+
+        # LOCAL error — the gate: rerun THIS op in fp32 on the GPU's
+        # own inputs; isolates the kernel from accumulated upstream
+        # drift (which e.g. softmax amplifies exponentially)
+        node = node_by_name[name]
+        fn = _CPU_OPS.get(node.op)
+        err = gerr
+        have_local = False
+        if fn is not None:
+            ins, ok = [], True
+            for i in node.inputs:
+                if i in graph.consts:
+                    ins.append(cx.consts[i])
+                elif i in plan.bufs:
+                    a = _logical(i)
+                    ins.append(torch.from_numpy(np.ascontiguousarray(a)))
+                elif i in graph.inputs:
+                    ins.append(xt)
+                else:
+                    ok = False
+                    break
+            if ok:
+                try:
+                    lref = fn(node, ins, cx.consts)
+                    if isinstance(lref, (tuple, list)):
+                        lref = lref[0]
+                    lr = lref.float().cpu().numpy()
+                    g2, l2 = got, lr
+                    if g2.shape != l2.shape:
+                        sl = tuple(slice(0, min(a, b))
+                                   for a, b in zip(g2.shape, l2.shape))
+                        g2, l2 = g2[sl], l2[sl]
+                    lscale = float(np.abs(l2).max()) + 1e-6
+                    err = float(np.abs(g2 - l2).max()) / lscale
+                    have_local = True
+                except Exception:
+                    pass  # fall back to the global metric for this node
+        op = node.op
         depth = depth_of.get(name, 0)
         budget = budgets.get(op, DEFAULT_BUDGET)
-        if depth_scale:
+        if depth_scale and not have_local:
+            # global metric accumulates drift with depth
             budget *= float(np.sqrt(depth + 1.0))
-        report.rows.append((name, op, depth, err, budget))
+        report.rows.append((name, op, depth, err, budget, gerr))
     return report

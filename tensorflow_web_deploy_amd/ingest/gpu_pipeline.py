@@ -431,9 +431,13 @@ class GpuPreprocessor:
             import os as _os
             jpeg = (_os.environ.get("TWD_GPU_JPEG", "1") != "0"
                     and self.c == 3)
+            # slot size bounds the largest decodable image (h*w*c bytes)
+            # AND the per-batch H2D volume (n * slot_bytes): 512 KiB
+            # fits ~400x400x3 but ships 3.4x padding for 224^2 uploads
+            sb = int(os.environ.get("TWD_SLOT_BYTES", 1 << 19))
             self._shm_pool = ShmDecodePool(
                 self.c, workers=int(nw) if nw else None,
-                slot_bytes=1 << 19,
+                slot_bytes=sb,
                 slots=max(self.phases * 129 + 2,
                           self.phases * self.max_batch + 2),
                 jpeg=jpeg, phases=self.phases)
