@@ -46,6 +46,19 @@ elif args.shape == "attn":
                           k.stride(0), k.stride(2), k.stride(1),
                           v.stride(0), v.stride(2), v.stride(1),
                           o.stride(0), o.stride(2), o.stride(1), 0.125)
+elif args.shape == "inc1x7":
+    # Inception-v3 17x17 factorized conv class (~300 TF round 1)
+    X = torch.randn(64, 17, 17, 160, device="cuda").bfloat16()
+    W2 = torch.randn(160, 1152, device="cuda").bfloat16()  # K=1120 pad 1152
+    Y = torch.zeros(64, 17, 17, 160, dtype=torch.bfloat16, device="cuda")
+    for _ in range(args.iters):
+        E.conv2d_bf16(X, W2, Y, None, None, 1, 7, 1, 1, 0, 3, False)
+elif args.shape == "inc7x1":
+    X = torch.randn(64, 17, 17, 160, device="cuda").bfloat16()
+    W2 = torch.randn(160, 1152, device="cuda").bfloat16()
+    Y = torch.zeros(64, 17, 17, 160, dtype=torch.bfloat16, device="cuda")
+    for _ in range(args.iters):
+        E.conv2d_bf16(X, W2, Y, None, None, 7, 1, 1, 1, 3, 0, False)
 elif args.shape == "s13x3":
     X = torch.randn(64, 56, 56, 64, device="cuda").bfloat16()
     W2 = torch.randn(64, 576, device="cuda").bfloat16()
