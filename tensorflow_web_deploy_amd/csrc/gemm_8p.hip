@@ -1,32 +1,34 @@
-// gemm_8p.hip — 256x256 GEMM on the guide's 8-phase quadrant schedule
-// (cdna_hip_programming.md §5 "256² 8-phase template"), derived from
-// the spec (no example source available):
+// gemm_8p.hip — 256x256 GEMM, guide §5 "8-phase template" geometry, v6.
 //
-//   * 512 threads / 8 waves; LDS 128 KiB = 2 dbuf x 4 HALF-tiles
-//     (A rows 0-127 / 128-255, B cols 0-127 / 128-255), each
-//     [128][64] bf16.
-//   * Each K-tile runs as 4 phases; phase q computes C-QUADRANT
-//     Q[q] (order (0,0),(1,0),(1,1),(0,1)) for the full K=64: the
-//     phase touches ONLY A-half qr and B-half qc, so halves become
-//     consumable progressively.
-//   * Each phase stages ONE half-tile (2 glds) of tile t+1 in
-//     first-need order A0,B0,A1,B1 and gates with a counted
-//     s_waitcnt vmcnt(6) (phases 0-2; none at 3): the wait leaves
-//     exactly the 3 newest half-tiles in flight — the property the
-//     guide states as "3 half-tiles prefetched ahead".
-//   * Per phase: stage -> vmcnt -> barrier -> 12 ds_read_b128 ->
-//     lgkmcnt(0) -> setprio(1) 16 MFMA setprio(0) -> barrier.
-//   * st_16x32 LDS swizzle (byte ^= ((byte>>9)&1)<<5, i.e. col-group
-//     bit 1 flips with row bit 2), applied on the glds SOURCE address
-//     and the ds_read address — the LDS image itself stays linear per
-//     instruction (rule 21 generalized).
+// v6 insight (PMC round 2, profiles/round2_gemm_pmc.md): the v5
+// reconstruction parked 42.5% of wave cycles at 4096^3 because its
+// waves computed a 64x32 piece of EVERY C-quadrant and therefore read
+// ALL FOUR LDS half-tiles each K-tile — every wave's schedule coupled
+// to every staged half (3 vmcnt waits/tile). The template's geometry
+// (guide: "per-wave output 128x64 = BM/WARPS_M x BN/WARPS_N", waves
+// 2Mx4N) gives each wave a CONTIGUOUS 128x64 block, so a wave touches
+// exactly ONE A-half (its wr) and ONE B-half (wc>>1):
 //
-// Scope: MODE=linear A (dense/1x1), bias + act + optional residual,
-// LDS-restage epilogue. UNROUTED: measured faster than
-// gemm_conv_kernel_256 only at 8192^3 (1006 vs 954 TF) and slower at
-// 4096^3 (875 vs 1130) and on model shapes — NOTES_ROUND2.md item 2
-// records the five-variant exploration. bench/gemm8p_bench.py
-// reproduces numerics + perf via the gemm8p_bf16 binding.
+//   * 512 thr / 8 waves; LDS 128 KiB = 2 dbuf x 4 half-tiles
+//     (A rows 0-127/128-255, B cols 0-127/128-255), each [128][64] bf16.
+//   * Wave (wr=w>>2, wc=w&3) owns C rows [wr*128,+128) cols [wc*64,+64):
+//     acc[8][4] f32x4 (128 VGPR).
+//   * K-tile = 4 phases, quadrant order (m,n) = (0,0),(1,0),(1,1),(0,1)
+//     over the wave's OWN block: 16 MFMA each; ds_reads 12/8/4/0 with
+//     A-set + B-pair register reuse (24 ds_read_b128 per tile).
+//   * ONE stage per phase (2 glds: halves of tile t+1 in order
+//     A0,B0,A1,B1) and ONE counted wait per K-tile: after staging
+//     A0(t+1) at phase 0, vmcnt(2) proves all four halves of tile t
+//     and leaves the new stage in flight. Never vmcnt(0) in the loop.
+//   * Phase shape per guide: [reads] stage [wait P0] barrier
+//     lgkmcnt(0) setprio(1) 16xMFMA setprio(0) barrier. P0's reads sit
+//     AFTER its barrier (they are only proven by P0's own vmcnt).
+//   * st_16x32 LDS swizzle on the glds SOURCE + ds_read address.
+//
+// Scope: linear A (dense/1x1 shapes), bias + act + optional residual,
+// LDS-restage epilogue. Routed for large square-ish shapes when it
+// beats gemm_conv_kernel_256 (see dispatch in gemm_conv.hip / bench
+// history in NOTES).
 #include <hip/hip_bf16.h>
 #include <hip/hip_runtime.h>
 
@@ -74,20 +76,11 @@ __global__ __launch_bounds__(512) void gemm_8p_kernel(
   const auto rsrcB = __builtin_amdgcn_make_buffer_rsrc(
       (void*)B, 0, (int)(((int64_t)(N - 1) * Kp + Kp) * 2), 0x00020000);
 
-  // staging geometry: half = [128 rows][64 elems]; instruction j covers
-  // rows j*64 + T/8 (T = global thread id), col-group T%8 (8 elems).
-  // st_16x32 swizzle on the SOURCE: colgrp_src = colgrp ^ ((row>>2)&1)<<1
+  // staging geometry: half = [128 rows][64 elems]; glds j covers rows
+  // j*64 + T/8, col-group T%8; st_16x32 swizzle applied on the SOURCE
+  // so the LDS image is "swizzled-linear" and ds_reads un-swizzle.
   const int T = tid;
   const int s_colgrp = T & 7;
-  // stage half h of K-tile t into buffer t&1 (t may be nt: parity
-  // still alternates, k reads clamp to the last tile — harmless
-  // redundant loads on the final iteration). h: 0=A0 1=A1 2=B0 3=B1.
-  //
-  // glds writes lane l's 16 B at LDS base + l*16 (wave-uniform base);
-  // for a [row][64-elem] half, (j*64 + w*8 + l/8)*128 + (l%8)*16 ==
-  // j*8192 + w*1024 + l*16, so base = half + j*8192 + wave*1024 gives
-  // the LINEAR image and the st_16x32 swizzle lives purely in the
-  // per-lane SOURCE address.
   auto stage_half = [&](int h, int t, int nt_) {
     const int kb = (t < nt_ ? t : nt_ - 1) * BK8;
     const int buf = t & 1;
@@ -116,68 +109,54 @@ __global__ __launch_bounds__(512) void gemm_8p_kernel(
     return row * 128 + ((cg ^ (((row >> 2) & 1) << 1)) << 4);
   };
 
-  // quadrant order and wave sub-tile: within quadrant (128x128), wave
-  // covers rows (w>>2)*64, cols (w&3)*32 -> 4 m-frags x 2 n-frags
-  const int wr2 = wave >> 2;   // 0..1
-  const int wc2 = wave & 3;    // 0..3
-  const int QR[4] = {0, 1, 1, 0};
-  const int QC[4] = {0, 0, 1, 1};
-  const int SO[4] = {0, 2, 1, 3};  // stage order A0,B0,A1,B1
+  const int wr = wave >> 2;        // A-half this wave reads
+  const int wc = wave & 3;
+  const int bh = wc >> 1;          // B-half this wave reads
+  const int bcol0 = (wc & 1) * 64; // col strip within that half
 
-  f32x4_8p acc[4][4][2];  // [quadrant][mi][ni]
+  f32x4_8p acc[8][4];
 #pragma unroll
-  for (int q = 0; q < 4; ++q)
+  for (int mi = 0; mi < 8; ++mi)
 #pragma unroll
-    for (int mi = 0; mi < 4; ++mi)
-#pragma unroll
-      for (int ni = 0; ni < 2; ++ni) {
-        float b0 = 0.f;
-        if (bias) {
-          int col = col0 + QC[q] * 128 + wc2 * 32 + ni * 16 + (lane & 15);
-          b0 = col < N ? bias[col] : 0.f;
-        }
-        acc[q][mi][ni] = f32x4_8p{b0, b0, b0, b0};
+    for (int ni = 0; ni < 4; ++ni) {
+      float b0 = 0.f;
+      if (bias) {
+        int col = col0 + wc * 64 + ni * 16 + (lane & 15);
+        b0 = col < N ? bias[col] : 0.f;
       }
+      acc[mi][ni] = f32x4_8p{b0, b0, b0, b0};
+    }
 
   const int nt = Kp / BK8;
-  // prologue: all four halves of tile 0
-  stage_half(0, 0, nt);
-  stage_half(2, 0, nt);
-  stage_half(1, 0, nt);
-  stage_half(3, 0, nt);
 
-  // Fragment reuse across the quadrant order (0,0),(1,0),(1,1),(0,1):
-  // q0 and q3 use the SAME A-frags (half 0, same wave rows), q1/q2
-  // share half 1; q0/q1 share B-pair 0, q2/q3 pair 1. So each A set
-  // and B pair is read ONCE per K-tile (24 ds_reads, not 40) and held
-  // in registers. A sets are issued where the previous phase's MFMA
-  // burst hides them; B pairs are the only exposed reads (4 each).
-  bf16x8_8p af[2][4][2], bf[2][2];  // af[set][mi][kk], bf[ni][kk]
+  // register frags: two A m-sets (rows mi*16, m-set s covers mi s*4..+4)
+  // and one live B n-pair (cols ni*16, pair p covers ni p*2..+2)
+  bf16x8_8p af[2][4][2];  // [m-set][mi'][kk]
+  bf16x8_8p bf[2][2];     // [ni'][kk] — current pair only
 
-  auto issue_a = [&](int set, int buf) {
-    const char* Ah = smem8p + HALF_OFF(buf, set);
+  auto read_a = [&](int set, int buf) {
+    const char* Ah = smem8p + HALF_OFF(buf, wr);
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi) {
-        int row = wr2 * 64 + mi * 16 + (lane & 15);
+        int row = (set * 4 + mi) * 16 + (lane & 15);
         af[set][mi][kk] = *(const bf16x8_8p*)(
             Ah + lds_b(row, kk * 4 + (lane >> 4)));
       }
   };
-  auto issue_b = [&](int pair, int buf) {
-    const char* Bh = smem8p + HALF_OFF(buf, 2 + pair);
+  auto read_b = [&](int pair, int buf) {
+    const char* Bh = smem8p + HALF_OFF(buf, 2 + bh);
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
       for (int ni = 0; ni < 2; ++ni) {
-        int row = wc2 * 32 + ni * 16 + (lane & 15);
+        int row = bcol0 + (pair * 2 + ni) * 16 + (lane & 15);
         bf[ni][kk] = *(const bf16x8_8p*)(
             Bh + lds_b(row, kk * 4 + (lane >> 4)));
       }
   };
-
-  auto mfma_q = [&](int q, int set) {
+  auto mfma_q = [&](int mset, int npair) {
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk)
@@ -185,97 +164,85 @@ __global__ __launch_bounds__(512) void gemm_8p_kernel(
       for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
         for (int ni = 0; ni < 2; ++ni)
-          acc[q][mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[set][mi][kk], bf[ni][kk], acc[q][mi][ni], 0, 0, 0);
+          acc[mset * 4 + mi][npair * 2 + ni] =
+              __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  af[mset][mi][kk], bf[ni][kk],
+                  acc[mset * 4 + mi][npair * 2 + ni], 0, 0, 0);
     __builtin_amdgcn_s_setprio(0);
   };
 
-  // prologue: verify tile0 A0/B0, sync, pre-read A set 0
-  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  // prologue: all four halves of tile 0; drain fully once
+  stage_half(0, 0, nt);
+  stage_half(2, 0, nt);
+  stage_half(1, 0, nt);
+  stage_half(3, 0, nt);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
-  issue_a(0, 0);
 
-  // Waits as derived for the v3 schedule (sim in git history):
-  //   q0: vmcnt(2) pre-barrier (t:A1 proven for q1); stage t+1:A0
-  //       AFTER the barrier (its next-tile counterpart would overwrite
-  //       the half the slowest wave still reads).
-  //   q1: stage t+1:B0; vmcnt(4) (t:B1 proven for q2).
-  //   q2: stage t+1:A1; no wait.
-  //   q3: stage t+1:B1; vmcnt(4) (t+1:A0 proven -> pre-read next A0
-  //       from the other buffer under q3's MFMAs).
   for (int t = 0; t < nt; ++t) {
     const int buf = t & 1;
-    // ---- q0 ----
+    // ---- phase 0: quadrant (m0, n0) ----
+    stage_half(0, t + 1, nt);
+    // ONE counted wait per K-tile: proves all four halves of tile t,
+    // leaves the stage just issued (2 glds) in flight
     asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
     __builtin_amdgcn_s_barrier();
-    stage_half(0, t + 1, nt);
-    issue_b(0, buf);
+    read_a(0, buf);
+    read_b(0, buf);
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    issue_a(1, buf);  // half 1 for q1/q2, hidden under q0's MFMAs
     mfma_q(0, 0);
     __builtin_amdgcn_s_barrier();
-    // ---- q1 ----
+    // ---- phase 1: (m1, n0) ----
+    read_a(1, buf);
     stage_half(2, t + 1, nt);
-    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    mfma_q(1, 0);
+    __builtin_amdgcn_s_barrier();
+    // ---- phase 2: (m1, n1) ----
+    read_b(1, buf);
+    stage_half(1, t + 1, nt);
     __builtin_amdgcn_s_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     mfma_q(1, 1);
     __builtin_amdgcn_s_barrier();
-    // ---- q2 ----
-    stage_half(1, t + 1, nt);
-    __builtin_amdgcn_s_barrier();
-    issue_b(1, buf);
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    mfma_q(2, 1);
-    __builtin_amdgcn_s_barrier();
-    // ---- q3 ----
+    // ---- phase 3: (m0, n1) ---- (no reads: af[0] + bf[1] both live)
     stage_half(3, t + 1, nt);
-    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
     __builtin_amdgcn_s_barrier();
-    mfma_q(3, 0);
-    // next tile's A0 AFTER q3's MFMAs consume set 0 (prefetching it
-    // above would create a register WAR the scheduler resolves by
-    // serializing — wrong data cannot occur, but the MFMAs would wait
-    // on next-tile loads; a 3rd A set doesn't fit in 256 VGPRs)
-    issue_a(0, buf ^ 1);
+    mfma_q(0, 1);
     __builtin_amdgcn_s_barrier();
   }
   asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
-  // ---- epilogue: restage C through LDS for wide dwordx4 stores ----
-  // (the scattered per-element form measured 888 vs 1010 TF at 4k/8k:
-  // the 2-byte stores cost ~2x the loop's own margin at nt=64)
-  asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
+
+  // ---- epilogue: restage C through LDS for wide dwordx4 stores ----
   {
     __bf16* ct = (__bf16*)smem8p;  // [256][256] bf16 = 128 KiB
+    const int rb = wr * 128;
+    const int cb = wc * 64;
 #pragma unroll
-    for (int q = 0; q < 4; ++q) {
-      const int rb = QR[q] * 128 + wr2 * 64;
-      const int cb = QC[q] * 128 + wc2 * 32;
+    for (int mi = 0; mi < 8; ++mi)
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
+      for (int ni = 0; ni < 4; ++ni) {
+        int col = cb + ni * 16 + (lane & 15);
 #pragma unroll
-        for (int ni = 0; ni < 2; ++ni) {
-          int col = cb + ni * 16 + (lane & 15);
-#pragma unroll
-          for (int r = 0; r < 4; ++r) {
-            int row = rb + mi * 16 + (lane >> 4) * 4 + r;
-            float v = acc[q][mi][ni][r];
-            if (residual) {
-              int gr = row0 + row, gc = col0 + col;
-              if (gr < M && gc < N)
-                v += (float)residual[(int64_t)gr * ldc + gc];
-            }
-            if (act == 1)
-              v = v > 0.f ? v : 0.f;
-            else if (act == 2)
-              v = v < 0.f ? 0.f : (v > 6.f ? 6.f : v);
-            else if (act == 5)
-              v = 0.5f * v * (1.f + erff(v * 0.70710678f));
-            ct[row * 256 + col] = (__bf16)v;
+        for (int r = 0; r < 4; ++r) {
+          int row = rb + mi * 16 + (lane >> 4) * 4 + r;
+          float v = acc[mi][ni][r];
+          if (residual) {
+            int gr = row0 + row, gc = col0 + col;
+            if (gr < M && gc < N)
+              v += (float)residual[(int64_t)gr * ldc + gc];
           }
+          if (act == 1)
+            v = v > 0.f ? v : 0.f;
+          else if (act == 2)
+            v = v < 0.f ? 0.f : (v > 6.f ? 6.f : v);
+          else if (act == 5)
+            v = 0.5f * v * (1.f + erff(v * 0.70710678f));
+          ct[row * 256 + col] = (__bf16)v;
         }
-    }
+      }
     __builtin_amdgcn_s_barrier();
     const bool full = (row0 + 256 <= M) && (col0 + 256 <= N);
 #pragma unroll 1
