@@ -1164,20 +1164,16 @@ static void dispatch(const void* A, const void* B, void* C, const float* bias,
                                   a_bytes, act, cp, stream, ws, splitk, ldc);
     return;
   }
-  // mid-N (Inception 17x17 factorized convs, N=160/192): TBN=128 pads
-  // the column dim to 256 and wastes 25-37% of the MFMA work; a
-  // (128,64) tile cuts the padded-N waste (160: 96->32, 192: 64->0)
-  // at the cost of one extra A-panel re-read. A/B: TWD_N64=0 disables.
-  if (N > 96 && N <= 224) {
-    int w64 = (N + 63) / 64 * 64 - N;
-    int w128 = (N + 127) / 128 * 128 - N;
-    const char* n64 = getenv("TWD_N64");
-    if (w64 * 2 < w128 && !(n64 && n64[0] == '0')) {
-      launch_tile<128, 64, MODE, 2>(A, B, C, bias, residual, M, N, K, Kp,
-                                    a_bytes, act, cp, stream, ws, splitk,
-                                    ldc);
-      return;
-    }
+  // Mid-N (Inception 17x17, N=160/192) pad-waste routing to (128,64)
+  // was A/B-tested and REGRESSED (inc1x7 308->298 TF, Inception e2e
+  // 19.6k->19.3k img/s): the class is latency/wait-bound, not
+  // MFMA-bound, so cutting padded-N waste buys nothing while the
+  // extra A-panel re-read costs. Kept opt-in for future A/Bs.
+  if (N > 96 && N <= 224 && getenv("TWD_N64") != nullptr) {
+    launch_tile<128, 64, MODE, 2>(A, B, C, bias, residual, M, N, K, Kp,
+                                  a_bytes, act, cp, stream, ws, splitk,
+                                  ldc);
+    return;
   }
   // mid-size M underfills the chip at 128^2 (e.g. Inception 17x17
   // factorized convs: 290 blocks = ~1 block/CU); halve the M-tile to
