@@ -1076,6 +1076,12 @@ int splitk_plan(int M, int N, int K) {
   int TBM = N <= 96 ? 256 : 128;
   long blocks = ((M + TBM - 1) / TBM) * (long)((N + TBN - 1) / TBN);
   int nt = (K + BK - 1) / BK;
+  // A/B override: TWD_SPLITK=n forces the split for underfilled shapes
+  const char* fs = getenv("TWD_SPLITK");
+  if (fs && blocks < 256 && nt >= 8) {
+    int v = atoi(fs);
+    if (v > 1) return v;
+  }
   // Split only when the chip is badly underfilled: the f32 slab
   // round-trip costs ~bytes/6TBps, so moderate underfill (>=128 blocks,
   // or >=64 blocks at small K) is cheaper unsplit (measured: s3 3x3
