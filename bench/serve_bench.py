@@ -33,22 +33,68 @@ def _png_bytes(h, w, gray=False):
 def _client_proc(url, img, nthreads, per_thread_qps, duration, q):
     import threading
     import time
-    import urllib.request
+    import socket
     lat = []
     lock = threading.Lock()
     stop_at = time.perf_counter() + duration
-
-    def post_once():
-        t0 = time.perf_counter()
-        req = urllib.request.Request(url, data=img, headers={
-            "Content-Type": "application/octet-stream"})
-        with urllib.request.urlopen(req, timeout=30) as r:
-            r.read()
-        dt = time.perf_counter() - t0
-        with lock:
-            lat.append(dt)
+    # pre-serialized request over a persistent keep-alive socket: the
+    # urllib client cost ~3x this in client CPU (new TCP + header
+    # parsing per request), and the load generator shares the box with
+    # the decode workers — client cycles ARE server throughput.
+    from urllib.parse import urlparse
+    u = urlparse(url)
+    host, port = u.hostname, u.port
+    req_bytes = (f"POST {u.path} HTTP/1.1\r\nHost: {host}\r\n"
+                 f"Content-Type: application/octet-stream\r\n"
+                 f"Content-Length: {len(img)}\r\n\r\n"
+                 ).encode() + img
 
     def run():
+        sock = None
+        buf = bytearray()
+
+        def post_once():
+            nonlocal sock, buf
+            t0 = time.perf_counter()
+            if sock is None:
+                sock = socket.create_connection((host, port), timeout=30)
+                sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+                buf = bytearray()
+            try:
+                sock.sendall(req_bytes)
+                # read headers
+                while b"\r\n\r\n" not in buf:
+                    d = sock.recv(65536)
+                    if not d:
+                        raise ConnectionError("closed")
+                    buf += d
+                head, _, rest = bytes(buf).partition(b"\r\n\r\n")
+                hl = head.decode("latin-1", "replace").lower()
+                clen = 0
+                for line in hl.split("\r\n"):
+                    if line.startswith("content-length:"):
+                        clen = int(line.split(":", 1)[1])
+                body = bytearray(rest)
+                while len(body) < clen:
+                    d = sock.recv(65536)
+                    if not d:
+                        raise ConnectionError("closed")
+                    body += d
+                buf = body[clen:]
+                if "connection: close" in hl:
+                    sock.close()
+                    sock = None
+            except Exception:
+                try:
+                    sock.close()
+                except Exception:
+                    pass
+                sock = None
+                raise
+            dt = time.perf_counter() - t0
+            with lock:
+                lat.append(dt)
+
         if per_thread_qps > 0:
             interval = 1.0 / per_thread_qps
             nxt = time.perf_counter()
