@@ -46,6 +46,13 @@ elif args.shape == "attn":
                           k.stride(0), k.stride(2), k.stride(1),
                           v.stride(0), v.stride(2), v.stride(1),
                           o.stride(0), o.stride(2), o.stride(1), 0.125)
+elif args.shape == "stem":
+    # Inception conv1: 149^2 x32 -> 3x3 VALID -> 147^2 x32 (120 us, 213 TF)
+    X = torch.randn(64, 149, 149, 32, device="cuda").bfloat16()
+    W2 = torch.randn(32, 320, device="cuda").bfloat16()  # K=288 pad 320
+    Y = torch.zeros(64, 147, 147, 32, dtype=torch.bfloat16, device="cuda")
+    for _ in range(args.iters):
+        E.conv2d_bf16(X, W2, Y, None, None, 3, 3, 1, 1, 0, 0, False)
 elif args.shape == "inc1x7":
     # Inception-v3 17x17 factorized conv class (~300 TF round 1)
     X = torch.randn(64, 17, 17, 160, device="cuda").bfloat16()
