@@ -27,3 +27,15 @@ Y2 = torch.zeros_like(X2)
 ms = t(lambda: E.conv2d_bf16(X2, W3, Y2, None, None, 1, 7, 1, 1, 0, 3, 0))
 fl2 = 2*64*17*17*192*7*192
 print(f"inc1x7_c192: {ms*1000:.1f} us  {fl2/ms/1e9:.0f} TF")
+Xs = torch.randn(64, 149, 149, 32, device="cuda").bfloat16()
+Ws = torch.randn(32, 320, device="cuda").bfloat16()
+Ys = torch.zeros(64, 147, 147, 32, dtype=torch.bfloat16, device="cuda")
+ms = t(lambda: E.conv2d_bf16(Xs, Ws, Ys, None, None, 3, 3, 1, 1, 0, 0, 0))
+fls = 2*64*147*147*32*288
+print(f"stem1 3x3C32: {ms*1000:.1f} us  {fls/ms/1e9:.0f} TF")
+X4 = torch.randn(64, 35, 35, 64, device="cuda").bfloat16()
+W4 = torch.randn(96, 576, device="cuda").bfloat16()
+Y4 = torch.zeros(64, 35, 35, 96, dtype=torch.bfloat16, device="cuda")
+ms = t(lambda: E.conv2d_bf16(X4, W4, Y4, None, None, 3, 3, 1, 1, 1, 1, 0))
+fl4 = 2*64*35*35*96*576
+print(f"b35 3x3 64->96: {ms*1000:.1f} us  {fl4/ms/1e9:.0f} TF")

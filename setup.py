@@ -19,6 +19,7 @@ ext = cpp_extension.CUDAExtension(
     sources=[
         os.path.join(CSRC, "bindings.cpp"),
         os.path.join(CSRC, "gemm_conv.hip"),
+        os.path.join(CSRC, "conv_direct.hip"),
         os.path.join(CSRC, "pool_reduce.hip"),
         os.path.join(CSRC, "elementwise.hip"),
         os.path.join(CSRC, "layernorm.hip"),

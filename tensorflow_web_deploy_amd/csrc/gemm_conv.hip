@@ -1246,6 +1246,18 @@ void launch_conv2d_bf16(const void* X, const void* B, void* C,
                         int splitk, int ldc) {
   int M = Nimg * HO * WO;
   int K = R * S * Cin;
+  // direct-conv path (conv_direct.hip): im2col-free for C%32==0
+  // small/mid-channel R*S>=3 shapes (Inception stems / 35x35 / 17x17
+  // classes — VALU-issue-bound on the gather path, see profiles/).
+  // TWD_DIRECT=0 disables for A/B.
+  {
+    static const char* dis = getenv("TWD_DIRECT");
+    if (!(dis && dis[0] == '0') && !residual && splitk <= 1 &&
+        launch_conv_direct(X, B, C, bias, Nimg, H, W, Cin, R, S, Cout,
+                           sh, sw, pt, pl, HO, WO, Kp, act, stream,
+                           ldc ? ldc : Cout))
+      return;
+  }
   ConvParams cp;
   cp.C = Cin; cp.H = H; cp.W = W; cp.R = R; cp.S = S;
   cp.sh = sh; cp.sw = sw; cp.pt = pt; cp.pl = pl; cp.HO = HO; cp.WO = WO;

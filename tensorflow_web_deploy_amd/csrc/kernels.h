@@ -11,6 +11,11 @@ void launch_gemm_bf16(const void* A, const void* B, void* C,
                       int K, int Kp, int64_t a_bytes, int act,
                       hipStream_t stream, float* ws = nullptr,
                       int splitk = 1, int ldc = 0);
+bool launch_conv_direct(const void* X, const void* W2, void* Y,
+                        const float* bias, int Nimg, int H, int W, int C,
+                        int R, int S, int Cout, int sh, int sw, int pt,
+                        int pl, int HO, int WO, int Kp, int act,
+                        hipStream_t stream, int ldc);
 void launch_conv2d_bf16(const void* X, const void* B, void* C,
                         const float* bias, const void* residual, int Nimg,
                         int H, int W, int Cin, int R, int S, int Cout, int sh,

@@ -312,3 +312,51 @@ def test_dwconv_bf16(E, C, stride, act, res):
                   _bf(rr) if res else None, y, R, R, stride, stride,
                   pt, pl, act)
     torch.testing.assert_close(y.float().cpu(), ref, atol=0.1, rtol=0.05)
+
+
+@pytest.mark.parametrize("N,H,W,C,R,S,O,stride,same", [
+    (2, 21, 21, 32, 3, 3, 32, 1, False),   # inception stem conv1 class
+    (2, 21, 21, 32, 3, 3, 64, 1, True),    # stem conv2 class (SAME)
+    (2, 17, 17, 160, 1, 7, 160, 1, True),  # 17x17 1x7
+    (2, 17, 17, 160, 7, 1, 192, 1, True),  # 17x17 7x1
+    (2, 35, 35, 64, 3, 3, 96, 1, True),    # 35x35 3x3
+    (2, 23, 23, 32, 3, 3, 32, 2, True),    # s2 through the direct path
+])
+def test_conv_direct_shapes(E, N, H, W, C, R, S, O, stride, same):
+    """conv_direct.hip routing (im2col-free): these shapes satisfy
+    C%32==0 / Cout%16==0 / R*S>=3 and must match the fp32 reference
+    through the same conv2d_bf16 binding."""
+    import torch.nn.functional as F
+    torch.manual_seed(4)
+    X = torch.randn(N, H, W, C) * 0.5
+    Wt = torch.randn(R, S, C, O) * (2.0 / (R * S * C)) ** 0.5
+    xc = _bf(X).float().cpu().permute(0, 3, 1, 2)
+    wc = Wt.permute(3, 2, 0, 1)
+    if same:
+        from tensorflow_web_deploy_amd.engine.executor import tf_same_pad
+        xc = tf_same_pad(xc, R, S, stride, stride)
+    ref = F.conv2d(xc, wc, stride=stride).permute(0, 2, 3, 1).contiguous()
+
+    Xg = _bf(X)
+    K = R * S * C
+    Kp = (K + 63) // 64 * 64
+    W2 = torch.zeros(O, Kp)
+    W2[:, :K] = Wt.permute(3, 0, 1, 2).reshape(O, K)
+    W2g = _bf(W2)
+    if same:
+        from tensorflow_web_deploy_amd.engine.gpu_executor import _same_pads
+        HO, WO, pt, pl = _same_pads(H, W, R, S, stride, stride)
+    else:
+        HO, WO = (H - R) // stride + 1, (W - S) // stride + 1
+        pt = pl = 0
+    Y = torch.zeros(N, HO, WO, O, dtype=torch.bfloat16, device="cuda")
+    E.conv2d_bf16(Xg, W2g, Y, None, None, R, S, stride, stride, pt, pl,
+                  False)
+    torch.testing.assert_close(Y.float().cpu(), ref, rtol=3e-2,
+                               atol=2e-2 * K ** 0.5)
+    # bias + relu epilogue
+    bias = torch.randn(O, device="cuda")
+    E.conv2d_bf16(Xg, W2g, Y, bias, None, R, S, stride, stride, pt, pl, 1)
+    ref2 = torch.relu(ref + bias.cpu())
+    torch.testing.assert_close(Y.float().cpu(), ref2, rtol=3e-2,
+                               atol=2e-2 * K ** 0.5)
