@@ -236,8 +236,19 @@ bool launch_conv_direct(const void* X, const void* W2, void* Y,
                         int R, int S, int Cout, int sh, int sw, int pt,
                         int pl, int HO, int WO, int Kp, int act,
                         hipStream_t stream, int ldc) {
+  // Routing: measured A/B (profiles/round2_inception.md) — the direct
+  // kernel wins on BIG-MAP small-channel shapes (Inception stems,
+  // 147^2 C=32: 103 -> 77 us) where the im2col gather is VALU-bound
+  // and the weight set is tiny; it LOSES on small maps (17x17 1x7:
+  // 21 -> 47 us — B-frag streaming at 1 wave/SIMD) and mid maps
+  // (35x35: 27 -> 34 us). Keep it to the measured-win region;
+  // TWD_DIRECT=2 opens the full C%32 surface for experiments.
+  static const char* wide = getenv("TWD_DIRECT");
+  const bool open_all = wide && wide[0] == '2';
   if (C % 32 != 0 || Cout % 16 != 0 || Cout < 16 || Cout > 192 ||
       R * S < 3 || act > 2)
+    return false;
+  if (!open_all && !(C <= 64 && Cout <= 64 && WO >= 96))
     return false;
   const int NF = Cout / 16;
   // MFW from the VGPR budget: MFW * NF <= 24 and MFW <= 10
