@@ -124,32 +124,75 @@ __global__ __launch_bounds__(256) void conv_direct_kernel(
 
   const int kq = lane >> 4;       // k-quarter 0..3 (8 elems)
   const int KC = C >> 5;          // 32-wide k-steps per tap
-  bf16x8_cd bfrag[NF];
-  for (int rs = 0; rs < R * S; ++rs) {
-    const int r = rs / S, s = rs - (rs / S) * S;
-    for (int kc = 0; kc < KC; ++kc) {
-      const int kbase = (rs * C + kc * 32 + kq * 8) * 2;
-      // B-fragments: 16 couts x 32 k, streamed from global (L2-hot)
-#pragma unroll
-      for (int nf = 0; nf < NF; ++nf) {
-        int o = nf * 16 + (lane & 15);
-        bfrag[nf] = *(const bf16x8_cd*)((const char*)W2 + o * Kp * 2 +
-                                        kbase);
-      }
-#pragma unroll
-      for (int i = 0; i < MFW; ++i) {
-        // A-fragment: 16 pixels x 32 c from the LDS band (contiguous)
-        const int row = px_oy[i] * sh + r;
-        const int col = px_ox[i] * sw + s;
-        bf16x8_cd af = *(const bf16x8_cd*)(
-            smem_cd + row * rowstride + (col * C + kc * 32 + kq * 8) * 2);
-#pragma unroll
-        for (int nf = 0; nf < NF; ++nf)
-          acc[i][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af, bfrag[nf], acc[i][nf], 0, 0, 0);
-      }
-    }
+  const int T = R * S * KC;       // fused (tap, k-step) loop count
+  const int wrow = (lane & 15);   // weight row within an n-frag
+  // double-buffered B prefetch with NAMED even/odd register sets (a
+  // runtime-indexed frag array would spill to scratch — rule 20): the
+  // loads for step t+1 issue before step t's MFMA burst so the
+  // global(L2) latency hides under it — without this the kernel
+  // read-use serializes at 1 wave/SIMD on small-map shapes (measured
+  // 2x loss vs the GEMM path)
+  bf16x8_cd bfA[NF], bfB[NF];
+  // incremental (rs, kc) cursors: n = prefetch target, c = compute
+  int rs_n = 0, kc_n = 0, rs_c = 0, kc_c = 0;
+  // macro-stamped bodies so the frag arrays are always referenced by
+  // NAME (a pointer/reference parameter would make them addressable
+  // and spill to scratch)
+#define CD_LOAD_B(bf)                                                   \
+  do {                                                                  \
+    const int kbase = ((rs_n * C) + kc_n * 32 + kq * 8) * 2;            \
+    _Pragma("unroll") for (int nf = 0; nf < NF; ++nf) bf[nf] =          \
+        *(const bf16x8_cd*)((const char*)W2 +                           \
+                            (nf * 16 + wrow) * Kp * 2 + kbase);         \
+    if (++kc_n == KC) {                                                 \
+      kc_n = 0;                                                         \
+      ++rs_n;                                                           \
+    }                                                                   \
+  } while (0)
+
+#define CD_COMPUTE(bf)                                                  \
+  do {                                                                  \
+    const int r = rs_c / S, s = rs_c - (rs_c / S) * S;                  \
+    const int abase = (kc_c * 32 + kq * 8) * 2;                         \
+    bf16x8_cd af0, af1;                                                 \
+    {                                                                   \
+      const int row = px_oy[0] * sh + r;                                \
+      const int col = px_ox[0] * sw + s;                                \
+      af0 = *(const bf16x8_cd*)(smem_cd + row * rowstride +             \
+                                col * C * 2 + abase);                   \
+    }                                                                   \
+    _Pragma("unroll") for (int i = 0; i < MFW; ++i) {                   \
+      if (i + 1 < MFW) {                                                \
+        const int row = px_oy[i + 1] * sh + r;                          \
+        const int col = px_ox[i + 1] * sw + s;                          \
+        if ((i & 1) == 0)                                               \
+          af1 = *(const bf16x8_cd*)(smem_cd + row * rowstride +         \
+                                    col * C * 2 + abase);               \
+        else                                                            \
+          af0 = *(const bf16x8_cd*)(smem_cd + row * rowstride +         \
+                                    col * C * 2 + abase);               \
+      }                                                                 \
+      _Pragma("unroll") for (int nf = 0; nf < NF; ++nf) acc[i][nf] =    \
+          __builtin_amdgcn_mfma_f32_16x16x32_bf16(                      \
+              (i & 1) ? af1 : af0, bf[nf], acc[i][nf], 0, 0, 0);        \
+    }                                                                   \
+    if (++kc_c == KC) {                                                 \
+      kc_c = 0;                                                         \
+      ++rs_c;                                                           \
+    }                                                                   \
+  } while (0)
+
+  CD_LOAD_B(bfA);
+  int t = 0;
+  for (; t + 2 <= T; t += 2) {
+    if (t + 1 < T) CD_LOAD_B(bfB);
+    CD_COMPUTE(bfA);
+    if (t + 2 < T) CD_LOAD_B(bfA);
+    CD_COMPUTE(bfB);
   }
+  if (t < T) CD_COMPUTE(bfA);  // odd tail
+#undef CD_LOAD_B
+#undef CD_COMPUTE
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();  // input band no longer needed
 
