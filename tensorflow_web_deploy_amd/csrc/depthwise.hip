@@ -94,12 +94,105 @@ __global__ __launch_bounds__(256) void dwconv_kernel(
   }
 }
 
+// stride-1 two-pixel variant: a thread computes wo0 and wo0+1, so each
+// input column loads ONCE for both (12 instead of 18 loads at 3x3 —
+// NOTES_ROUND2 item 9 follow-up). VEC-only (C % 8 == 0).
+__global__ __launch_bounds__(256) void dwconv2_kernel(
+    const __bf16* __restrict__ x, const __bf16* __restrict__ w,
+    const float* __restrict__ bias, const __bf16* __restrict__ residual,
+    __bf16* __restrict__ y, int H, int W, int C, int HO, int WO, int R,
+    int S, int pt, int pl, int act, int c8, int wo2, int total) {
+  int idx = blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= total) return;
+  int c = (idx % c8) * 8;
+  int p = idx / c8;  // flattened (n, ho, wpair)
+  int wp = p % wo2;
+  int t = p / wo2;
+  int ho = t % HO;
+  int n = t / HO;
+  const int wo0 = wp * 2;
+  const bool two = wo0 + 1 < WO;
+
+  float a0[8], a1[8];
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    float b = bias ? bias[c + e] : 0.f;
+    a0[e] = b;
+    a1[e] = b;
+  }
+
+  const int hb = ho - pt;
+  const int wb = wo0 - pl;
+  const __bf16* xn = x + ((int64_t)n * H * W * C) + c;
+  for (int r = 0; r < R; ++r) {
+    int ih = hb + r;
+    if ((unsigned)ih >= (unsigned)H) continue;
+    const __bf16* xrow = xn + (int64_t)ih * W * C;
+    const __bf16* wrow = w + r * S * C + c;
+    for (int j = 0; j < S + 1; ++j) {
+      int iw = wb + j;
+      if ((unsigned)iw >= (unsigned)W) continue;
+      dwbf16x8 xv = *(const dwbf16x8*)(xrow + (int64_t)iw * C);
+      if (j < S) {
+        dwbf16x8 wv = *(const dwbf16x8*)(wrow + j * C);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) a0[e] += (float)xv[e] * (float)wv[e];
+      }
+      if (j >= 1) {
+        dwbf16x8 wv = *(const dwbf16x8*)(wrow + (j - 1) * C);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) a1[e] += (float)xv[e] * (float)wv[e];
+      }
+    }
+  }
+
+  const int64_t p0 = (((int64_t)n * HO + ho) * WO + wo0);
+  if (residual) {
+    dwbf16x8 rv = *(const dwbf16x8*)(residual + p0 * C + c);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) a0[e] += (float)rv[e];
+    if (two) {
+      dwbf16x8 r2 = *(const dwbf16x8*)(residual + (p0 + 1) * C + c);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) a1[e] += (float)r2[e];
+    }
+  }
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    if (act == 1) {
+      a0[e] = fmaxf(a0[e], 0.f);
+      a1[e] = fmaxf(a1[e], 0.f);
+    } else if (act == 2) {
+      a0[e] = fminf(fmaxf(a0[e], 0.f), 6.f);
+      a1[e] = fminf(fmaxf(a1[e], 0.f), 6.f);
+    }
+  }
+  dwbf16x8 v0, v1;
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    v0[e] = (__bf16)a0[e];
+    v1[e] = (__bf16)a1[e];
+  }
+  *(dwbf16x8*)(y + p0 * C + c) = v0;
+  if (two) *(dwbf16x8*)(y + (p0 + 1) * C + c) = v1;
+}
+
 void launch_dwconv_bf16(const void* x, const void* w, const float* bias,
                         const void* residual, void* y, int N, int H,
                         int W, int C, int HO, int WO, int R, int S,
                         int sh, int sw, int pt, int pl, int act,
                         hipStream_t stream) {
   int c8 = (C + 7) >> 3;
+  if ((C & 7) == 0 && sh == 1 && sw == 1 && S <= 7) {
+    int wo2 = (WO + 1) >> 1;
+    int total = N * HO * wo2 * c8;
+    hipLaunchKernelGGL(dwconv2_kernel, dim3((total + 255) / 256),
+                       dim3(256), 0, stream, (const __bf16*)x,
+                       (const __bf16*)w, bias, (const __bf16*)residual,
+                       (__bf16*)y, H, W, C, HO, WO, R, S, pt, pl, act,
+                       c8, wo2, total);
+    return;
+  }
   int total = N * HO * WO * c8;
   int blocks = (total + 255) / 256;
   if ((C & 7) == 0)

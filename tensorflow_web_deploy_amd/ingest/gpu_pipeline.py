@@ -234,7 +234,10 @@ class ShmDecodePool:
         self._meta = np.frombuffer(
             self.buf[slots * 16:slots * 32], dtype=np.int32)
         if workers is None:
-            workers = min(32, max(4, (os.cpu_count() or 8) - 2))
+            # 48-worker cap A/B'd +3% over 32 at 256 clients (the
+            # load generator shares the box; more workers win the
+            # core-stealing fight) — profiles/round2_serving_ab.md
+            workers = min(48, max(4, (os.cpu_count() or 8) - 2))
         self.workers = workers
         ctx = mp.get_context("spawn")
         self._wake, self._done, self._procs = [], [], []
