@@ -12,6 +12,7 @@
 #include <hip/hip_runtime.h>
 
 #include <cstdint>
+#include <cstdlib>
 
 #include "kernels.h"
 
@@ -183,7 +184,11 @@ void launch_dwconv_bf16(const void* x, const void* w, const float* bias,
                         int sh, int sw, int pt, int pl, int act,
                         hipStream_t stream) {
   int c8 = (C + 7) >> 3;
-  if ((C & 7) == 0 && sh == 1 && sw == 1 && S <= 7) {
+  // A/B'd NEGATIVE (MobileNetV2 51.1k -> 48.6k img/s): halving the
+  // thread count costs more latency-hiding than the shared column
+  // loads save (L2 was already absorbing the re-reads). Parked.
+  static const bool dw2 = getenv("TWD_DW2") != nullptr;
+  if (dw2 && (C & 7) == 0 && sh == 1 && sw == 1 && S <= 7) {
     int wo2 = (WO + 1) >> 1;
     int total = N * HO * wo2 * c8;
     hipLaunchKernelGGL(dwconv2_kernel, dim3((total + 255) / 256),
