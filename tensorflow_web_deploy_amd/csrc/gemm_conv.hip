@@ -1185,7 +1185,9 @@ static void dispatch(const void* A, const void* B, void* C, const float* bias,
   // factorized convs: 290 blocks = ~1 block/CU); halve the M-tile to
   // double the grid when it buys >= ~1.5 blocks/CU
   long b128 = ((M + BM - 1) / BM) * (long)((N + BN - 1) / BN);
-  if (splitk == 1 && b128 >= 128 && b128 < 320 && getenv("TWD_NO_6412") == nullptr)
+  static const char* b64min_s = getenv("TWD_6412_MIN");
+  const long b64min = b64min_s ? atol(b64min_s) : 128;
+  if (splitk == 1 && b128 >= b64min && b128 < 320 && getenv("TWD_NO_6412") == nullptr)
     launch_tile<64, 128, MODE, 2>(A, B, C, bias, residual, M, N, K, Kp,
                                   a_bytes, act, cp, stream, ws, splitk, ldc);
   else
