@@ -1185,12 +1185,13 @@ static void dispatch(const void* A, const void* B, void* C, const float* bias,
   // factorized convs: 290 blocks = ~1 block/CU); halve the M-tile to
   // double the grid when it buys >= ~1.5 blocks/CU
   long b128 = ((M + BM - 1) / BM) * (long)((N + BN - 1) / BN);
-  // [48,128)-block shapes: (64,128) pays only at BIG K (Inception
-  // E-block 3x3s, K>=3456: +1.3% e2e) — at ResNet's K=2048 1x1s the
-  // smaller tile's extra staging loses 3% e2e. K>=3072 separates the
-  // two cleanly (measured both ways, profiles/).
+  // [48,128)-block shapes: (64,128) pays on gather convs (Inception
+  // E/D blocks: +1.3% e2e) but loses on plain 1x1 GEMMs (ResNet
+  // K=2048 2048->512: -3% e2e — the linear A re-reads cheaply, so the
+  // bigger tile's fewer B re-reads win). Gate on the gather MODE.
   static const char* b64min_s = getenv("TWD_6412_MIN");
-  const long b64min = b64min_s ? atol(b64min_s) : (K >= 3072 ? 48 : 128);
+  const long b64min =
+      b64min_s ? atol(b64min_s) : ((MODE == 1 || K >= 3072) ? 48 : 128);
   if (splitk == 1 && b128 >= b64min && b128 < 320 && getenv("TWD_NO_6412") == nullptr)
     launch_tile<64, 128, MODE, 2>(A, B, C, bias, residual, M, N, K, Kp,
                                   a_bytes, act, cp, stream, ws, splitk, ldc);
