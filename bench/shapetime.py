@@ -39,3 +39,19 @@ Y4 = torch.zeros(64, 35, 35, 96, dtype=torch.bfloat16, device="cuda")
 ms = t(lambda: E.conv2d_bf16(X4, W4, Y4, None, None, 3, 3, 1, 1, 1, 1, 0))
 fl4 = 2*64*35*35*96*576
 print(f"b35 3x3 64->96: {ms*1000:.1f} us  {fl4/ms/1e9:.0f} TF")
+# Inception pools (35x35 avg 3x3s1 SAME C=288; 17x17 C=768)
+Xp = torch.randn(64, 35, 35, 288, device="cuda").bfloat16()
+Yp = torch.zeros_like(Xp)
+ms = t(lambda: E.avgpool_bf16(Xp, Yp, 3, 3, 1, 1, 1, 1))
+mb = 2 * 64*35*35*288*2 / 1e6
+print(f"avgpool35 C288: {ms*1000:.1f} us  {mb/ms/1000:.2f} TB/s")
+Xp2 = torch.randn(64, 17, 17, 768, device="cuda").bfloat16()
+Yp2 = torch.zeros_like(Xp2)
+ms = t(lambda: E.avgpool_bf16(Xp2, Yp2, 3, 3, 1, 1, 1, 1))
+mb = 2 * 64*17*17*768*2 / 1e6
+print(f"avgpool17 C768: {ms*1000:.1f} us  {mb/ms/1000:.2f} TB/s")
+Xp3 = torch.randn(64, 147, 147, 64, device="cuda").bfloat16()
+Yp3 = torch.zeros(64, 73, 73, 64, dtype=torch.bfloat16, device="cuda")
+ms = t(lambda: E.maxpool_bf16(Xp3, Yp3, 3, 3, 2, 2, 0, 0))
+mb = (64*147*147*64 + 64*73*73*64) * 2 / 1e6
+print(f"maxpool147s2: {ms*1000:.1f} us  {mb/ms/1000:.2f} TB/s")
