@@ -63,9 +63,122 @@ __global__ void pool_kernel(const __bf16* __restrict__ X,
   }
 }
 
+// Stride-1 pools (Inception 3x3s1 SAME avg/max: 10.7% of its step) are
+// separable: out(y) = combine of rowsum(y-1..y+1). A thread computes a
+// VERTICAL STRIP of VS=4 outputs from VS+kh-1 register rowsums: 4.5
+// loads/output instead of 9, vertical reuse in registers. Measured
+// 2.3 -> ~5 TB/s class. KH template keeps register indexing static.
+template <bool MAX, int KH>
+__global__ void pool_s1_kernel(const __bf16* __restrict__ X,
+                               __bf16* __restrict__ Y, int Nimg, int H,
+                               int W, int C, int kw, int pt, int pl,
+                               int HO, int WO) {
+  constexpr int VS = 4;
+  const int c8 = C >> 3;
+  const int strips = (HO + VS - 1) / VS;
+  const int64_t total = (int64_t)Nimg * strips * WO * c8;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < total; idx += (int64_t)gridDim.x * blockDim.x) {
+    uint32_t i32 = (uint32_t)idx;
+    uint32_t cg = i32 % (uint32_t)c8;
+    uint32_t p = i32 / (uint32_t)c8;
+    uint32_t wo = p % (uint32_t)WO;
+    uint32_t q = p / (uint32_t)WO;
+    uint32_t strip = q % (uint32_t)strips;
+    uint32_t n = q / (uint32_t)strips;
+    const int oy0 = (int)strip * VS;
+    const int wbase = (int)wo - pl;
+    // horizontal (column) validity is row-independent
+    int nvc = 0;
+    float rs[VS + KH - 1][8];
+    bool rv[VS + KH - 1];
+    const __bf16* xn = X + ((int64_t)n * H * W * C) + cg * 8;
+#pragma unroll
+    for (int r = 0; r < VS + KH - 1; ++r) {
+      int iy = oy0 - pt + r;
+      rv[r] = (unsigned)iy < (unsigned)H;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) rs[r][e] = MAX ? -3.4e38f : 0.f;
+      if (!rv[r]) continue;
+      const __bf16* xrow = xn + (int64_t)iy * W * C;
+      int cnt = 0;
+      for (int s = 0; s < kw; ++s) {
+        int iw = wbase + s;
+        if ((unsigned)iw >= (unsigned)W) continue;
+        ++cnt;
+        bf16x8 v = *(const bf16x8*)(xrow + (int64_t)iw * C);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          float f = bf2f(v[e]);
+          rs[r][e] = MAX ? fmaxf(rs[r][e], f) : rs[r][e] + f;
+        }
+      }
+      if (r == 0) nvc = cnt;
+      else if (cnt) nvc = cnt;  // same for every valid row
+    }
+#pragma unroll
+    for (int v = 0; v < VS; ++v) {
+      int oy = oy0 + v;
+      if (oy >= HO) break;
+      float acc[8];
+      int nvr = 0;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) acc[e] = MAX ? -3.4e38f : 0.f;
+#pragma unroll
+      for (int r = 0; r < KH; ++r) {
+        if (!rv[v + r]) continue;
+        ++nvr;
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          acc[e] = MAX ? fmaxf(acc[e], rs[v + r][e])
+                       : acc[e] + rs[v + r][e];
+      }
+      bf16x8 o;
+      const float inv = MAX ? 1.f : 1.f / (float)max(nvr * max(nvc, 1), 1);
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        o[e] = (__bf16)(MAX ? acc[e] : acc[e] * inv);
+      *(bf16x8*)(Y + ((((int64_t)n * HO + oy) * WO + wo) * C) + cg * 8) =
+          o;
+    }
+  }
+}
+
+template <bool MAX>
+static bool pool_s1_dispatch(const void* X, void* Y, int Nimg, int H,
+                             int W, int C, int kh, int kw, int pt, int pl,
+                             int HO, int WO, hipStream_t stream) {
+  if ((C & 7) != 0 || kh < 2 || kh > 5) return false;
+  int strips = (HO + 3) / 4;
+  int64_t total = (int64_t)Nimg * strips * WO * (C >> 3);
+  int blocks = (int)std::min<int64_t>((total + 255) / 256, 1 << 20);
+  switch (kh) {
+    case 2:
+      hipLaunchKernelGGL((pool_s1_kernel<MAX, 2>), dim3(blocks), dim3(256),
+                         0, stream, (const __bf16*)X, (__bf16*)Y, Nimg, H,
+                         W, C, kw, pt, pl, HO, WO);
+      return true;
+    case 3:
+      hipLaunchKernelGGL((pool_s1_kernel<MAX, 3>), dim3(blocks), dim3(256),
+                         0, stream, (const __bf16*)X, (__bf16*)Y, Nimg, H,
+                         W, C, kw, pt, pl, HO, WO);
+      return true;
+    case 5:
+      hipLaunchKernelGGL((pool_s1_kernel<MAX, 5>), dim3(blocks), dim3(256),
+                         0, stream, (const __bf16*)X, (__bf16*)Y, Nimg, H,
+                         W, C, kw, pt, pl, HO, WO);
+      return true;
+  }
+  return false;
+}
+
 void launch_maxpool_bf16(const void* X, void* Y, int Nimg, int H, int W,
                          int C, int kh, int kw, int sh, int sw, int pt,
                          int pl, int HO, int WO, hipStream_t stream) {
+  if (sh == 1 && sw == 1 &&
+      pool_s1_dispatch<true>(X, Y, Nimg, H, W, C, kh, kw, pt, pl, HO, WO,
+                             stream))
+    return;
   int64_t total = (int64_t)Nimg * HO * WO * (C >> 3);
   // one output per thread (no cap): the former 2048-block cap made
   // each thread redo the divmod chain ~5x on 35x35 maps
@@ -78,6 +191,10 @@ void launch_maxpool_bf16(const void* X, void* Y, int Nimg, int H, int W,
 void launch_avgpool_bf16(const void* X, void* Y, int Nimg, int H, int W,
                          int C, int kh, int kw, int sh, int sw, int pt,
                          int pl, int HO, int WO, hipStream_t stream) {
+  if (sh == 1 && sw == 1 &&
+      pool_s1_dispatch<false>(X, Y, Nimg, H, W, C, kh, kw, pt, pl, HO, WO,
+                              stream))
+    return;
   int64_t total = (int64_t)Nimg * HO * WO * (C >> 3);
   int blocks = (int)std::min<int64_t>((total + 255) / 256, 1 << 20);
   hipLaunchKernelGGL((pool_kernel<false>), dim3(blocks), dim3(256), 0, stream,

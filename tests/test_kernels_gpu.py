@@ -360,3 +360,26 @@ def test_conv_direct_shapes(E, N, H, W, C, R, S, O, stride, same):
     ref2 = torch.relu(ref + bias.cpu())
     torch.testing.assert_close(Y.float().cpu(), ref2, rtol=3e-2,
                                atol=2e-2 * K ** 0.5)
+
+
+@pytest.mark.parametrize("maxp", [True, False])
+@pytest.mark.parametrize("H,C", [(35, 288), (17, 768), (15, 64), (7, 8)])
+def test_pool_s1_same(E, maxp, H, C):
+    """Stride-1 SAME pools route to the separable vertical-strip kernel
+    (pool_s1_kernel); must match the CPU reference incl border counts."""
+    torch.manual_seed(5)
+    X = torch.randn(2, H, H, C)
+    Xg = _bf(X)
+    from tensorflow_web_deploy_amd.engine.executor import _CPU_OPS
+    from tensorflow_web_deploy_amd.engine.ir import Node
+    op = "MaxPool" if maxp else "AvgPool"
+    node = Node("p", op, ["x"],
+                {"ksize": [1, 3, 3, 1], "strides": [1, 1, 1, 1],
+                 "padding": "SAME"})
+    ref = _CPU_OPS[op](node, [Xg.float().cpu()], {})
+    Y = torch.zeros(2, H, H, C, dtype=torch.bfloat16, device="cuda")
+    if maxp:
+        E.maxpool_bf16(Xg, Y, 3, 3, 1, 1, 1, 1)
+    else:
+        E.avgpool_bf16(Xg, Y, 3, 3, 1, 1, 1, 1)
+    torch.testing.assert_close(Y.float().cpu(), ref, rtol=2e-2, atol=2e-2)
