@@ -291,7 +291,10 @@ bool launch_conv_direct(const void* X, const void* W2, void* Y,
   if (C % 32 != 0 || Cout % 16 != 0 || Cout < 16 || Cout > 192 ||
       R * S < 3 || act > 2)
     return false;
-  if (!open_all && !(C <= 64 && Cout <= 64 && WO >= 96))
+  // Cout<=48: the Cout=64 stem (conv2) measured 153 us via direct vs
+  // 89 us on the GEMM path in-model (NF=4 halves MFW, and the wider
+  // epilogue tile halves the LDS bands) — only the narrow-N stems win
+  if (!open_all && !(C <= 64 && Cout <= 48 && WO >= 96))
     return false;
   const int NF = Cout / 16;
   // MFW from the VGPR budget: MFW * NF <= 24 and MFW <= 10
