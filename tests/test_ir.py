@@ -200,3 +200,34 @@ def test_optimize_preserves_semantics(family):
     p_raw = Executor(raw).run({"input": x})[0]
     p_opt = Executor(opt).run({"input": x})[0]
     torch.testing.assert_close(p_raw, p_opt, atol=2e-4, rtol=2e-4)
+
+
+def test_assign_streams_inception_branches():
+    """Inter-branch stream assignment (gpu_executor.assign_streams is a
+    pure function): Inception blocks fork 2-4 independent conv chains —
+    sibling chains must land on different streams, joins and trunk stay
+    on stream 0, and chain membership never crosses a multi-input op."""
+    from tensorflow_web_deploy_amd.conf.config import Config
+    from tensorflow_web_deploy_amd.engine.gpu_executor import assign_streams
+    from tensorflow_web_deploy_amd.engine.service import build_graph
+    cfg = Config(model_name="inception_v3", device="cpu",
+                 input_height=299, input_width=299, input_channels=3,
+                 num_classes=1000)
+    g = build_graph(cfg)
+    sid = assign_streams(g.nodes, g.consts, 2)
+    n_side = sum(1 for v in sid.values() if v)
+    assert n_side >= 20, f"expected many side-chain nodes, got {n_side}"
+    prods = {n.name: n for n in g.nodes}
+    # no multi-input (join) node off stream 0
+    for n in g.nodes:
+        ins = {i.split(":")[0] for i in n.inputs if i.split(":")[0] in prods}
+        if len(ins) > 1:
+            assert sid[n.name] == 0, n.name
+    # ViT / ResNet-style linear trunks: residual Adds are joins -> 0;
+    # spot-check ResNet stays effectively single-stream
+    cfg2 = Config(model_name="resnet50", device="cpu", input_height=224,
+                  input_width=224, input_channels=3, num_classes=1000)
+    g2 = build_graph(cfg2)
+    sid2 = assign_streams(g2.nodes, g2.consts, 2)
+    frac = sum(1 for v in sid2.values() if v) / max(1, len(sid2))
+    assert frac < 0.5, frac
