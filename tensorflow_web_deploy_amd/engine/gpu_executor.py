@@ -819,6 +819,28 @@ class GpuExecutor:
         # one event after its last step; consumers wait before their
         # first step. The input-cast steps (before any node span) stay
         # on stream 0, so side chains fork via these edges.
+        # a node may emit NO steps (eliminated ConcatV2, view-only
+        # Reshape): dependencies must pass THROUGH it to the nodes that
+        # actually wrote the underlying buffer
+        eff_cache: Dict[str, tuple] = {}
+
+        def eff_producers(name):
+            if name in eff_cache:
+                return eff_cache[name]
+            if name in span_of:
+                out = (name,)
+            elif name in prods:
+                acc = []
+                for i in prods[name].inputs:
+                    b = i.split(":")[0]
+                    if b in prods:
+                        acc.extend(eff_producers(b))
+                out = tuple(dict.fromkeys(acc))
+            else:
+                out = ()
+            eff_cache[name] = out
+            return out
+
         rec_of: Dict[str, int] = {}   # node -> event id
         waits = [[] for _ in range(n_steps)]
         recs = [None] * n_steps
@@ -829,16 +851,20 @@ class GpuExecutor:
             my = sid_of.get(n.name, 0)
             for i in n.inputs:
                 base = i.split(":")[0]
-                if base not in span_of or base not in prods:
+                if base not in prods:
                     continue
-                pside = sid_of.get(base, 0)
-                if pside == my:
-                    continue
-                if base not in rec_of:
-                    rec_of[base] = nev
-                    recs[span_of[base][1] - 1] = nev
-                    nev += 1
-                waits[span_of[n.name][0]].append(rec_of[base])
+                for pr in eff_producers(base):
+                    pside = sid_of.get(pr, 0)
+                    if pside == my:
+                        continue
+                    if pr not in rec_of:
+                        rec_of[pr] = nev
+                        recs[span_of[pr][1] - 1] = nev
+                        nev += 1
+                    ev = rec_of[pr]
+                    w = waits[span_of[n.name][0]]
+                    if ev not in w:
+                        w.append(ev)
         # join: record an event at the end of every side-stream node
         # that produces a graph OUTPUT (none in practice) — outputs are
         # on stream 0 by construction; also make stream 0's final step
