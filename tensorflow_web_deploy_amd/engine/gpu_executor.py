@@ -51,8 +51,10 @@ def _pad64(n):
 
 # ops a side-stream chain may contain (pure native-kernel steps with no
 # torch temporaries — keeps graph capture and the caching allocator
-# trivially safe on side streams)
-_CHAIN_OPS = ("ConvBlock", "DepthwiseConvBlock", "MaxPool", "AvgPool")
+# trivially safe on side streams). MatMul-family included for ViT's
+# 3-way Q/K/V fork (each projection fills only ~0.6 blocks/CU alone).
+_CHAIN_OPS = ("ConvBlock", "DepthwiseConvBlock", "MaxPool", "AvgPool",
+              "MatMul", "Dense", "DenseND")
 
 
 def assign_streams(nodes, consts, n_streams: int = 2):
