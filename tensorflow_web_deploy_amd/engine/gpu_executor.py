@@ -69,9 +69,20 @@ def assign_streams(nodes, consts, n_streams: int = 2):
     everything else stays on stream 0. Pure function (CPU-tested).
     """
     prods = {n.name: n for n in nodes}
+
+    def node_inputs(n):
+        """tensor inputs INCLUDING fused-residual references (carried
+        in attrs, not inputs — missing them would drop a cross-stream
+        dependency edge)"""
+        ins = list(n.inputs)
+        r = n.attrs.get("residual") if n.attrs else None
+        if r:
+            ins.append(r)
+        return ins
+
     consumers: Dict[str, list] = {}
     for n in nodes:
-        for i in n.inputs:
+        for i in node_inputs(n):
             base = i.split(":")[0]
             if base in prods:
                 consumers.setdefault(base, []).append(n.name)
@@ -86,7 +97,7 @@ def assign_streams(nodes, consts, n_streams: int = 2):
             cur = c
             while True:
                 nd = prods[cur]
-                ins = [i.split(":")[0] for i in nd.inputs
+                ins = [i.split(":")[0] for i in node_inputs(nd)
                        if i.split(":")[0] in prods]
                 if nd.op not in _CHAIN_OPS or len(set(ins)) != 1:
                     break  # join / non-chain op: stays on stream 0
@@ -837,7 +848,12 @@ class GpuExecutor:
                 out = (name,)
             elif name in prods:
                 acc = []
-                for i in prods[name].inputs:
+                nd = prods[name]
+                srcs = list(nd.inputs)
+                r = nd.attrs.get("residual") if nd.attrs else None
+                if r:
+                    srcs.append(r)
+                for i in srcs:
                     b = i.split(":")[0]
                     if b in prods:
                         acc.extend(eff_producers(b))
@@ -855,7 +871,11 @@ class GpuExecutor:
             if n.name not in span_of:
                 continue
             my = sid_of.get(n.name, 0)
-            for i in n.inputs:
+            srcs = list(n.inputs)
+            r = n.attrs.get("residual") if n.attrs else None
+            if r:
+                srcs.append(r)
+            for i in srcs:
                 base = i.split(":")[0]
                 if base not in prods:
                     continue
