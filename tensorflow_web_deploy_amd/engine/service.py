@@ -44,6 +44,11 @@ def build_graph(cfg: Config) -> Graph:
             lambda: mobilenet_v2.build_mobilenet_v2()
     except ImportError:
         pass
+    try:
+        from ..models import vgg16
+        builders["vgg16"] = lambda: vgg16.build_vgg16()
+    except ImportError:
+        pass
     if cfg.model_name not in builders:
         raise ValueError(f"unknown built-in model {cfg.model_name}")
     g = builders[cfg.model_name]()

@@ -171,7 +171,7 @@ def test_multi_output_split_through_executor():
 
 
 @pytest.mark.parametrize("family", ["lenet", "resnet50", "mobilenet_v2",
-                                    "inception_v3", "vit_b16"])
+                                    "inception_v3", "vit_b16", "vgg16"])
 def test_optimize_preserves_semantics(family):
     """optimize() (BN fold, block fusion, concat elimination, ...) must
     not change the function: raw-graph execution == optimized-graph
@@ -183,12 +183,13 @@ def test_optimize_preserves_semantics(family):
     from tensorflow_web_deploy_amd.engine.executor import Executor
     from tensorflow_web_deploy_amd.models import (inception_v3, lenet,
                                                   mobilenet_v2, resnet50,
-                                                  vit)
+                                                  vgg16, vit)
     build = {"lenet": lenet.build_lenet,
              "resnet50": resnet50.build_resnet50,
              "mobilenet_v2": mobilenet_v2.build_mobilenet_v2,
              "inception_v3": inception_v3.build_inception_v3,
-             "vit_b16": vit.build_vit_b16}[family]
+             "vit_b16": vit.build_vit_b16,
+             "vgg16": vgg16.build_vgg16}[family]
     g = build(seed=3)
     raw = ir.from_graphdef(g, outputs=["softmax"])
     opt = ir.optimize(ir.from_graphdef(g, outputs=["softmax"]))

@@ -51,3 +51,8 @@ def test_layers_mobilenet_v2():
 @pytest.mark.timeout(300)
 def test_layers_lenet():
     _check("lenet", 28, 1, 10, batch=8)
+
+
+@pytest.mark.timeout(900)
+def test_layers_vgg16():
+    _check("vgg16", 224, 3, 1000, batch=2)

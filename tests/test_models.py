@@ -70,3 +70,11 @@ def test_vit_layernorm_fusion_equivalence():
     y_raw = Executor(raw).run({"input": xv})[0]
     y_opt = Executor(opt).run({"input": xv})[0]
     torch.testing.assert_close(y_raw, y_opt, rtol=1e-5, atol=1e-6)
+
+
+def test_vgg16():
+    eng, ops = _run("vgg16", 224, 224, 3, 1000)
+    assert ops["ConvBlock"] == 13
+    assert ops["MaxPool"] == 5
+    # fc6/fc7/fc8 fuse to Dense blocks
+    assert ops.get("Dense", 0) + ops.get("DenseBlock", 0) == 3
