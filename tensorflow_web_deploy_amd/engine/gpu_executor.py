@@ -810,10 +810,7 @@ class GpuExecutor:
         plan.sched is None when everything is on stream 0."""
         import os
         ns = int(os.environ.get("TWD_EXEC_STREAMS", "2"))
-        # >2 streams segfaulted in hipGraph capture on ROCm 7.2 (4-way
-        # fork, round-2 A/B); 2 streams is validated (+15% Inception)
-        # and covers the pairwise-branch overlap. Clamp.
-        ns = min(ns, 2)
+        ns = min(ns, 4)
         if ns < 2:
             plan.sched = None
             return
