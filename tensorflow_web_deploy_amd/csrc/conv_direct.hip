@@ -34,7 +34,6 @@ typedef float f32x4_cd __attribute__((ext_vector_type(4)));
 
 extern __shared__ __attribute__((aligned(16))) char smem_cd[];
 
-__device__ __forceinline__ float cd_bf2f(__bf16 v) { return (float)v; }
 
 // 40-bit magic division helpers (same scheme as gemm_conv.hip)
 __device__ __forceinline__ uint32_t cd_mdiv(uint32_t x, uint64_t magic) {
