@@ -49,6 +49,12 @@ def build_graph(cfg: Config) -> Graph:
         builders["vgg16"] = lambda: vgg16.build_vgg16()
     except ImportError:
         pass
+    try:
+        from ..models import densenet121
+        builders["densenet121"] = \
+            lambda: densenet121.build_densenet121()
+    except ImportError:
+        pass
     if cfg.model_name not in builders:
         raise ValueError(f"unknown built-in model {cfg.model_name}")
     g = builders[cfg.model_name]()

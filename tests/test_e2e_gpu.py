@@ -110,3 +110,17 @@ def test_vgg16_gpu_vs_cpu():
         assert pg[i].argmax() in top5[i]
     corr = np.corrcoef(pg.ravel(), pc.ravel())[0, 1]
     assert corr > 0.95, corr
+
+
+def test_densenet121_gpu_vs_cpu():
+    gpu, cpu = _engines("densenet121", 224, 224, 3, 1000)
+    x = (np.random.default_rng(9).random((2, 224, 224, 3),
+                                          dtype=np.float32) - 0.5)
+    pg = gpu.infer(x)
+    pc = cpu.infer(x)
+    np.testing.assert_allclose(pg.sum(1), 1.0, atol=1e-3)
+    top5 = np.argsort(pc, 1)[:, -5:]
+    for i in range(pg.shape[0]):
+        assert pg[i].argmax() in top5[i]
+    corr = np.corrcoef(pg.ravel(), pc.ravel())[0, 1]
+    assert corr > 0.95, corr

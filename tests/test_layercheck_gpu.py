@@ -56,3 +56,8 @@ def test_layers_lenet():
 @pytest.mark.timeout(900)
 def test_layers_vgg16():
     _check("vgg16", 224, 3, 1000, batch=2)
+
+
+@pytest.mark.timeout(900)
+def test_layers_densenet121():
+    _check("densenet121", 224, 3, 1000, batch=2)

@@ -78,3 +78,13 @@ def test_vgg16():
     assert ops["MaxPool"] == 5
     # fc6/fc7/fc8 fuse to Dense blocks
     assert ops.get("Dense", 0) + ops.get("DenseBlock", 0) == 3
+
+
+def test_densenet121():
+    eng, ops = _run("densenet121", 224, 224, 3, 1000)
+    # 120 convs (stem + 58x2 dense-layer convs + 3 transitions), all
+    # pre-activation: BN runs standalone as BnScaleBias
+    assert ops["ConvBlock"] == 120
+    assert ops["BnScaleBias"] == 62
+    assert ops["ConcatV2"] == 58
+    assert ops["AvgPool"] == 3
