@@ -171,7 +171,8 @@ def test_multi_output_split_through_executor():
 
 
 @pytest.mark.parametrize("family", ["lenet", "resnet50", "mobilenet_v2",
-                                    "inception_v3", "vit_b16", "vgg16"])
+                                    "inception_v3", "vit_b16", "vgg16",
+                                    "densenet121"])
 def test_optimize_preserves_semantics(family):
     """optimize() (BN fold, block fusion, concat elimination, ...) must
     not change the function: raw-graph execution == optimized-graph
@@ -181,7 +182,8 @@ def test_optimize_preserves_semantics(family):
 
     from tensorflow_web_deploy_amd.engine import ir
     from tensorflow_web_deploy_amd.engine.executor import Executor
-    from tensorflow_web_deploy_amd.models import (inception_v3, lenet,
+    from tensorflow_web_deploy_amd.models import (densenet121,
+                                                  inception_v3, lenet,
                                                   mobilenet_v2, resnet50,
                                                   vgg16, vit)
     build = {"lenet": lenet.build_lenet,
@@ -189,7 +191,8 @@ def test_optimize_preserves_semantics(family):
              "mobilenet_v2": mobilenet_v2.build_mobilenet_v2,
              "inception_v3": inception_v3.build_inception_v3,
              "vit_b16": vit.build_vit_b16,
-             "vgg16": vgg16.build_vgg16}[family]
+             "vgg16": vgg16.build_vgg16,
+             "densenet121": densenet121.build_densenet121}[family]
     g = build(seed=3)
     raw = ir.from_graphdef(g, outputs=["softmax"])
     opt = ir.optimize(ir.from_graphdef(g, outputs=["softmax"]))
