@@ -375,3 +375,52 @@ def test_metrics_prometheus(server):
         assert r.status == 200
         assert "twd_requests_total" in text
         assert "twd_responses_total" in text
+
+
+def test_frontend_limit_edges(server):
+    """C++ frontend hardening: runaway headers -> 400, Content-Length
+    beyond 32 MiB (incl. negative wrap) -> 413, garbage request line ->
+    connection closed, pipelined requests on one socket all answered."""
+    import socket
+
+    def raw(data, expect_close=False, nresp=1):
+        s = socket.create_connection(("127.0.0.1", server.port), timeout=10)
+        s.sendall(data)
+        buf = b""
+        try:
+            while buf.count(b"HTTP/1.1") < nresp:
+                d = s.recv(65536)
+                if not d:
+                    break
+                buf += d
+        except socket.timeout:
+            pass
+        s.close()
+        return buf
+
+    # runaway header block (>64 KiB without CRLFCRLF) -> 400
+    r = raw(b"POST /predict HTTP/1.1\r\nX-Junk: " + b"a" * 70000)
+    assert b"400" in r.split(b"\r\n")[0]
+
+    # oversized Content-Length -> 413 before any body is read
+    r = raw(b"POST /predict HTTP/1.1\r\nHost: x\r\n"
+            b"Content-Length: 999999999999\r\n\r\n")
+    assert b"413" in r.split(b"\r\n")[0]
+
+    # negative Content-Length wraps to huge -> 413 (never a hang)
+    r = raw(b"POST /predict HTTP/1.1\r\nHost: x\r\n"
+            b"Content-Length: -1\r\n\r\n")
+    assert b"413" in r.split(b"\r\n")[0]
+
+    # garbage request line -> closed without a crash
+    r = raw(b"NONSENSE\r\n\r\n")
+    assert b"predictions" not in r
+
+    # two pipelined healthz on one socket -> two responses
+    r = raw(b"GET /healthz HTTP/1.1\r\nHost: x\r\n\r\n"
+            b"GET /healthz HTTP/1.1\r\nHost: x\r\n\r\n", nresp=2)
+    assert r.count(b"200 OK") == 2
+
+    # server still healthy afterwards
+    status, j = _post(server.port, _png_bytes())
+    assert status == 200
