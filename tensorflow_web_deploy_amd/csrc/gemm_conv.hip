@@ -436,6 +436,13 @@ void gemm_conv_kernel_256(const __bf16* __restrict__ A,
   const int lane = tid & 63;
   const int wave = __builtin_amdgcn_readfirstlane(tid >> 6);
 
+  // tiles_n < 0 flags 4x4 tile clustering (host A/B, TWD_TILE2D): an
+  // XCD's contiguous logical range then walks 4x4 tile blocks, so its
+  // co-resident workgroups share 4 A-panels AND 4 B-panels in L2
+  // instead of 2 B + 16 A (column-major) — ~2.25x less HBM per K-tile
+  // when the kernel is L2-miss-bound.
+  const bool t2d = tiles_n < 0;
+  if (t2d) tiles_n = -tiles_n;
   int nwg = tiles_m * tiles_n;
   int bid = blockIdx.x;
   {
@@ -443,8 +450,17 @@ void gemm_conv_kernel_256(const __bf16* __restrict__ A,
     int xcd = bid & 7, off = bid >> 3;
     bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
   }
-  const int tn = bid / tiles_m;
-  const int tm = bid - tn * tiles_m;
+  int tn, tm;
+  if (t2d) {
+    const int bpr = tiles_m >> 2;  // 4x4 blocks per block-row
+    const int blk = bid >> 4, wi = bid & 15;
+    const int bm = blk % bpr, bn = blk / bpr;
+    tm = bm * 4 + (wi & 3);
+    tn = bn * 4 + (wi >> 2);
+  } else {
+    tn = bid / tiles_m;
+    tm = bid - tn * tiles_m;
+  }
   const int row0 = tm * BM2;
   const int col0 = tn * BN2;
 
@@ -1117,6 +1133,12 @@ static void dispatch(const void* A, const void* B, void* C, const float* bias,
     int tiles_m = (M + BM2 - 1) / BM2;
     int tiles_n = (N + BN2 - 1) / BN2;
     dim3 grid(tiles_m * tiles_n);
+    // 4x4 tile clustering (see kernel comment): needs both tile dims
+    // divisible by 4; negated tiles_n is the wire flag
+    static const char* t2d = getenv("TWD_TILE2D");
+    int tiles_n_arg = tiles_n;
+    if (t2d && t2d[0] == '1' && (tiles_m & 3) == 0 && (tiles_n & 3) == 0)
+      tiles_n_arg = -tiles_n;
     if (use_256p()) {
       if (residual)
         hipLaunchKernelGGL((gemm_conv_kernel_256p<MODE, true>), grid,
@@ -1137,13 +1159,13 @@ static void dispatch(const void* A, const void* B, void* C, const float* bias,
                          dim3(NTHREADS2), LDS2_BYTES, stream,
                          (const __bf16*)A, (const __bf16*)B, (__bf16*)C, bias,
                          (const __bf16*)residual, M, N, K, Kp, (int)a_bytes,
-                         tiles_m, tiles_n, act, cp, ldc);
+                         tiles_m, tiles_n_arg, act, cp, ldc);
     else
       hipLaunchKernelGGL((gemm_conv_kernel_256<MODE, false>), grid,
                          dim3(NTHREADS2), LDS2_BYTES, stream,
                          (const __bf16*)A, (const __bf16*)B, (__bf16*)C, bias,
                          (const __bf16*)nullptr, M, N, K, Kp, (int)a_bytes,
-                         tiles_m, tiles_n, act, cp, ldc);
+                         tiles_m, tiles_n_arg, act, cp, ldc);
     return;
   }
   // DEPTH=3 measured WORSE here (120 KiB LDS -> 1 block/CU, losing the
