@@ -1134,11 +1134,16 @@ static void dispatch(const void* A, const void* B, void* C, const float* bias,
     int tiles_n = (N + BN2 - 1) / BN2;
     dim3 grid(tiles_m * tiles_n);
     // 4x4 tile clustering (see kernel comment): needs both tile dims
-    // divisible by 4; negated tiles_n is the wire flag
+    // divisible by 4; negated tiles_n is the wire flag. Default ON in
+    // the L2-overflow regime (nwg >= 512: 8192^3 measured 1030 -> 1293
+    // TF, +26%); smaller grids already L2-fit and measured flat-to-
+    // slightly-negative at 4096^3. TWD_TILE2D=0/1 forces.
     static const char* t2d = getenv("TWD_TILE2D");
     int tiles_n_arg = tiles_n;
-    if (t2d && t2d[0] == '1' && (tiles_m & 3) == 0 && (tiles_n & 3) == 0)
-      tiles_n_arg = -tiles_n;
+    const bool div4 = (tiles_m & 3) == 0 && (tiles_n & 3) == 0;
+    const bool want = t2d ? (t2d[0] == '1')
+                          : ((long)tiles_m * tiles_n >= 512);
+    if (div4 && want) tiles_n_arg = -tiles_n;
     if (use_256p()) {
       if (residual)
         hipLaunchKernelGGL((gemm_conv_kernel_256p<MODE, true>), grid,
