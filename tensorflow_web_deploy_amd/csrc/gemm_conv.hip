@@ -112,6 +112,10 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
   const int wave = __builtin_amdgcn_readfirstlane(tid >> 6);
 
   // ---- XCD-aware bijective remap (guide §5: T1, bijective variant) ----
+  // tiles_n < 0 flags 4x4 tile clustering (L2 panel sharing in the
+  // overflow regime — see gemm_conv_kernel_256's comment)
+  const bool t2d = tiles_n < 0;
+  if (t2d) tiles_n = -tiles_n;
   int nwg = tiles_m * tiles_n;
   int bid = blockIdx.x;
   {
@@ -120,8 +124,16 @@ void gemm_conv_kernel(const __bf16* __restrict__ A,
     bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
   }
   // tm fastest => consecutive blocks share the B panel (L2 reuse)
-  const int tn = bid / tiles_m;
-  const int tm = bid - tn * tiles_m;
+  int tn, tm;
+  if (t2d) {
+    const int bpr = tiles_m >> 2;
+    const int blk = bid >> 4, wi = bid & 15;
+    tm = (blk % bpr) * 4 + (wi & 3);
+    tn = (blk / bpr) * 4 + (wi >> 2);
+  } else {
+    tn = bid / tiles_m;
+    tm = bid - tn * tiles_m;
+  }
   const int row0 = tm * TBM;
   const int col0 = tn * TBN;
 
@@ -1069,19 +1081,27 @@ static void launch_tile(const void* A, const void* B, void* C,
     return;
   }
   dim3 grid(tiles_m * tiles_n);
+  static const char* t2dv = getenv("TWD_TILE2D");
+  int tiles_n_arg = tiles_n;
+  {
+    const bool div4 = (tiles_m & 3) == 0 && (tiles_n & 3) == 0;
+    const bool want = t2dv ? (t2dv[0] == '1')
+                           : ((long)tiles_m * tiles_n >= 512);
+    if (div4 && want) tiles_n_arg = -tiles_n;
+  }
   if (residual)
     hipLaunchKernelGGL((gemm_conv_kernel<TBM, TBN, MODE, true, DEPTH>), grid,
                        dim3(THREADS), LDS_TOTAL, stream, (const __bf16*)A,
                        (const __bf16*)B, (__bf16*)C, bias,
                        (const __bf16*)residual, M, N, K, Kp, (int)a_bytes,
-                       tiles_m, tiles_n, act, cp, (int64_t)0,
+                       tiles_m, tiles_n_arg, act, cp, (int64_t)0,
                        (int64_t)0, (int64_t)0, nullptr, 1 << 28, ldc);
   else
     hipLaunchKernelGGL((gemm_conv_kernel<TBM, TBN, MODE, false, DEPTH>), grid,
                        dim3(THREADS), LDS_TOTAL, stream, (const __bf16*)A,
                        (const __bf16*)B, (__bf16*)C, bias,
                        (const __bf16*)nullptr, M, N, K, Kp, (int)a_bytes,
-                       tiles_m, tiles_n, act, cp, (int64_t)0,
+                       tiles_m, tiles_n_arg, act, cp, (int64_t)0,
                        (int64_t)0, (int64_t)0, nullptr, 1 << 28, ldc);
 }
 
