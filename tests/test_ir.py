@@ -235,3 +235,30 @@ def test_assign_streams_inception_branches():
     sid2 = assign_streams(g2.nodes, g2.consts, 2)
     frac = sum(1 for v in sid2.values() if v) / max(1, len(sid2))
     assert frac < 0.5, frac
+
+
+def test_tile_cluster_remap_bijective():
+    """Mirror of the 4x4 tile-clustering + XCD remap in gemm_conv.hip:
+    every (tm, tn) tile must be produced exactly once for any grid with
+    both dims divisible by 4 (a hole would silently skip output tiles)."""
+    def remap(bid, nwg):
+        q, r = nwg >> 3, nwg & 7
+        xcd, off = bid & 7, bid >> 3
+        return (xcd * (q + 1) if xcd < r
+                else r * (q + 1) + (xcd - r) * q) + off
+
+    for tiles_m, tiles_n in [(16, 16), (32, 32), (4, 128), (128, 4),
+                             (20, 8), (8, 20), (196, 4)]:
+        nwg = tiles_m * tiles_n
+        seen = set()
+        bpr = tiles_m >> 2
+        for hw_bid in range(nwg):
+            bid = remap(hw_bid, nwg)
+            blk, wi = bid >> 4, bid & 15
+            bm, bn = blk % bpr, blk // bpr
+            tm = bm * 4 + (wi & 3)
+            tn = bn * 4 + (wi >> 2)
+            assert 0 <= tm < tiles_m and 0 <= tn < tiles_n, \
+                (tiles_m, tiles_n, hw_bid, tm, tn)
+            seen.add((tm, tn))
+        assert len(seen) == nwg, (tiles_m, tiles_n, len(seen))
