@@ -240,6 +240,12 @@ class BlobPre:
     def warm(self):
         self.pool.warm()
 
+    # GpuPreprocessor-interface alias (serve_bench warms either engine
+    # through `backend.pre.warm_pool()`; round-2 rehearsal caught the
+    # AttributeError on the DP path)
+    def warm_pool(self):
+        self.warm()
+
     def desc(self):
         p = self.pool
         return {"path": p.path, "slot_bytes": p.slot_bytes,

@@ -283,3 +283,11 @@ def test_pipelined_fast_server_world8_gloo():
     finally:
         srv.stop()
         eng.shutdown()
+
+
+def test_blobpre_has_serving_warm_interface():
+    """serve_bench warms any engine through backend.pre.warm_pool();
+    the DP path's BlobPre must expose it (round-2 rehearsal caught the
+    AttributeError that broke `serve_bench --gpus N`)."""
+    from tensorflow_web_deploy_amd.parallel.replicas import BlobPre
+    assert hasattr(BlobPre, "warm_pool") and hasattr(BlobPre, "warm")
