@@ -18,6 +18,7 @@
 #include <hip/hip_bf16.h>
 #include <algorithm>
 #include <cstdint>
+#include <cstdlib>
 
 #include "kernels.h"
 
@@ -245,6 +246,237 @@ void attn_fused_kernel(const __bf16* __restrict__ Q,
   }
 }
 
+// ---------------------------------------------------------------------
+// v2 (TWD_ATTN2=1): swapped-QK^T in-register softmax (guide T12 class).
+// mfma(K, Q) puts the scores KEY-major / QUERY-per-lane, so the online
+// softmax reduces 8 in-lane values + two stride-16/32 shuffles per
+// query (vs 4 shuffle steps x 8 rows), and P reaches the PV A-fragment
+// through cvt_pk + 4 ds_bpermute lane exchanges — the wave-private
+// P-LDS bounce (32 addressed ds_write_b16 + 2 lgkm drains per tile)
+// disappears. Fragment algebra in NOTES_ROUND3.md "T12 worked notes".
+__global__ __launch_bounds__(512)
+void attn_fused2_kernel(const __bf16* __restrict__ Q,
+                        const __bf16* __restrict__ K,
+                        const __bf16* __restrict__ V,
+                        __bf16* __restrict__ O,
+                        int batch2, int S, int Sp,
+                        int64_t sQ1, int64_t sQ2, int lqd,
+                        int64_t sK1, int64_t sK2, int lkd,
+                        int64_t sV1, int64_t sV2, int lvd,
+                        int64_t sO1, int64_t sO2, int lod,
+                        float scale) {
+  const int b1 = blockIdx.x / batch2;
+  const int b2 = blockIdx.x - b1 * batch2;
+  Q += b1 * sQ1 + b2 * sQ2;
+  K += b1 * sK1 + b2 * sK2;
+  V += b1 * sV1 + b2 * sV2;
+  O += b1 * sO1 + b2 * sO2;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = __builtin_amdgcn_readfirstlane(tid >> 6);
+  const int jq = lane >> 4;  // lane quarter
+
+  const int vt_ld = Sp + 8;
+  __bf16* k_lds = (__bf16*)smem;
+  __bf16* vt_lds = k_lds + Sp * AT_D;
+
+  // ---- stage K (swizzled) + Vt — identical to v1 ----
+  for (int idx = tid; idx < Sp * 8; idx += 512) {
+    int row = idx >> 3;
+    int slot = idx & 7;
+    bf16x8 kv;
+    if (row < S) {
+      kv = *(const bf16x8*)(K + (int64_t)row * lkd + slot * 8);
+    } else {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) kv[e] = (__bf16)0.f;
+    }
+    *(bf16x8*)(k_lds + row * AT_D + ((slot ^ (row & 7)) << 3)) = kv;
+    bf16x8 vv;
+    if (row < S) {
+      vv = *(const bf16x8*)(V + (int64_t)row * lvd + slot * 8);
+    } else {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) vv[e] = (__bf16)0.f;
+    }
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      int dd = (e + lane) & 7;
+      vt_lds[(slot * 8 + dd) * vt_ld + row] = vv[dd];
+    }
+  }
+
+  // ---- Q fragments (B-operand now; same per-lane map as v1's A) ----
+  const int q0 = wave * AT_QW;
+  bf16x8 qf[AT_QM][2];
+#pragma unroll
+  for (int m = 0; m < AT_QM; ++m)
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      int row = q0 + m * 16 + (lane & 15);
+      int d = kk * 32 + (lane >> 4) * 8;
+      if (row < S) {
+        qf[m][kk] = *(const bf16x8*)(Q + (int64_t)row * lqd + d);
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          qf[m][kk][e] = (__bf16)(bf2f(qf[m][kk][e]) * scale);
+      } else {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) qf[m][kk][e] = (__bf16)0.f;
+      }
+    }
+
+  __syncthreads();
+
+  // per-lane online-softmax state: ONE scalar per query group (the
+  // lane's query is qn*16 + (lane&15))
+  float m2[AT_QM], l2[AT_QM];
+#pragma unroll
+  for (int qn = 0; qn < AT_QM; ++qn) {
+    m2[qn] = -3.4e38f;
+    l2[qn] = 0.f;
+  }
+  f32x4 oacc[AT_QM][4];
+#pragma unroll
+  for (int m = 0; m < AT_QM; ++m)
+#pragma unroll
+    for (int n = 0; n < 4; ++n) oacc[m][n] = {0.f, 0.f, 0.f, 0.f};
+
+  const int ntile = Sp / AT_KT;
+  for (int t = 0; t < ntile; ++t) {
+    const int kbase = t * AT_KT;
+    // ---- S^T block = K_t @ Q^T : sacc[km][qn], rows = keys ----
+    f32x4 sacc[2][AT_QM];
+#pragma unroll
+    for (int km = 0; km < 2; ++km)
+#pragma unroll
+      for (int qn = 0; qn < AT_QM; ++qn) sacc[km][qn] = {0, 0, 0, 0};
+#pragma unroll
+    for (int km = 0; km < 2; ++km) {
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        int krow = kbase + km * 16 + (lane & 15);
+        int slot = kk * 4 + (lane >> 4);
+        bf16x8 kf = *(bf16x8*)(k_lds + krow * AT_D +
+                               ((slot ^ (krow & 7)) << 3));
+#pragma unroll
+        for (int qn = 0; qn < AT_QM; ++qn)
+          sacc[km][qn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              kf, qf[qn][kk], sacc[km][qn], 0, 0, 0);
+      }
+    }
+    // mask padded keys: key = kbase + km*16 + jq*4 + r
+#pragma unroll
+    for (int km = 0; km < 2; ++km)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int key = kbase + km * 16 + jq * 4 + r;
+        if (key >= S)
+#pragma unroll
+          for (int qn = 0; qn < AT_QM; ++qn)
+            sacc[km][qn][r] = -3.4e38f;
+      }
+    // ---- per-query stats + P (all in registers) ----
+#pragma unroll
+    for (int qn = 0; qn < AT_QM; ++qn) {
+      float mx = sacc[0][qn][0];
+#pragma unroll
+      for (int km = 0; km < 2; ++km)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          mx = fmaxf(mx, sacc[km][qn][r]);
+      mx = fmaxf(mx, __shfl_xor(mx, 16, 64));
+      mx = fmaxf(mx, __shfl_xor(mx, 32, 64));
+      const float mold = m2[qn];
+      const float mnew = fmaxf(mold, mx);
+      const float alpha = __expf(mold - mnew);
+      float rs = 0.f;
+#pragma unroll
+      for (int km = 0; km < 2; ++km)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float p = __expf(sacc[km][qn][r] - mnew);
+          sacc[km][qn][r] = p;
+          rs += p;
+        }
+      rs += __shfl_xor(rs, 16, 64);
+      rs += __shfl_xor(rs, 32, 64);
+      l2[qn] = l2[qn] * alpha + rs;
+      m2[qn] = mnew;
+      // rescale O rows: oacc rows are q' = (lane>>4)*4 + r; fetch that
+      // query's alpha from the lane holding it (same quarter)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int src = (lane & 48) | ((jq * 4 + r) & 15);
+        float a_r = __builtin_bit_cast(
+            float, __builtin_amdgcn_ds_bpermute(
+                       src << 2, __builtin_bit_cast(int, alpha)));
+#pragma unroll
+        for (int n = 0; n < 4; ++n) oacc[qn][n][r] *= a_r;
+      }
+      // ---- pack P -> PV A-fragment via cvt_pk + bpermute ----
+      // c[i]: bf16x2 of keys (jq*4 + 2i', +1) for i<2 (km=0) and
+      // 16 + ... for i>=2 (km=1)
+      uint32_t c[4];
+#pragma unroll
+      for (int km = 0; km < 2; ++km)
+#pragma unroll
+        for (int pr = 0; pr < 2; ++pr) {
+          uint32_t d;
+          asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2"
+                       : "=v"(d)
+                       : "v"(sacc[km][qn][pr * 2]),
+                         "v"(sacc[km][qn][pr * 2 + 1]));
+          c[km * 2 + pr] = d;
+        }
+      // target lane jq' wants quarters {2(jq'&1), 2(jq'&1)+1} of
+      // km = jq'>>1 at its own lane&15
+      const uint32_t sel0 = jq >= 2 ? c[2] : c[0];
+      const uint32_t sel1 = jq >= 2 ? c[3] : c[1];
+      const int lo = ((lane & 15) | ((2 * (jq & 1)) << 4)) << 2;
+      const int hi = ((lane & 15) | ((2 * (jq & 1) + 1) << 4)) << 2;
+      union {
+        uint32_t u[4];
+        bf16x8 v;
+      } pf;
+      pf.u[0] = (uint32_t)__builtin_amdgcn_ds_bpermute(lo, (int)sel0);
+      pf.u[1] = (uint32_t)__builtin_amdgcn_ds_bpermute(lo, (int)sel1);
+      pf.u[2] = (uint32_t)__builtin_amdgcn_ds_bpermute(hi, (int)sel0);
+      pf.u[3] = (uint32_t)__builtin_amdgcn_ds_bpermute(hi, (int)sel1);
+      // ---- PV: O[qn] += P @ Vt^T ----
+#pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        int drow = n * 16 + (lane & 15);
+        bf16x8 vf = *(const bf16x8*)(vt_lds + drow * vt_ld + kbase +
+                                     (lane >> 4) * 8);
+        oacc[qn][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            pf.v, vf, oacc[qn][n], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- epilogue: O / l (broadcast inv like alpha), store ----
+#pragma unroll
+  for (int m = 0; m < AT_QM; ++m) {
+    const float invl = 1.f / l2[m];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int qr = q0 + m * 16 + jq * 4 + r;
+      if (qr >= S) continue;
+      int src = (lane & 48) | ((jq * 4 + r) & 15);
+      float inv = __builtin_bit_cast(
+          float, __builtin_amdgcn_ds_bpermute(
+                     src << 2, __builtin_bit_cast(int, invl)));
+#pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        int d = n * 16 + (lane & 15);
+        O[(int64_t)qr * lod + d] = (__bf16)(oacc[m][n][r] * inv);
+      }
+    }
+  }
+}
+
 void launch_attn_fused_bf16(const void* Q, const void* K, const void* V,
                             void* O, int batch1, int batch2, int S,
                             int64_t sQ1, int64_t sQ2, int lqd, int64_t sK1,
@@ -256,6 +488,15 @@ void launch_attn_fused_bf16(const void* Q, const void* K, const void* V,
   size_t lds = (size_t)Sp * AT_D * 2 + (size_t)AT_D * vt_ld * 2 +
                8 * AT_QW * AT_PLD * 2;
   dim3 grid(batch1 * batch2);
+  static const char* v2 = getenv("TWD_ATTN2");
+  if (v2 && v2[0] == '1') {
+    hipLaunchKernelGGL(attn_fused2_kernel, grid, dim3(512), lds, stream,
+                       (const __bf16*)Q, (const __bf16*)K,
+                       (const __bf16*)V, (__bf16*)O, batch2, S, Sp, sQ1,
+                       sQ2, lqd, sK1, sK2, lkd, sV1, sV2, lvd, sO1, sO2,
+                       lod, scale);
+    return;
+  }
   hipLaunchKernelGGL(attn_fused_kernel, grid, dim3(512), lds, stream,
                      (const __bf16*)Q, (const __bf16*)K, (const __bf16*)V,
                      (__bf16*)O, batch2, S, Sp, sQ1, sQ2, lqd, sK1, sK2,
