@@ -431,19 +431,27 @@ void attn_fused2_kernel(const __bf16* __restrict__ Q,
           c[km * 2 + pr] = d;
         }
       // target lane jq' wants quarters {2(jq'&1), 2(jq'&1)+1} of
-      // km = jq'>>1 at its own lane&15
-      const uint32_t sel0 = jq >= 2 ? c[2] : c[0];
-      const uint32_t sel1 = jq >= 2 ? c[3] : c[1];
+      // km = jq'>>1 at its own lane&15. bpermute contributions come
+      // from the SOURCE lane, so permute every c-reg and select the
+      // km half on the RECEIVER side (a source-side select picked the
+      // wrong half for jq' in {1,2} — first-debug lesson).
       const int lo = ((lane & 15) | ((2 * (jq & 1)) << 4)) << 2;
       const int hi = ((lane & 15) | ((2 * (jq & 1) + 1) << 4)) << 2;
+      uint32_t bl[4], bh[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        bl[i] = (uint32_t)__builtin_amdgcn_ds_bpermute(lo, (int)c[i]);
+        bh[i] = (uint32_t)__builtin_amdgcn_ds_bpermute(hi, (int)c[i]);
+      }
+      const bool kmHi = jq >= 2;
       union {
         uint32_t u[4];
         bf16x8 v;
       } pf;
-      pf.u[0] = (uint32_t)__builtin_amdgcn_ds_bpermute(lo, (int)sel0);
-      pf.u[1] = (uint32_t)__builtin_amdgcn_ds_bpermute(lo, (int)sel1);
-      pf.u[2] = (uint32_t)__builtin_amdgcn_ds_bpermute(hi, (int)sel0);
-      pf.u[3] = (uint32_t)__builtin_amdgcn_ds_bpermute(hi, (int)sel1);
+      pf.u[0] = kmHi ? bl[2] : bl[0];
+      pf.u[1] = kmHi ? bl[3] : bl[1];
+      pf.u[2] = kmHi ? bh[2] : bh[0];
+      pf.u[3] = kmHi ? bh[3] : bh[1];
       // ---- PV: O[qn] += P @ Vt^T ----
 #pragma unroll
       for (int n = 0; n < 4; ++n) {
