@@ -423,12 +423,13 @@ void attn_fused2_kernel(const __bf16* __restrict__ Q,
       for (int km = 0; km < 2; ++km)
 #pragma unroll
         for (int pr = 0; pr < 2; ++pr) {
-          uint32_t d;
-          asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2"
-                       : "=v"(d)
-                       : "v"(sacc[km][qn][pr * 2]),
-                         "v"(sacc[km][qn][pr * 2 + 1]));
-          c[km * 2 + pr] = d;
+          // plain RNE casts; the v_cvt_pk_bf16_f32 asm form corrupted
+          // one packed half under register pressure (debug iteration 4)
+          uint32_t lo16 = (uint32_t)__builtin_bit_cast(
+              uint16_t, (__bf16)sacc[km][qn][pr * 2]);
+          uint32_t hi16 = (uint32_t)__builtin_bit_cast(
+              uint16_t, (__bf16)sacc[km][qn][pr * 2 + 1]);
+          c[km * 2 + pr] = lo16 | (hi16 << 16);
         }
       // target lane jq' wants quarters {2(jq'&1), 2(jq'&1)+1} of
       // km = jq'>>1 at its own lane&15. bpermute contributions come
