@@ -388,9 +388,28 @@ void attn_fused2_kernel(const __bf16* __restrict__ Q,
           mx = fmaxf(mx, sacc[km][qn][r]);
       mx = fmaxf(mx, __shfl_xor(mx, 16, 64));
       mx = fmaxf(mx, __shfl_xor(mx, 32, 64));
+      // defer-max (guide T13): keep the running max while the tile's
+      // max stays within THR — skips the alpha exp, the broadcast and
+      // the O rescale on most tiles. Exact modulo normalization scale
+      // (P bounded by e^THR; f32 accumulators); the decision covers
+      // THIS tile's P, taken wave-uniformly before exponentiation.
       const float mold = m2[qn];
-      const float mnew = fmaxf(mold, mx);
-      const float alpha = __expf(mold - mnew);
+      float mnew = mold, alpha = 1.f;
+      if (__any(mx > mold + 8.f)) {
+        mnew = fmaxf(mold, mx);
+        alpha = __expf(mold - mnew);
+        // rescale O rows: oacc rows are q' = (lane>>4)*4 + r; fetch
+        // that query's alpha from the lane holding it (same quarter)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int src = (lane & 48) | ((jq * 4 + r) & 15);
+          float a_r = __builtin_bit_cast(
+              float, __builtin_amdgcn_ds_bpermute(
+                         src << 2, __builtin_bit_cast(int, alpha)));
+#pragma unroll
+          for (int n = 0; n < 4; ++n) oacc[qn][n][r] *= a_r;
+        }
+      }
       float rs = 0.f;
 #pragma unroll
       for (int km = 0; km < 2; ++km)
@@ -404,17 +423,6 @@ void attn_fused2_kernel(const __bf16* __restrict__ Q,
       rs += __shfl_xor(rs, 32, 64);
       l2[qn] = l2[qn] * alpha + rs;
       m2[qn] = mnew;
-      // rescale O rows: oacc rows are q' = (lane>>4)*4 + r; fetch that
-      // query's alpha from the lane holding it (same quarter)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int src = (lane & 48) | ((jq * 4 + r) & 15);
-        float a_r = __builtin_bit_cast(
-            float, __builtin_amdgcn_ds_bpermute(
-                       src << 2, __builtin_bit_cast(int, alpha)));
-#pragma unroll
-        for (int n = 0; n < 4; ++n) oacc[qn][n][r] *= a_r;
-      }
       // ---- pack P -> PV A-fragment via cvt_pk + bpermute ----
       // c[i]: bf16x2 of keys (jq*4 + 2i', +1) for i<2 (km=0) and
       // 16 + ... for i>=2 (km=1)
