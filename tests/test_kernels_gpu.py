@@ -383,3 +383,33 @@ def test_pool_s1_same(E, maxp, H, C):
     else:
         E.avgpool_bf16(Xg, Y, 3, 3, 1, 1, 1, 1)
     torch.testing.assert_close(Y.float().cpu(), ref, rtol=2e-2, atol=2e-2)
+
+
+def test_attn_defer_max_spiked(E):
+    """T13 defer-max hazard coverage (guide §5.4 rule 26): a huge key
+    arriving LATE must flip the rescale branch and still match the
+    fp32 reference within the documented defer-path bound."""
+    torch.manual_seed(1)
+    B, H, S, D = 4, 2, 196, 64
+    q = (torch.randn(B, S, H, D) * 0.5).bfloat16().cuda()
+    k = (torch.randn(B, S, H, D) * 0.5).bfloat16().cuda()
+    v = (torch.randn(B, S, H, D) * 0.5).bfloat16().cuda()
+    # spike one key in the LAST tile and one matching query so the
+    # running max jumps past THR mid-stream
+    k[0, 180, 0] = 10.0
+    q[0, 3, 0] = 10.0
+    o = torch.zeros_like(q)
+    E.attn_fused_bf16(q, k, v, o, B, H, S,
+                      q.stride(0), q.stride(2), q.stride(1),
+                      k.stride(0), k.stride(2), k.stride(1),
+                      v.stride(0), v.stride(2), v.stride(1),
+                      o.stride(0), o.stride(2), o.stride(1), 0.125)
+    torch.cuda.synchronize()
+    qf = q.float().permute(0, 3, 1, 2)
+    qf = q.float().permute(0, 2, 1, 3)
+    kf = k.float().permute(0, 2, 1, 3)
+    vf = v.float().permute(0, 2, 1, 3)
+    ref = (torch.softmax(qf @ kf.transpose(-1, -2) * 0.125, -1)
+           @ vf).permute(0, 2, 1, 3)
+    err = (o.float() - ref).abs().max().item()
+    assert err < 2e-2, err
