@@ -405,7 +405,6 @@ def test_attn_defer_max_spiked(E):
                       v.stride(0), v.stride(2), v.stride(1),
                       o.stride(0), o.stride(2), o.stride(1), 0.125)
     torch.cuda.synchronize()
-    qf = q.float().permute(0, 3, 1, 2)
     qf = q.float().permute(0, 2, 1, 3)
     kf = k.float().permute(0, 2, 1, 3)
     vf = v.float().permute(0, 2, 1, 3)
