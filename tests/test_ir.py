@@ -262,3 +262,51 @@ def test_tile_cluster_remap_bijective():
                 (tiles_m, tiles_n, hw_bid, tm, tn)
             seen.add((tm, tn))
         assert len(seen) == nwg, (tiles_m, tiles_n, len(seen))
+
+
+def test_stream_schedule_event_wiring():
+    """_build_stream_schedule (pure logic): cross-stream edges must get
+    record/wait events, deps must pass THROUGH step-less nodes
+    (eliminated concats / view reshapes), and fused-residual attrs
+    count as edges — the two round-2 bug classes."""
+    from tensorflow_web_deploy_amd.engine.gpu_executor import GpuExecutor
+    from tensorflow_web_deploy_amd.engine.ir import Graph, Node
+
+    # fork: x -> (a-chain | b-chain) -> cat(step-less) -> tail
+    # tail also takes a 'residual' attr ref to b.
+    nodes = [
+        Node("x", "Placeholder", []),
+        Node("a", "ConvBlock", ["x"], {}),
+        Node("b", "ConvBlock", ["x"], {}),
+        Node("cat", "ConcatV2", ["a", "b"], {}),
+        Node("tail", "ConvBlock", ["cat"], {"residual": "b"}),
+    ]
+    g = Graph(nodes, {}, ["x"], ["tail"])
+
+    class _P:
+        pass
+
+    plan = _P()
+    # spans: x steps [0,1), a [1,2), b [2,3), cat NO steps, tail [3,4)
+    spans = [("x", 0), ("a", 1), ("b", 2), ("cat", 3), ("tail", 3)]
+    ex = GpuExecutor.__new__(GpuExecutor)  # no GPU init needed
+    ex._build_stream_schedule(plan, g, spans, 4)
+    assert plan.sched is not None
+    sched = {i: (sid, waits, rec) for (i, sid, waits, rec) in plan.sched}
+    # a and b are sibling heavy consumers of x -> different streams
+    sid_a, sid_b = sched[1][0], sched[2][0]
+    assert {sid_a, sid_b} == {0, 1}
+    # whichever chain is on stream 1 must RECORD an event, and tail
+    # (stream 0, reading through the step-less cat AND the residual)
+    # must WAIT on it
+    side_step = 1 if sid_a == 1 else 2
+    rec = sched[side_step][2]
+    assert rec is not None
+    assert rec in sched[3][1], (plan.sched,)
+    # the stream-0 sibling needs no event for the stream-0 consumer
+    main_step = 2 if side_step == 1 else 1
+    # x (stream 0) -> side chain start must wait on x's event only if
+    # x recorded one; x is a producer consumed cross-stream -> event
+    assert sched[0][2] is not None
+    assert sched[0][2] in sched[side_step][1]
+    _ = main_step
