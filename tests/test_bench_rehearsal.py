@@ -48,3 +48,26 @@ def test_bench_single_process_cpu():
     rec = json.loads(out.stdout.strip().splitlines()[-1])
     assert rec["metric"] == "images_per_sec"
     assert rec["n_gpus"] == 1
+
+
+@pytest.mark.timeout(900)
+def test_bench_torchrun_world8_cpu():
+    """Full-width rehearsal of the driver's 8-GPU SCALE launch line
+    (gloo instead of nccl): one JSON line, dp8 config."""
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
+         "--master-port", "29533", os.path.join(ROOT, "bench.py"),
+         "--gpus", "8", "--steps", "2", "--warmup", "1", "--batch", "4",
+         "--model", "lenet", "--device", "cpu"],
+        capture_output=True, text=True, timeout=840, env=env, cwd=ROOT)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [ln for ln in out.stdout.splitlines()
+             if ln.startswith("{") and '"metric"' in ln]
+    assert len(lines) == 1, out.stdout
+    rec = json.loads(lines[0])
+    assert rec["n_gpus"] == 8
+    assert rec["config"]["parallelism"] == "dp8"
