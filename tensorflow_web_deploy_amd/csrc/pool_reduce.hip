@@ -68,11 +68,12 @@ __global__ void pool_kernel(const __bf16* __restrict__ X,
 // VERTICAL STRIP of VS=4 outputs from VS+kh-1 register rowsums: 4.5
 // loads/output instead of 9, vertical reuse in registers. Measured
 // 2.3 -> ~5 TB/s class. KH template keeps register indexing static.
-template <bool MAX, int KH, int VS = 4>
+template <bool MAX, int KH>
 __global__ void pool_s1_kernel(const __bf16* __restrict__ X,
                                __bf16* __restrict__ Y, int Nimg, int H,
                                int W, int C, int kw, int pt, int pl,
                                int HO, int WO) {
+  constexpr int VS = 4;
   const int c8 = C >> 3;
   const int strips = (HO + VS - 1) / VS;
   const int64_t total = (int64_t)Nimg * strips * WO * c8;
@@ -148,19 +149,9 @@ static bool pool_s1_dispatch(const void* X, void* Y, int Nimg, int H,
                              int W, int C, int kh, int kw, int pt, int pl,
                              int HO, int WO, hipStream_t stream) {
   if ((C & 7) != 0 || kh < 2 || kh > 5) return false;
-  // deeper strips amortize more vertical rowsum reuse on big maps
-  // (3x3: 18 loads / 8 outputs vs 18/4)
-  const bool deep = HO >= 16 && kh == 3;
-  const int vs = deep ? 8 : 4;
-  int strips = (HO + vs - 1) / vs;
+  int strips = (HO + 3) / 4;
   int64_t total = (int64_t)Nimg * strips * WO * (C >> 3);
   int blocks = (int)std::min<int64_t>((total + 255) / 256, 1 << 20);
-  if (deep) {
-    hipLaunchKernelGGL((pool_s1_kernel<MAX, 3, 8>), dim3(blocks),
-                       dim3(256), 0, stream, (const __bf16*)X, (__bf16*)Y,
-                       Nimg, H, W, C, kw, pt, pl, HO, WO);
-    return true;
-  }
   switch (kh) {
     case 2:
       hipLaunchKernelGGL((pool_s1_kernel<MAX, 2>), dim3(blocks), dim3(256),
